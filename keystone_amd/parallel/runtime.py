@@ -1,0 +1,211 @@
+"""Distributed runtime for keystone-mi355x.
+
+KeystoneML delegated its cluster runtime to Apache Spark (``SparkContext``,
+``RDD[T]``, ``treeReduce``/``treeAggregate``, broadcast, shuffle — ref:
+external Spark dependency).  The MI355X rebuild replaces
+that layer with an SPMD one-process-per-GPU runtime: every rank holds a
+row-shard of each dataset resident in its GPU's 288 GB HBM3E, and the
+aggregation patterns Spark expressed as ``treeReduce``/``treeAggregate``
+become RCCL collectives over xGMI (``torch.distributed`` backend ``nccl``,
+which IS RCCL on ROCm).
+
+Design notes (MI355X-first):
+  * xGMI is point-to-point (7 links x ~153 GB/s per GPU) — ring all-reduce is
+    bound by a single link, so for the dominant medium-size reductions
+    (per-block Gram matrices, ~134 MB fp32 at block=4096) we prefer
+    reduce-scatter + all-gather which stripes across links; RCCL picks this
+    internally for allreduce at these sizes, so we expose plain
+    ``all_reduce`` and tune bucket/block sizes at the call sites.
+  * "Broadcast a solved model block" is usually replaced by *replicated
+    deterministic local solves*: every rank runs the same Cholesky on the
+    same all-reduced Gram — zero communication, bitwise-identical by RCCL
+    determinism for a fixed topology.
+  * CPU testing: the identical code path runs under the ``gloo`` backend
+    with ``world_size`` 2..4 (Spark's ``local[k]`` test fixture analogue,
+    ref: LocalSparkContext test trait).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+__all__ = ["Runtime", "get_runtime", "init_runtime", "shutdown_runtime"]
+
+
+@dataclass
+class Runtime:
+    """Per-process handle on the SPMD job: rank, world size, device, and the
+    collective primitives the solver/stat estimators use.
+
+    Replaces Spark's driver/executor split: there is no driver — every rank
+    executes the same program, and "driver-side" solves are replicated.
+    """
+
+    rank: int = 0
+    world_size: int = 1
+    device: torch.device = field(default_factory=lambda: torch.device("cpu"))
+    backend: Optional[str] = None  # None => single-process, no collectives
+
+    # ---------------------------------------------------------------- info
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1 and dist.is_initialized()
+
+    @property
+    def is_gpu(self) -> bool:
+        return self.device.type == "cuda"
+
+    def barrier(self) -> None:
+        if self.is_distributed:
+            dist.barrier()
+
+    # --------------------------------------------------------- collectives
+    # Spark treeReduce(_+_) / treeAggregate  ->  RCCL all-reduce over xGMI.
+    def all_reduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
+        """In-place sum/max/min across ranks. No-op when single-process."""
+        if self.is_distributed:
+            red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+                   "min": dist.ReduceOp.MIN}[op]
+            dist.all_reduce(t, op=red)
+        return t
+
+    def all_reduce_many_(self, tensors: List[torch.Tensor], op: str = "sum") -> None:
+        """Fused all-reduce of several tensors (one flat buffer => one RCCL
+        call; stripes better across the 7 xGMI links than many small calls)."""
+        if not self.is_distributed or not tensors:
+            return
+        flat = torch.cat([t.reshape(-1) for t in tensors])
+        self.all_reduce_(flat, op)
+        off = 0
+        for t in tensors:
+            n = t.numel()
+            t.copy_(flat[off:off + n].view_as(t))
+            off += n
+
+    def all_gather(self, t: torch.Tensor) -> List[torch.Tensor]:
+        if not self.is_distributed:
+            return [t]
+        outs = [torch.empty_like(t) for _ in range(self.world_size)]
+        dist.all_gather(outs, t.contiguous())
+        return outs
+
+    def all_gather_rows(self, t: torch.Tensor) -> torch.Tensor:
+        """Gather row-sharded 2-D tensors with possibly unequal first dims."""
+        if not self.is_distributed:
+            return t
+        n_local = torch.tensor([t.shape[0]], dtype=torch.int64)
+        counts = [torch.zeros(1, dtype=torch.int64) for _ in range(self.world_size)]
+        dist.all_gather(counts, n_local)
+        counts = [int(c.item()) for c in counts]
+        n_max = max(counts)
+        pad = t
+        if t.shape[0] < n_max:
+            pad = torch.zeros((n_max,) + t.shape[1:], dtype=t.dtype, device=t.device)
+            pad[: t.shape[0]] = t
+        outs = [torch.empty_like(pad) for _ in range(self.world_size)]
+        dist.all_gather(outs, pad.contiguous())
+        return torch.cat([o[:c] for o, c in zip(outs, counts)], dim=0)
+
+    def broadcast_(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if self.is_distributed:
+            dist.broadcast(t, src=src)
+        return t
+
+    def broadcast_obj(self, obj, src: int = 0):
+        if not self.is_distributed:
+            return obj
+        box = [obj if self.rank == src else None]
+        dist.broadcast_object_list(box, src=src)
+        return box[0]
+
+    def all_to_all(self, send: List[torch.Tensor]) -> List[torch.Tensor]:
+        """Keyed exchange (Spark shuffle analogue) for GPU-resident rows."""
+        if not self.is_distributed:
+            return send
+        sizes = [torch.tensor([s.shape[0]], dtype=torch.int64) for s in send]
+        recv_sizes = [torch.zeros(1, dtype=torch.int64) for _ in range(self.world_size)]
+        dist.all_to_all(recv_sizes, sizes)
+        trailing = send[0].shape[1:]
+        recv = [torch.empty((int(n.item()),) + trailing, dtype=send[0].dtype,
+                            device=send[0].device) for n in recv_sizes]
+        dist.all_to_all(recv, [s.contiguous() for s in send])
+        return recv
+
+    # ------------------------------------------------------------- helpers
+    def shard_bounds(self, n_total: int) -> tuple:
+        """Row range [lo, hi) owned by this rank under balanced sharding."""
+        base, rem = divmod(n_total, self.world_size)
+        lo = self.rank * base + min(self.rank, rem)
+        hi = lo + base + (1 if self.rank < rem else 0)
+        return lo, hi
+
+    def synchronize(self) -> None:
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+
+
+_RUNTIME: Optional[Runtime] = None
+
+
+def init_runtime(device: Optional[str] = None, backend: Optional[str] = None,
+                 timeout_s: int = 600) -> Runtime:
+    """Initialise the process's runtime from torchrun-style env vars.
+
+    Single-process (no RANK in env): a trivial runtime, no process group.
+    Multi-process: ``nccl`` (=RCCL) when CUDA is available, else ``gloo``.
+    """
+    global _RUNTIME
+    if _RUNTIME is not None:
+        return _RUNTIME
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_gpu = torch.cuda.is_available() if device is None else device.startswith("cuda")
+    if use_gpu:
+        dev = torch.device("cuda", local_rank % max(1, torch.cuda.device_count()))
+        torch.cuda.set_device(dev)
+    else:
+        dev = torch.device("cpu")
+
+    if world > 1:
+        if backend is None:
+            backend = "nccl" if use_gpu else "gloo"
+        if not dist.is_initialized():
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29571")
+            dist.init_process_group(backend=backend, rank=rank, world_size=world,
+                                    timeout=datetime.timedelta(seconds=timeout_s))
+    else:
+        backend = None
+
+    _RUNTIME = Runtime(rank=rank, world_size=world, device=dev, backend=backend)
+    return _RUNTIME
+
+
+def get_runtime() -> Runtime:
+    """The process-wide runtime (lazily created single-process if unset)."""
+    global _RUNTIME
+    if _RUNTIME is None:
+        _RUNTIME = init_runtime()
+    return _RUNTIME
+
+
+def set_runtime(rt: Runtime) -> None:
+    """Install an explicit runtime (used by in-process multi-rank tests)."""
+    global _RUNTIME
+    _RUNTIME = rt
+
+
+def shutdown_runtime() -> None:
+    global _RUNTIME
+    if _RUNTIME is not None and dist.is_initialized():
+        dist.destroy_process_group()
+    _RUNTIME = None

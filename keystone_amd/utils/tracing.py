@@ -1,0 +1,48 @@
+"""Structured timing/trace records (aux subsystem).
+
+KeystoneML had no always-on tracer (Spark UI + the sampling Profiler fed
+its AutoCacheRule).  Here the executor emits one record per operator
+execution; enable JSONL output with ``enable_trace(path)`` or env
+``KEYSTONE_TRACE=/path/trace.jsonl``.  GPU kernel-level profiling is done
+externally with rocprofv3 (see profiles/).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Optional, TextIO
+
+_trace_file: Optional[TextIO] = None
+_enabled = False
+
+
+def enable_trace(path: Optional[str] = None) -> None:
+    global _trace_file, _enabled
+    path = path or os.environ.get("KEYSTONE_TRACE")
+    if path:
+        _trace_file = open(path, "a")
+    _enabled = True
+
+
+def disable_trace() -> None:
+    global _trace_file, _enabled
+    if _trace_file:
+        _trace_file.close()
+    _trace_file = None
+    _enabled = False
+
+
+def trace_event(kind: str, **fields) -> None:
+    if not _enabled:
+        return
+    rec = {"t": time.time(), "kind": kind, **fields}
+    if _trace_file:
+        _trace_file.write(json.dumps(rec) + "\n")
+        _trace_file.flush()
+
+
+# auto-enable from env
+if os.environ.get("KEYSTONE_TRACE"):
+    enable_trace()
