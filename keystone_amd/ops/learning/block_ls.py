@@ -1,0 +1,268 @@
+"""Block-coordinate-descent least squares — the flagship distributed solver.
+
+ref: nodes.learning.BlockLeastSquaresEstimator,
+BlockLinearMapper, BlockWeightedLeastSquaresEstimator.  Ridge regression
+for very wide features (d up to 100k–400k): the feature dimension is split
+into blocks (default 4096 cols); each epoch, each block's Gram and
+residual-correlation are computed per rank and reduced, the block system is
+solved, and the residual is updated.  BCD with per-block regularizer
+``(X_b^T X_b + lam I) W_b = X_b^T (Y - sum_{b'!=b} X_b' W_b')`` has the
+exact joint ridge solution as its fixpoint (each update solves one block
+row of the joint normal equations), which the tests assert.
+
+MI355X mapping of the Spark comm pattern:
+  * per-block ``treeReduce`` of ``A_b^T A_b`` (4096^2 fp32 = 67 MB) and
+    ``A_b^T R`` -> ONE fused RCCL all-reduce over xGMI (reduce-scatter +
+    all-gather internally, striping the 7 p2p links);
+  * "driver Cholesky solve" -> replicated fp64 solve on every rank
+    (bitwise-deterministic, no broadcast needed);
+  * residual update -> local GEMM per rank (hipBLASLt), no communication.
+
+The weighted variant reproduces KeystoneML's class-rebalanced loss (the
+ImageNet/VOC FV solver): for score column c, rows of class c get weight
+``alpha/n_c`` and all rows get ``(1-alpha)/n``.  Spark shuffled rows by
+class (``groupBy``); here a one-time RCCL all-to-all re-shards rows so each
+rank owns whole classes, after which per-class Grams and the per-class
+batched Cholesky are rank-local and only the shared Gram and solved block
+columns cross xGMI.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional, Sequence
+
+import torch
+
+from ...parallel.dataset import Dataset, TensorDataset
+from ...workflow.pipeline import (BatchTransformer, LabelEstimator,
+                                  Transformer, TupleDataset)
+from .linear import _solve_spd
+
+__all__ = ["BlockLinearMapper", "BlockLeastSquaresEstimator",
+           "BlockWeightedLeastSquaresEstimator"]
+
+
+class BlockLinearMapper(Transformer):
+    """Apply block-partitioned weights sequentially (ref: BlockLinearMapper):
+    ``out = sum_b X_b W_b + intercept``.  Accepts a full feature tensor
+    (split internally) or the TupleDataset produced by VectorSplitter —
+    supporting the lazy per-block featurization pattern where 256k-dim rows
+    are never materialized at once."""
+
+    def __init__(self, blocks: List[torch.Tensor],
+                 intercept: Optional[torch.Tensor] = None):
+        self.blocks = blocks
+        self.intercept = intercept
+
+    @property
+    def block_sizes(self) -> List[int]:
+        return [int(b.shape[0]) for b in self.blocks]
+
+    def apply(self, x):
+        if isinstance(x, (tuple, list)):
+            parts = list(x)
+        else:
+            parts = list(torch.split(x, self.block_sizes, dim=-1))
+        out = None
+        for p, W in zip(parts, self.blocks):
+            W = W.to(device=p.device, dtype=p.dtype)
+            out = p @ W if out is None else out + p @ W
+        if self.intercept is not None:
+            out = out + self.intercept.to(device=out.device, dtype=out.dtype)
+        return out
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        return self.apply(X)
+
+    def apply_dataset(self, ds: Dataset):
+        if isinstance(ds, TupleDataset):
+            out = None
+            for part, W in zip(ds.parts, self.blocks):
+                X = part.data
+                W = W.to(device=X.device, dtype=X.dtype)
+                out = X @ W if out is None else out + X @ W
+            if self.intercept is not None:
+                out = out + self.intercept.to(device=out.device, dtype=out.dtype)
+            return TensorDataset(out, ds.runtime)
+        if isinstance(ds, TensorDataset):
+            return ds.map_batch(self.apply_batch)
+        return ds.map(self.apply)
+
+    # persistence: store blocks as stacked state
+    def get_state(self):
+        st = {f"block_{i}": b for i, b in enumerate(self.blocks)}
+        if self.intercept is not None:
+            st["intercept"] = self.intercept
+        return st
+
+    def get_config(self):
+        return {"n_blocks": len(self.blocks)}
+
+    @classmethod
+    def from_saved(cls, config, state):
+        blocks = [state[f"block_{i}"] for i in range(config["n_blocks"])]
+        return cls(blocks, state.get("intercept"))
+
+
+def _as_block_getters(data: Dataset, block_size: int):
+    """Return (list of zero-arg block getters, runtime, n_local)."""
+    if isinstance(data, TupleDataset):
+        parts = [p.data for p in data.parts]
+        return [lambda p=p: p for p in parts], data.runtime, parts[0].shape[0]
+    if isinstance(data, TensorDataset):
+        X = data.data
+        splits = list(torch.split(X, block_size, dim=-1))
+        return [lambda s=s: s for s in splits], data.runtime, X.shape[0]
+    raise TypeError("Block solvers need a TensorDataset or VectorSplitter output")
+
+
+class BlockLeastSquaresEstimator(LabelEstimator):
+    """ref: BlockLeastSquaresEstimator(blockSize, numIter, lambda)."""
+
+    def __init__(self, block_size: int = 4096, num_iter: int = 10,
+                 lam: float = 0.0, fit_intercept: bool = True):
+        self.block_size = block_size
+        self.num_iter = num_iter
+        self.lam = lam
+        self.fit_intercept = fit_intercept
+
+    def _fit(self, data: Dataset, labels: TensorDataset) -> BlockLinearMapper:
+        getters, rt, n_local = _as_block_getters(data, self.block_size)
+        Y = labels.data
+        if Y.dim() == 1:
+            Y = Y.unsqueeze(1)
+        dtype = getters[0]().dtype
+        Y = Y.to(dtype)
+
+        x_means, y_mean = None, None
+        if self.fit_intercept:
+            n = torch.tensor([float(n_local)], dtype=torch.float64, device=Y.device)
+            sums = [g().sum(0, dtype=torch.float64) for g in getters]
+            sy = Y.sum(0, dtype=torch.float64)
+            rt.all_reduce_many_([n, sy] + sums)
+            n_tot = n.item()
+            x_means = [(s / n_tot).to(dtype) for s in sums]
+            y_mean = (sy / n_tot).to(dtype)
+            Y = Y - y_mean
+
+        nb = len(getters)
+        Ws = [None] * nb
+        R = Y.clone()  # residual Y - X W, local shard rows
+        for _epoch in range(self.num_iter):
+            for b in range(nb):
+                Xb = getters[b]()
+                if x_means is not None:
+                    Xb = Xb - x_means[b]
+                # add this block's current contribution back into the target
+                T = R if Ws[b] is None else R + Xb @ Ws[b]
+                gram = Xb.T @ Xb            # [db, db] local GEMM (hipBLASLt)
+                rhs = Xb.T @ T              # [db, k]
+                rt.all_reduce_many_([gram, rhs])   # the #1 comm hot spot
+                Wb = _solve_spd(gram, rhs, self.lam)
+                R = T - Xb @ Wb
+                Ws[b] = Wb
+
+        intercept = None
+        if self.fit_intercept:
+            intercept = y_mean - sum(m @ W for m, W in zip(x_means, Ws))
+        return BlockLinearMapper(Ws, intercept)
+
+
+class BlockWeightedLeastSquaresEstimator(LabelEstimator):
+    """Class-rebalanced block least squares (ref:
+    BlockWeightedLeastSquaresEstimator(blockSize, numIter, lambda,
+    mixtureWeight) — the ImageNet/VOC FV solver).
+
+    Per-column weighted ridge: for score column c, row i has weight
+    ``w_ic = (1-alpha)/n + [class_i == c] * alpha/n_c``.  Rows are
+    re-sharded by class once (RCCL all-to-all) so per-class Grams and the
+    per-class solves are local; per block per epoch the cross-class terms
+    need one all-reduce of ``X_b^T X_b`` and one of ``X_b^T R~`` and an
+    all-gather of the solved block columns.
+    """
+
+    def __init__(self, block_size: int = 4096, num_iter: int = 10,
+                 lam: float = 0.0, mixture_weight: float = 0.5):
+        self.block_size = block_size
+        self.num_iter = num_iter
+        self.lam = lam
+        self.mixture_weight = mixture_weight
+
+    def _fit(self, data: Dataset, labels: TensorDataset) -> BlockLinearMapper:
+        getters, rt, n_local = _as_block_getters(data, self.block_size)
+        Y = labels.data
+        if Y.dim() == 1:
+            raise ValueError("weighted solver needs +-1 indicator labels [n, K]")
+        dtype = getters[0]().dtype
+        Y = Y.to(dtype)
+        K = Y.shape[1]
+        cls = Y.argmax(dim=1)
+        alpha = self.mixture_weight
+
+        # ---- one-time class re-shard (Spark groupBy(class) -> all-to-all)
+        X_full = torch.cat([g() for g in getters], dim=1)
+        if rt.is_distributed:
+            order = torch.argsort(cls % rt.world_size, stable=True)
+            X_full, Y, cls = X_full[order], Y[order], cls[order]
+            send_x, send_y, send_c = [], [], []
+            dest = cls % rt.world_size
+            for r in range(rt.world_size):
+                m = dest == r
+                send_x.append(X_full[m])
+                send_y.append(Y[m])
+                send_c.append(cls[m].unsqueeze(1).to(dtype))
+            X_full = torch.cat(rt.all_to_all(send_x), dim=0)
+            Y = torch.cat(rt.all_to_all(send_y), dim=0)
+            cls = torch.cat(rt.all_to_all(send_c), dim=0).squeeze(1).long()
+
+        n = torch.tensor([float(X_full.shape[0])], dtype=torch.float64,
+                         device=Y.device)
+        counts = torch.bincount(cls, minlength=K).to(torch.float64)
+        rt.all_reduce_many_([n, counts])
+        n_tot = n.item()
+        counts = counts.clamp_min(1.0)
+
+        own = [c for c in range(K) if (c % max(1, rt.world_size)) == rt.rank] \
+            if rt.is_distributed else list(range(K))
+        own_masks = {c: (cls == c) for c in own}
+
+        blocks = list(torch.split(X_full, self.block_size, dim=1))
+        nb = len(blocks)
+        base_w = (1.0 - alpha) / n_tot
+
+        Ws: List[Optional[torch.Tensor]] = [None] * nb
+        R = Y.clone()
+        for _epoch in range(self.num_iter):
+            for b in range(nb):
+                Xb = blocks[b]
+                db = Xb.shape[1]
+                T = R if Ws[b] is None else R + Xb @ Ws[b]
+                gram_all = Xb.T @ Xb                   # shared Gram
+                xtr_all = Xb.T @ T                     # [db, K]
+                rt.all_reduce_many_([gram_all, xtr_all])
+
+                Wb = torch.zeros(db, K, dtype=dtype, device=Xb.device)
+                if own:
+                    # batched per-class weighted systems, all rank-local
+                    grams, rhss = [], []
+                    for c in own:
+                        m = own_masks[c]
+                        Xc = Xb[m]
+                        wc = alpha / counts[c].item()
+                        G = base_w * gram_all + wc * (Xc.T @ Xc)
+                        r = base_w * xtr_all[:, c] + wc * (Xc.T @ T[m, c])
+                        grams.append(G)
+                        rhss.append(r)
+                    Gb = torch.stack(grams).to(torch.float64)
+                    rb = torch.stack(rhss).to(torch.float64).unsqueeze(-1)
+                    Gb += self.lam * torch.eye(db, dtype=torch.float64,
+                                               device=Gb.device)
+                    L = torch.linalg.cholesky(Gb)
+                    sol = torch.cholesky_solve(rb, L).squeeze(-1).to(dtype)
+                    for j, c in enumerate(own):
+                        Wb[:, c] = sol[j]
+                if rt.is_distributed:
+                    rt.all_reduce_(Wb)   # disjoint columns -> sum == gather
+                R = T - Xb @ Wb
+                Ws[b] = Wb
+        return BlockLinearMapper(Ws)

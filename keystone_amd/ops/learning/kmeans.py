@@ -1,0 +1,88 @@
+"""K-means++ (ref: nodes.learning.KMeansPlusPlusEstimator / KMeansModel,
+SURVEY.md §2.2).
+
+Assignment uses the distance-GEMM formulation ``||x||^2 + ||c||^2 - 2 x C^T``
+(one hipBLASLt GEMM per Lloyd iteration per shard); per-centroid (sum,
+count) accumulators are index_add'ed locally and all-reduced — Spark's
+aggregate became one fused RCCL call.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ...parallel.dataset import TensorDataset
+from ...workflow.pipeline import BatchTransformer, Estimator
+
+__all__ = ["KMeansPlusPlusEstimator", "KMeansModel"]
+
+
+class KMeansModel(BatchTransformer):
+    """x -> one-hot cluster indicator (KeystoneML's KMeansModel.apply)."""
+
+    def __init__(self, centers: torch.Tensor):
+        self.centers = centers  # [K, d]
+
+    def assign(self, X: torch.Tensor) -> torch.Tensor:
+        C = self.centers.to(device=X.device, dtype=X.dtype)
+        d2 = (X * X).sum(1, keepdim=True) - 2.0 * (X @ C.T) + (C * C).sum(1)
+        return d2.argmin(dim=1)
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        idx = self.assign(X)
+        out = torch.zeros(X.shape[0], self.centers.shape[0],
+                          dtype=X.dtype, device=X.device)
+        out.scatter_(1, idx.unsqueeze(1), 1.0)
+        return out
+
+    @classmethod
+    def from_saved(cls, config, state):
+        return cls(state["centers"])
+
+
+class KMeansPlusPlusEstimator(Estimator):
+    def __init__(self, num_clusters: int, max_iterations: int = 20,
+                 seed: int = 0, tol: float = 1e-6):
+        self.num_clusters = num_clusters
+        self.max_iterations = max_iterations
+        self.seed = seed
+        self.tol = tol
+
+    def _fit(self, data: TensorDataset) -> KMeansModel:
+        X = data.data
+        rt = data.runtime
+        K = self.num_clusters
+        dev, dt = X.device, X.dtype
+
+        # --- k-means++ init on a gathered sample (deterministic, replicated)
+        sample = data.sample_local(max(8 * K // max(1, rt.world_size), K),
+                                   self.seed)
+        S = sample.collect().to(torch.float64)
+        g = torch.Generator().manual_seed(self.seed)
+        centers = torch.empty(K, X.shape[1], dtype=torch.float64, device=S.device)
+        first = int(torch.randint(0, S.shape[0], (1,), generator=g).item())
+        centers[0] = S[first]
+        d2 = ((S - centers[0]) ** 2).sum(1)
+        for k in range(1, K):
+            probs = (d2 / d2.sum().clamp_min(1e-30)).cpu()
+            nxt = int(torch.multinomial(probs, 1, generator=g).item())
+            centers[k] = S[nxt]
+            d2 = torch.minimum(d2, ((S - centers[k]) ** 2).sum(1))
+        C = centers.to(device=dev, dtype=dt)
+
+        # --- Lloyd iterations
+        for _it in range(self.max_iterations):
+            d2m = (X * X).sum(1, keepdim=True) - 2.0 * (X @ C.T) + (C * C).sum(1)
+            idx = d2m.argmin(dim=1)
+            sums = torch.zeros(K, X.shape[1], dtype=torch.float64, device=dev)
+            sums.index_add_(0, idx, X.to(torch.float64))
+            cnt = torch.bincount(idx, minlength=K).to(torch.float64)
+            rt.all_reduce_many_([sums, cnt])
+            newC = torch.where(cnt.unsqueeze(1) > 0,
+                               sums / cnt.clamp_min(1.0).unsqueeze(1),
+                               C.to(torch.float64)).to(dt)
+            shift = float(((newC - C) ** 2).sum())
+            C = newC
+            if shift <= self.tol:
+                break
+        return KMeansModel(C)

@@ -1,0 +1,108 @@
+"""Small stat transformers.
+
+ref: nodes.stats.RandomSignNode, LinearRectifier,
+SignedHellingerMapper, NormalizeRows, ColumnSampler / Sampler.
+All elementwise/row-local ops subclass ElementwiseTransformer so the
+pipeline optimizer can fuse chains of them into one shard pass.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ...parallel.dataset import Dataset, ItemDataset, TensorDataset, from_rows
+from ...workflow.optimizer import ElementwiseTransformer
+from ...workflow.pipeline import Transformer
+
+
+class RandomSignNode(ElementwiseTransformer):
+    """Multiply each coordinate by a random fixed +-1 (ref: RandomSignNode)."""
+
+    def __init__(self, dim: int, seed: int = 0):
+        g = torch.Generator().manual_seed(seed)
+        self.dim = dim
+        self.seed = seed
+        self.signs = (torch.randint(0, 2, (dim,), generator=g,
+                                    dtype=torch.int8).float() * 2 - 1)
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        self.signs = self.signs.to(device=X.device, dtype=X.dtype)
+        return X * self.signs
+
+    @classmethod
+    def from_saved(cls, config, state):
+        obj = cls.__new__(cls)
+        obj.__dict__.update(config)
+        obj.signs = state["signs"]
+        return obj
+
+
+class LinearRectifier(ElementwiseTransformer):
+    """``max(x - alpha, 0.0) + maxVal`` style rectifier; KeystoneML's is
+    ``max(x, alpha)`` (ref: LinearRectifier(maxVal, alpha))."""
+
+    def __init__(self, max_val: float = 0.0, alpha: float = 0.0):
+        self.max_val = max_val
+        self.alpha = alpha
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        return torch.clamp(X - self.alpha, min=self.max_val)
+
+
+class SignedHellingerMapper(ElementwiseTransformer):
+    """``sign(x) * sqrt(|x|)`` — Fisher-Vector power normalization
+    (ref: SignedHellingerMapper)."""
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        return torch.sign(X) * torch.sqrt(torch.abs(X))
+
+
+class NormalizeRows(ElementwiseTransformer):
+    """Row-wise L2 normalization (ref: NormalizeRows)."""
+
+    def __init__(self, eps: float = 1e-12):
+        self.eps = eps
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        return X / X.norm(dim=-1, keepdim=True).clamp_min(self.eps)
+
+
+class Sampler(Transformer):
+    """Sample up to ``size`` items of the local shard (ref: Sampler)."""
+
+    def __init__(self, size: int, seed: int = 42):
+        self.size = size
+        self.seed = seed
+
+    def apply(self, x):
+        return x
+
+    def apply_dataset(self, ds: Dataset) -> Dataset:
+        # `size` is the TOTAL sample budget across ranks (KeystoneML sampled
+        # a global fraction); split it evenly.
+        per_rank = max(1, self.size // max(1, ds.runtime.world_size))
+        return ds.sample_local(per_rank, self.seed)
+
+
+class ColumnSampler(Transformer):
+    """From a dataset of [n_i, d] descriptor matrices, sample columns/rows to
+    build a flat sample matrix (ref: ColumnSampler — used to draw ~1M SIFT
+    descriptors for PCA/GMM fits)."""
+
+    def __init__(self, num_samples: int, seed: int = 42):
+        self.num_samples = num_samples
+        self.seed = seed
+
+    def apply(self, x: torch.Tensor) -> torch.Tensor:
+        return x
+
+    def apply_dataset(self, ds: Dataset) -> TensorDataset:
+        per_rank = max(1, self.num_samples // max(1, ds.runtime.world_size))
+        mats = [x for x in ds.items()]
+        if not mats:
+            raise ValueError("ColumnSampler: empty local shard")
+        all_rows = torch.cat([m.reshape(-1, m.shape[-1]) for m in mats], dim=0)
+        n = all_rows.shape[0]
+        g = torch.Generator().manual_seed(self.seed + 7919 * ds.runtime.rank)
+        idx = torch.randperm(n, generator=g)[:per_rank].to(all_rows.device)
+        return TensorDataset(all_rows[idx], ds.runtime)

@@ -1,0 +1,10 @@
+from .labels import (ClassLabelIndicatorsFromIntLabels,
+                     ClassLabelIndicatorsFromStringLabels, MaxClassifier,
+                     TopKClassifier)
+from .vectors import (Cast, Densify, Sparsify, VectorCombiner, VectorSplitter)
+
+__all__ = [
+    "ClassLabelIndicatorsFromIntLabels", "ClassLabelIndicatorsFromStringLabels",
+    "MaxClassifier", "TopKClassifier", "VectorCombiner", "VectorSplitter",
+    "Densify", "Sparsify", "Cast",
+]

@@ -125,9 +125,16 @@ class Runtime:
         return box[0]
 
     def all_to_all(self, send: List[torch.Tensor]) -> List[torch.Tensor]:
-        """Keyed exchange (Spark shuffle analogue) for GPU-resident rows."""
+        """Keyed exchange (Spark shuffle analogue) for GPU-resident rows.
+        RCCL all-to-all on GPU; gloo has no alltoall, so the CPU test path
+        exchanges through all_gather_object."""
         if not self.is_distributed:
             return send
+        if self.backend == "gloo":
+            gathered: List[Optional[list]] = [None] * self.world_size
+            dist.all_gather_object(gathered, [t.cpu() for t in send])
+            return [gathered[src][self.rank].to(send[0].device)
+                    for src in range(self.world_size)]
         sizes = [torch.tensor([s.shape[0]], dtype=torch.int64) for s in send]
         recv_sizes = [torch.zeros(1, dtype=torch.int64) for _ in range(self.world_size)]
         dist.all_to_all(recv_sizes, sizes)

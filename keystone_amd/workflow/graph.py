@@ -237,8 +237,9 @@ class Graph:
         return [nid for nid, deps in self.dependencies.items() if gid in deps]
 
     def topo_order(self) -> List[NodeId]:
-        """Topological order over operator nodes reachable from sinks."""
-        live = self.reachable_from_sinks()
+        """Topological order over operator nodes (reachable from sinks when
+        sinks exist; all nodes for unbound prefix pipelines)."""
+        live = self.reachable_from_sinks() if self.sinks else set(self.operators)
         seen, order = set(), []
 
         def visit(g):

@@ -1,0 +1,25 @@
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD MI355X GPU (run with -m gpu)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture
+def fresh_runtime():
+    """Single-process runtime, reset around the test."""
+    from keystone_amd.parallel import runtime as rt_mod
+    rt_mod.shutdown_runtime()
+    yield rt_mod.get_runtime()
+    rt_mod.shutdown_runtime()

@@ -1,0 +1,164 @@
+"""Multi-process distributed semantics on CPU/gloo (ref: LocalSparkContext
+local[2..4] test pattern).  Each test asserts the 2-rank fit
+equals the single-process fit on the concatenated data — the invariant that
+makes the RCCL path correct by construction."""
+
+import pytest
+import torch
+
+from tests.dist_util import run_distributed
+
+torch.manual_seed(0)
+N, D, K = 64, 12, 3
+X_ALL = torch.randn(N, D, dtype=torch.float64)
+W_TRUE = torch.randn(D, K, dtype=torch.float64)
+Y_ALL = X_ALL @ W_TRUE + 0.1 * torch.randn(N, K, dtype=torch.float64)
+LAB_ALL = torch.randint(0, K, (N,))
+
+
+def _shard(rank, world, t):
+    base = N // world
+    lo = rank * base
+    hi = N if rank == world - 1 else lo + base
+    return t[lo:hi]
+
+
+def _linear_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import LinearMapEstimator
+    ds = ka.from_rows(_shard(rank, world, X_ALL))
+    ys = ka.from_rows(_shard(rank, world, Y_ALL))
+    m = LinearMapEstimator(lam=0.5).fit(ds, ys)
+    return m.W
+
+
+def test_linear_map_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import LinearMapEstimator
+    ref = LinearMapEstimator(lam=0.5).fit(
+        ka.from_rows(X_ALL), ka.from_rows(Y_ALL)).W
+    results = run_distributed(_linear_fit, 2)
+    for W in results:
+        assert torch.allclose(W, ref, atol=1e-9)
+    # ranks agree bitwise (replicated solve on all-reduced Gram)
+    assert torch.equal(results[0], results[1])
+
+
+def _block_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    ds = ka.from_rows(_shard(rank, world, X_ALL))
+    ys = ka.from_rows(_shard(rank, world, Y_ALL))
+    m = BlockLeastSquaresEstimator(block_size=5, num_iter=30, lam=0.5).fit(ds, ys)
+    return torch.cat(m.blocks, dim=0), m.intercept
+
+
+def test_block_ls_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    ref = BlockLeastSquaresEstimator(block_size=5, num_iter=30, lam=0.5).fit(
+        ka.from_rows(X_ALL), ka.from_rows(Y_ALL))
+    ref_W = torch.cat(ref.blocks, dim=0)
+    results = run_distributed(_block_fit, 2)
+    for W, b in results:
+        assert torch.allclose(W, ref_W, atol=1e-8)
+        assert torch.allclose(b, ref.intercept, atol=1e-8)
+
+
+def _weighted_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    Y_ind = torch.full((N, K), -1.0, dtype=torch.float64)
+    Y_ind[torch.arange(N), LAB_ALL] = 1.0
+    ds = ka.from_rows(_shard(rank, world, X_ALL))
+    ys = ka.from_rows(_shard(rank, world, Y_ind))
+    m = BlockWeightedLeastSquaresEstimator(
+        block_size=6, num_iter=20, lam=1e-2, mixture_weight=0.5).fit(ds, ys)
+    return torch.cat(m.blocks, dim=0)
+
+
+def test_block_weighted_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    Y_ind = torch.full((N, K), -1.0, dtype=torch.float64)
+    Y_ind[torch.arange(N), LAB_ALL] = 1.0
+    ref = BlockWeightedLeastSquaresEstimator(
+        block_size=6, num_iter=20, lam=1e-2, mixture_weight=0.5).fit(
+        ka.from_rows(X_ALL), ka.from_rows(Y_ind))
+    ref_W = torch.cat(ref.blocks, dim=0)
+    results = run_distributed(_weighted_fit, 2)
+    for W in results:
+        assert torch.allclose(W, ref_W, atol=1e-6)
+
+
+def _scaler_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.stats import StandardScaler
+    ds = ka.from_rows(_shard(rank, world, X_ALL))
+    m = StandardScaler().fit(ds)
+    return m.mean, m.std
+
+
+def test_standard_scaler_distributed():
+    results = run_distributed(_scaler_fit, 2)
+    mean_ref = X_ALL.mean(0)
+    std_ref = X_ALL.std(0)
+    for mean, std in results:
+        assert torch.allclose(mean, mean_ref, atol=1e-10)
+        assert torch.allclose(std, std_ref, atol=1e-10)
+
+
+def _pca_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import DistributedPCAEstimator
+    ds = ka.from_rows(_shard(rank, world, X_ALL))
+    return DistributedPCAEstimator(4).fit(ds).components
+
+
+def test_tsqr_pca_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import PCAEstimator
+    ref = PCAEstimator(4).fit(ka.from_rows(X_ALL)).components.to(torch.float64)
+    results = run_distributed(_pca_fit, 2)
+    for V in results:
+        dots = (V.to(torch.float64).T @ ref).abs().diagonal()
+        assert torch.allclose(dots, torch.ones(4, dtype=torch.float64), atol=1e-6)
+    assert torch.equal(results[0], results[1])
+
+
+def _lbfgs_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import DenseLBFGSwithL2, LeastSquaresGradient
+    ds = ka.from_rows(_shard(rank, world, X_ALL).float())
+    ys = ka.from_rows(_shard(rank, world, Y_ALL).float())
+    return DenseLBFGSwithL2(LeastSquaresGradient(), 30, 1e-3).fit(ds, ys).W
+
+
+def test_lbfgs_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import DenseLBFGSwithL2, LeastSquaresGradient
+    ref = DenseLBFGSwithL2(LeastSquaresGradient(), 30, 1e-3).fit(
+        ka.from_rows(X_ALL.float()), ka.from_rows(Y_ALL.float())).W
+    results = run_distributed(_lbfgs_fit, 2)
+    for W in results:
+        assert torch.allclose(W, ref, atol=1e-4)
+    assert torch.equal(results[0], results[1])
+
+
+def _eval_run(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.evaluation import MulticlassClassifierEvaluator
+    pred = _shard(rank, world, LAB_ALL)
+    act = _shard(rank, world, LAB_ALL.roll(1))
+    return MulticlassClassifierEvaluator(K).evaluate(
+        ka.from_rows(pred), ka.from_rows(act)).confusion
+
+
+def test_evaluator_distributed_confusion():
+    from keystone_amd.evaluation import MulticlassClassifierEvaluator
+    import keystone_amd as ka
+    ref = MulticlassClassifierEvaluator(K).evaluate(
+        ka.from_rows(LAB_ALL), ka.from_rows(LAB_ALL.roll(1))).confusion
+    results = run_distributed(_eval_run, 2)
+    for conf in results:
+        assert torch.equal(conf, ref)

@@ -1,0 +1,275 @@
+"""Solver numerics vs local oracles (ref test strategy: SURVEY.md §4 —
+every solver checked against a dense local computation, Stats.aboutEq
+style; BlockLeastSquares converges to the exact normal-equation solution)."""
+
+import math
+
+import pytest
+import torch
+
+import keystone_amd as ka
+from keystone_amd.ops.learning import (BlockLeastSquaresEstimator,
+                                       BlockWeightedLeastSquaresEstimator,
+                                       DenseLBFGSwithL2,
+                                       GaussianKernelGenerator,
+                                       KernelRidgeRegression,
+                                       KMeansPlusPlusEstimator,
+                                       LeastSquaresEstimator,
+                                       LeastSquaresGradient,
+                                       LinearMapEstimator,
+                                       LocalLeastSquaresEstimator,
+                                       LogisticGradient,
+                                       LogisticRegressionEstimator,
+                                       NaiveBayesEstimator,
+                                       SparseLBFGSwithL2)
+from keystone_amd.ops.learning.gmm import GaussianMixtureModelEstimator
+from keystone_amd.ops.learning.pca import (ApproximatePCAEstimator,
+                                           DistributedPCAEstimator,
+                                           PCAEstimator)
+from keystone_amd.ops.learning.whitening import (LinearDiscriminantAnalysis,
+                                                 ZCAWhitenerEstimator)
+
+torch.manual_seed(0)
+
+
+def ridge_oracle(X, Y, lam):
+    X64, Y64 = X.to(torch.float64), Y.to(torch.float64)
+    d = X.shape[1]
+    G = X64.T @ X64 + lam * torch.eye(d, dtype=torch.float64)
+    return torch.linalg.solve(G, X64.T @ Y64)
+
+
+def test_linear_map_estimator_exact():
+    X = torch.randn(64, 12, dtype=torch.float64)
+    W_true = torch.randn(12, 3, dtype=torch.float64)
+    Y = X @ W_true
+    model = LinearMapEstimator(lam=0.1).fit(ka.from_rows(X), ka.from_rows(Y))
+    assert torch.allclose(model.W, ridge_oracle(X, Y, 0.1), atol=1e-8)
+
+
+def test_linear_map_estimator_intercept():
+    X = torch.randn(128, 6, dtype=torch.float64)
+    W_true = torch.randn(6, 2, dtype=torch.float64)
+    Y = X @ W_true + torch.tensor([1.0, -2.0], dtype=torch.float64)
+    m = LinearMapEstimator(lam=0.0, fit_intercept=True).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    pred = m.apply_batch(X)
+    assert torch.allclose(pred, Y, atol=1e-6)
+
+
+def test_local_least_squares():
+    X = torch.randn(40, 8, dtype=torch.float64)
+    Y = torch.randn(40, 2, dtype=torch.float64)
+    m = LocalLeastSquaresEstimator(lam=0.5).fit(ka.from_rows(X), ka.from_rows(Y))
+    assert torch.allclose(m.W, ridge_oracle(X, Y, 0.5), atol=1e-7)
+
+
+def test_block_least_squares_converges_to_exact():
+    X = torch.randn(200, 24, dtype=torch.float64)
+    W_true = torch.randn(24, 4, dtype=torch.float64)
+    Y = X @ W_true + 0.01 * torch.randn(200, 4, dtype=torch.float64)
+    lam = 0.5
+    m = BlockLeastSquaresEstimator(block_size=8, num_iter=60, lam=lam,
+                                   fit_intercept=False).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    W_bcd = torch.cat(m.blocks, dim=0)
+    assert torch.allclose(W_bcd, ridge_oracle(X, Y, lam), atol=1e-5)
+
+
+def test_block_least_squares_intercept():
+    X = torch.randn(300, 16, dtype=torch.float64)
+    W_true = torch.randn(16, 2, dtype=torch.float64)
+    b_true = torch.tensor([3.0, -1.0], dtype=torch.float64)
+    Y = X @ W_true + b_true
+    m = BlockLeastSquaresEstimator(block_size=8, num_iter=50, lam=1e-8).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    pred = m.apply_batch(X)
+    assert torch.allclose(pred, Y, atol=1e-4)
+
+
+def test_block_weighted_reduces_to_plain_when_balanced():
+    # alpha -> 0 makes every weight (1-alpha)/n = uniform: same solution as
+    # unweighted BCD with rescaled lambda
+    n, d, K = 120, 12, 3
+    X = torch.randn(n, d, dtype=torch.float64)
+    y = torch.randint(0, K, (n,))
+    Y = torch.full((n, K), -1.0, dtype=torch.float64)
+    Y[torch.arange(n), y] = 1.0
+    lam = 1e-3
+    m_w = BlockWeightedLeastSquaresEstimator(
+        block_size=6, num_iter=40, lam=lam, mixture_weight=1e-12).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    W_w = torch.cat(m_w.blocks, dim=0)
+    # weighted normal equations at alpha=0: (G/n + lam I) W = X^T Y / n
+    W_ref = ridge_oracle(X, Y, lam * n)
+    assert torch.allclose(W_w, W_ref, atol=1e-5)
+
+
+def test_block_weighted_oracle_per_class():
+    # direct check against the per-column weighted normal equations
+    n, d, K = 90, 10, 3
+    X = torch.randn(n, d, dtype=torch.float64)
+    y = torch.cat([torch.zeros(60), torch.ones(20), 2 * torch.ones(10)]).long()
+    Y = torch.full((n, K), -1.0, dtype=torch.float64)
+    Y[torch.arange(n), y] = 1.0
+    alpha, lam = 0.5, 1e-2
+    m = BlockWeightedLeastSquaresEstimator(
+        block_size=d, num_iter=30, lam=lam, mixture_weight=alpha).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    W = torch.cat(m.blocks, dim=0)
+    counts = torch.bincount(y, minlength=K).to(torch.float64)
+    for c in range(K):
+        w = torch.full((n,), (1 - alpha) / n, dtype=torch.float64)
+        w[y == c] += alpha / counts[c]
+        G = X.T @ (w.unsqueeze(1) * X) + lam * torch.eye(d, dtype=torch.float64)
+        rhs = X.T @ (w * Y[:, c])
+        ref = torch.linalg.solve(G, rhs)
+        assert torch.allclose(W[:, c], ref, atol=1e-5), f"class {c}"
+
+
+def test_lbfgs_least_squares():
+    X = torch.randn(150, 10)
+    W_true = torch.randn(10, 2)
+    Y = X @ W_true
+    m = DenseLBFGSwithL2(LeastSquaresGradient(), num_iterations=80,
+                         reg_param=0.0).fit(ka.from_rows(X), ka.from_rows(Y))
+    assert torch.allclose(m.W, W_true, atol=1e-2)
+
+
+def test_logistic_regression_separable():
+    n = 200
+    X = torch.randn(n, 5)
+    w = torch.randn(5)
+    y = (X @ w > 0).long()
+    model = LogisticRegressionEstimator(num_classes=2, num_iters=60,
+                                        reg_param=1e-4).fit(
+        ka.from_rows(X), ka.from_rows(y))
+    pred = model.apply_batch(X)
+    acc = (pred == y).float().mean().item()
+    assert acc > 0.97
+
+
+def test_sparse_lbfgs_matches_dense():
+    X = (torch.rand(100, 20) < 0.3).float() * torch.randn(100, 20)
+    Y = torch.randn(100, 2)
+    dense = DenseLBFGSwithL2(LeastSquaresGradient(), 40, 1e-3).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    sparse = SparseLBFGSwithL2(LeastSquaresGradient(), 40, 1e-3).fit(
+        ka.from_rows(X.to_sparse_csr()), ka.from_rows(Y))
+    assert torch.allclose(dense.W, sparse.W, atol=1e-4)
+
+
+def test_naive_bayes_oracle():
+    X = torch.poisson(torch.rand(80, 12) * 3)
+    y = torch.randint(0, 3, (80,))
+    model = NaiveBayesEstimator(num_classes=3, lam=1.0).fit(
+        ka.from_rows(X), ka.from_rows(y))
+    # oracle
+    counts = torch.zeros(3, 12, dtype=torch.float64)
+    counts.index_add_(0, y, X.to(torch.float64))
+    lt = (counts + 1).log() - (counts.sum(1, keepdim=True) + 12).log()
+    scores = model.apply_batch(X)
+    ref = X @ lt.T.float() + (torch.bincount(y, minlength=3).float() / 80).log()
+    assert torch.allclose(scores, ref, atol=1e-4)
+
+
+def test_kmeans_recovers_clusters():
+    centers = torch.tensor([[5.0, 5.0], [-5.0, -5.0], [5.0, -5.0]])
+    X = torch.cat([centers[i] + 0.2 * torch.randn(50, 2) for i in range(3)])
+    model = KMeansPlusPlusEstimator(3, max_iterations=30, seed=1).fit(
+        ka.from_rows(X))
+    C = model.centers
+    d = torch.cdist(C.double(), centers.double())
+    assert float(d.min(dim=1).values.max()) < 0.3
+    onehot = model.apply_batch(X)
+    assert onehot.shape == (150, 3)
+    assert torch.allclose(onehot.sum(1), torch.ones(150))
+
+
+def test_gmm_recovers_mixture():
+    torch.manual_seed(3)
+    mu = torch.tensor([[4.0, 0.0], [-4.0, 0.0]])
+    X = torch.cat([mu[0] + torch.randn(300, 2) * 0.5,
+                   mu[1] + torch.randn(300, 2) * 1.0])
+    gmm = GaussianMixtureModelEstimator(2, max_iterations=60, seed=0).fit(
+        ka.from_rows(X))
+    order = gmm.means[:, 0].argsort(descending=True)
+    assert torch.allclose(gmm.means[order], mu, atol=0.2)
+    assert torch.allclose(gmm.weights.sum(), torch.tensor(1.0), atol=1e-5)
+    gamma = gmm.apply_batch(X)
+    assert torch.allclose(gamma.sum(1), torch.ones(600), atol=1e-5)
+
+
+def test_pca_matches_svd_oracle():
+    X = torch.randn(100, 8) @ torch.diag(torch.tensor([5., 4., 3., 2., 1., .5, .2, .1]))
+    pca = PCAEstimator(3).fit(ka.from_rows(X))
+    _u, _s, vh = torch.linalg.svd(X.to(torch.float64), full_matrices=False)
+    Vo = vh.T[:, :3]
+    # compare up to sign
+    dots = (pca.components.to(torch.float64).T @ Vo).abs().diagonal()
+    assert torch.allclose(dots, torch.ones(3, dtype=torch.float64), atol=1e-5)
+
+
+def test_distributed_pca_matches_local_single_rank():
+    X = torch.randn(60, 6)
+    p1 = PCAEstimator(4).fit(ka.from_rows(X))
+    p2 = DistributedPCAEstimator(4).fit(ka.from_rows(X))
+    dots = (p1.components.T.to(torch.float64)
+            @ p2.components.to(torch.float64)).abs().diagonal()
+    assert torch.allclose(dots, torch.ones(4, dtype=torch.float64), atol=1e-4)
+
+
+def test_approximate_pca_close():
+    X = torch.randn(200, 30) @ torch.randn(30, 30)
+    exact = PCAEstimator(5).fit(ka.from_rows(X))
+    approx = ApproximatePCAEstimator(5, oversample=10, power_iters=2).fit(
+        ka.from_rows(X))
+    dots = (exact.components.T.to(torch.float64)
+            @ approx.components.to(torch.float64)).abs().diagonal()
+    assert float(dots.min()) > 0.95
+
+
+def test_zca_whitening_identity_cov():
+    X = torch.randn(500, 6) @ torch.randn(6, 6)
+    zca = ZCAWhitenerEstimator(eps=1e-6).fit(ka.from_rows(X))
+    Z = zca.apply_batch(X)
+    cov = (Z.T @ Z / Z.shape[0]).to(torch.float64)
+    assert torch.allclose(cov, torch.eye(6, dtype=torch.float64), atol=0.05)
+
+
+def test_lda_separates_classes():
+    mu = torch.tensor([[3.0, 0.0, 0.0], [-3.0, 0.0, 0.0]])
+    X = torch.cat([mu[0] + torch.randn(100, 3), mu[1] + torch.randn(100, 3)])
+    y = torch.cat([torch.zeros(100), torch.ones(100)]).long()
+    lda = LinearDiscriminantAnalysis(1).fit(ka.from_rows(X), ka.from_rows(y))
+    proj = lda.apply_batch(X).squeeze(1)
+    m0, m1 = proj[:100].mean(), proj[100:].mean()
+    s = proj[:100].std() + proj[100:].std()
+    assert abs(float(m0 - m1)) / float(s) > 1.5
+
+
+def test_krr_fits_nonlinear_function():
+    torch.manual_seed(1)
+    X = torch.rand(200, 1) * 6 - 3
+    Y = torch.sin(X)
+    kern = GaussianKernelGenerator(gamma=1.0)
+    model = KernelRidgeRegression(kern, lam=1e-6, block_size=64,
+                                  num_epochs=4).fit(ka.from_rows(X), ka.from_rows(Y))
+    pred = model.apply_batch(X)
+    err = (pred - Y).abs().max().item()
+    assert err < 0.05
+
+
+def test_least_squares_dispatcher():
+    X = torch.randn(50, 8)
+    Y = torch.randn(50, 2)
+    ds, ys = ka.from_rows(X), ka.from_rows(Y)
+    est = LeastSquaresEstimator(lam=0.1)
+    chosen = est.optimize(ds, ys)
+    assert isinstance(chosen, LocalLeastSquaresEstimator)
+    est2 = LeastSquaresEstimator(lam=0.1, dense_solve_max=4)
+    chosen2 = est2.optimize(ds, ys)
+    assert isinstance(chosen2, BlockLeastSquaresEstimator)
+    chosen3 = LeastSquaresEstimator().optimize(
+        ka.from_rows(X.to_sparse_csr()), ys)
+    assert isinstance(chosen3, SparseLBFGSwithL2)
