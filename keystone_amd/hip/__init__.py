@@ -1,0 +1,186 @@
+"""keystone_amd.hip — the CDNA4 (gfx950) kernel extension.
+
+Hand-written HIP kernels for the framework's hot paths, compiled in-tree to
+``libkeystone_hip.so`` with hipcc (no torch C++ ABI — launchers are
+``extern "C"`` and take raw device pointers + the caller's HIP stream; we
+call them via ctypes with torch's current stream so they serialize
+correctly with surrounding torch ops).
+
+Kernels (ref mapping in SURVEY.md §2.9):
+  * ``gemm_fused``      — fp32 MFMA GEMM with fused bias/cos/relu epilogue
+                          (CosineRandomFeatures, Convolver epilogues)
+  * ``hellinger_l2``    — fused signed-sqrt + row L2-norm (FV normalization)
+  * ``sym_rectify``     — SymmetricRectifier feature doubling
+  * ``fisher_vector``   — fused GMM-posterior + FV accumulation (K3)
+  * ``gmm_posteriors``  — GMM E-step responsibilities (K2)
+  * ``sift``            — dense SIFT descriptor family (K1)
+
+GPU ops FAIL LOUDLY when the extension is missing on a CUDA device — there
+is deliberately no silent eager fallback on GPU (set KEYSTONE_ALLOW_EAGER=1
+to override for debugging only).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+import subprocess
+import sys
+from typing import Optional
+
+import torch
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_SRC = os.path.join(_HERE, "src")
+_LIB_PATH = os.path.join(_HERE, "libkeystone_hip.so")
+
+_lib: Optional[ctypes.CDLL] = None
+
+
+def source_files():
+    return sorted(
+        os.path.join(_SRC, f) for f in os.listdir(_SRC)
+        if f.endswith(".hip") or f.endswith(".cpp"))
+
+
+def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> str:
+    """Compile every HIP source into the in-tree shared library."""
+    srcs = source_files()
+    if not force and os.path.exists(_LIB_PATH):
+        lib_m = os.path.getmtime(_LIB_PATH)
+        if all(os.path.getmtime(s) <= lib_m for s in srcs +
+               [os.path.join(_SRC, "common.h")]):
+            return _LIB_PATH
+    hipcc = os.environ.get("HIPCC", "hipcc")
+    cmd = [hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-shared",
+           "-fPIC", "-fvisibility=hidden", "-o", _LIB_PATH] + srcs
+    if verbose:
+        print("[keystone_amd.hip] " + " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return _LIB_PATH
+
+
+def _load() -> ctypes.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB_PATH):
+        raise RuntimeError(
+            "keystone_amd HIP extension not built (libkeystone_hip.so "
+            "missing). Run keystone_amd.hip.build() — GPU ops refuse to "
+            "fall back to eager silently.")
+    lib = ctypes.CDLL(_LIB_PATH)
+    c_p, c_i, c_f = ctypes.c_void_p, ctypes.c_int, ctypes.c_float
+    lib.ks_gemm_fused_f32.argtypes = [c_p, c_p, c_p, c_p, c_i, c_i, c_i,
+                                      c_f, c_f, c_i, c_p]
+    lib.ks_gemm_fused_f32.restype = c_i
+    lib.ks_hellinger_l2_f32.argtypes = [c_p, c_p, c_i, c_i, c_f, c_p]
+    lib.ks_hellinger_l2_f32.restype = c_i
+    lib.ks_sym_rectify_f32.argtypes = [c_p, c_p, c_i, c_i, c_f, c_p]
+    lib.ks_sym_rectify_f32.restype = c_i
+    _register_extra(lib)
+    _lib = lib
+    return lib
+
+
+def _register_extra(lib: ctypes.CDLL) -> None:
+    """argtypes for kernels added after v1 (ignore if absent)."""
+    c_p, c_i, c_f = ctypes.c_void_p, ctypes.c_int, ctypes.c_float
+    try:
+        lib.ks_fisher_vector_f32.argtypes = [c_p, c_p, c_p, c_p, c_p, c_i,
+                                             c_i, c_i, c_i, c_p]
+        lib.ks_fisher_vector_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
+        lib.ks_gmm_posteriors_f32.argtypes = [c_p, c_p, c_p, c_p, c_p, c_p,
+                                              c_i, c_i, c_i, c_p]
+        lib.ks_gmm_posteriors_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
+        lib.ks_sift_f32.argtypes = [c_p, c_p, c_i, c_i, c_i, c_i, c_i, c_i,
+                                    c_p]
+        lib.ks_sift_f32.restype = c_i
+    except AttributeError:
+        pass
+
+
+def available() -> bool:
+    try:
+        _load()
+        return True
+    except (RuntimeError, OSError):
+        return False
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _check(code: int, name: str) -> None:
+    if code != 0:
+        raise RuntimeError(f"{name} failed: hipError {code}")
+
+
+def _prep(t: torch.Tensor, name: str) -> torch.Tensor:
+    if not t.is_cuda:
+        raise RuntimeError(f"{name}: expected CUDA tensor")
+    if t.dtype != torch.float32:
+        raise RuntimeError(f"{name}: expected float32, got {t.dtype}")
+    return t.contiguous()
+
+
+# --------------------------------------------------------------------- ops
+EP_NONE, EP_BIAS, EP_COS, EP_RELU = 0, 1, 2, 3
+
+
+def gemm_fused(X: torch.Tensor, W: torch.Tensor,
+               bias: Optional[torch.Tensor], epilogue: int,
+               scale: float = 1.0, alpha: float = 0.0,
+               out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    lib = _load()
+    X = _prep(X, "gemm_fused.X")
+    W = _prep(W, "gemm_fused.W")
+    M, K = X.shape
+    K2, N = W.shape
+    assert K == K2, (X.shape, W.shape)
+    if out is None:
+        out = torch.empty(M, N, dtype=torch.float32, device=X.device)
+    bptr = None
+    if bias is not None:
+        bias = _prep(bias, "gemm_fused.bias")
+        bptr = ctypes.c_void_p(bias.data_ptr())
+    _check(lib.ks_gemm_fused_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(W.data_ptr()), bptr,
+        ctypes.c_void_p(out.data_ptr()), M, K, N, scale, alpha, epilogue,
+        _stream()), "ks_gemm_fused_f32")
+    return out
+
+
+def gemm_cos(X: torch.Tensor, W: torch.Tensor, b: torch.Tensor,
+             scale: float) -> torch.Tensor:
+    """cos(X @ W + b) * scale — CosineRandomFeatures hot path."""
+    return gemm_fused(X, W, b, EP_COS, scale=scale)
+
+
+def hellinger_l2(X: torch.Tensor, eps: float = 1e-12) -> torch.Tensor:
+    lib = _load()
+    X = _prep(X, "hellinger_l2.X")
+    rows, cols = X.shape
+    out = torch.empty_like(X)
+    _check(lib.ks_hellinger_l2_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        rows, cols, eps, _stream()), "ks_hellinger_l2_f32")
+    return out
+
+
+def sym_rectify(X: torch.Tensor, alpha: float = 0.0) -> torch.Tensor:
+    lib = _load()
+    X = _prep(X, "sym_rectify.X")
+    rows, cols = X.shape
+    out = torch.empty(rows, 2 * cols, dtype=torch.float32, device=X.device)
+    _check(lib.ks_sym_rectify_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        rows, cols, alpha, _stream()), "ks_sym_rectify_f32")
+    return out

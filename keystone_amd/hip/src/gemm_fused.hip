@@ -1,0 +1,179 @@
+// Fused fp32 MFMA GEMM with elementwise epilogues — the MI355X hot path for
+// CosineRandomFeatures (out = cos(X W + b) * scale) and friends.
+//
+// Replaces the reference's Breeze/OpenBLAS GEMM + per-element cos pass
+// (ref: nodes.stats.CosineRandomFeatures): the
+// D-wide intermediate X*W never round-trips through HBM.
+//
+// Design (CDNA4 / gfx950, cdna_hip_programming.md §3/§5):
+//   * v_mfma_f32_32x32x2_f32 — exact f32 at the 157 TF f32 vector rate
+//     (no xf32 on gfx950; guide §3 "FP32-input MFMA").
+//   * 128x128 macro tile, 4 waves, each wave a 2x2 grid of 32x32 MFMA
+//     tiles; BK=32 (16 MFMA k-steps of K=2 per tile pair).
+//   * LDS: A tile [128][33] f32 (pad 1 breaks the 32-bank row stride:
+//     bank (row*33+k)%32 = (row+k)%32, conflict-free per 32-lane group),
+//     B tile [32][128] (column-consecutive reads are conflict-free).
+//   * XCD-aware block swizzle for L2 locality on the 8-XCD chip.
+// Correctness-first v1 (guide's untuned shape measured ~122 TF of the
+// 155 TF ceiling); pipelining upgrades come with rocprof evidence.
+
+#include "common.h"
+#include <math.h>
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define APAD 1
+
+enum Epilogue { EP_NONE = 0, EP_BIAS = 1, EP_COS = 2, EP_RELU = 3 };
+
+template <int EPI>
+__global__ __launch_bounds__(256)
+void gemm_fused_f32_kernel(const float* __restrict__ A,  // [M, K] row-major
+                           const float* __restrict__ B,  // [K, N] row-major
+                           const float* __restrict__ bias,  // [N] or null
+                           float* __restrict__ C,        // [M, N]
+                           int M, int K, int N, float scale, float alpha) {
+    __shared__ float As[BM][BK + APAD];
+    __shared__ float Bs[BK][BN];
+
+    const unsigned nbn = (unsigned)ks_ceil_div(N, BN);
+    const unsigned nwg = nbn * (unsigned)ks_ceil_div(M, BM);
+    unsigned wg = ks_xcd_swizzle(blockIdx.x, nwg);
+    const int brow = (int)(wg / nbn) * BM;
+    const int bcol = (int)(wg % nbn) * BN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wr = wave >> 1, wc = wave & 1;   // wave's 64x64 quadrant
+    const int l31 = lane & 31;
+    const int lk = lane >> 5;                  // 0/1: k offset inside MFMA
+
+    f32x16 acc[2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.0f);
+
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        // ---- stage A: 128 rows x 32 k, float4 per thread x 4 passes
+        {
+            const int kv = (tid & 7) * 4;        // k offset (0..28 step 4)
+            const int r0 = tid >> 3;             // row 0..31
+#pragma unroll
+            for (int p = 0; p < 4; ++p) {
+                const int row = r0 + p * 32;
+                const int gr = brow + row;
+                float4 v = {0, 0, 0, 0};
+                if (gr < M) {
+                    const int gk = k0 + kv;
+                    if (gk + 3 < K) {
+                        v = *reinterpret_cast<const float4*>(&A[(size_t)gr * K + gk]);
+                    } else {
+#pragma unroll
+                        for (int q = 0; q < 4; ++q)
+                            if (gk + q < K) ((float*)&v)[q] = A[(size_t)gr * K + gk + q];
+                    }
+                }
+                As[row][kv + 0] = v.x;
+                As[row][kv + 1] = v.y;
+                As[row][kv + 2] = v.z;
+                As[row][kv + 3] = v.w;
+            }
+        }
+        // ---- stage B: 32 k x 128 n
+        {
+            const int nv = (tid & 31) * 4;       // col offset 0..124
+            const int kr0 = tid >> 5;            // k row 0..7
+#pragma unroll
+            for (int p = 0; p < 4; ++p) {
+                const int kr = kr0 + p * 8;
+                const int gk = k0 + kr;
+                const int gn = bcol + nv;
+                float4 v = {0, 0, 0, 0};
+                if (gk < K) {
+                    if (gn + 3 < N) {
+                        v = *reinterpret_cast<const float4*>(&B[(size_t)gk * N + gn]);
+                    } else {
+#pragma unroll
+                        for (int q = 0; q < 4; ++q)
+                            if (gn + q < N) ((float*)&v)[q] = B[(size_t)gk * N + gn + q];
+                    }
+                }
+                *reinterpret_cast<float4*>(&Bs[kr][nv]) = v;
+            }
+        }
+        __syncthreads();
+
+        // ---- MFMA inner loop: 16 k-steps of K=2
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += 2) {
+            float a0 = As[wr * 64 + 0 + l31][kk + lk];
+            float a1 = As[wr * 64 + 32 + l31][kk + lk];
+            float b0 = Bs[kk + lk][wc * 64 + 0 + l31];
+            float b1 = Bs[kk + lk][wc * 64 + 32 + l31];
+            acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+            acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+            acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+            acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: C/D layout col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
+#pragma unroll
+    for (int ti = 0; ti < 2; ++ti) {
+#pragma unroll
+        for (int tj = 0; tj < 2; ++tj) {
+            const int col = bcol + wc * 64 + tj * 32 + l31;
+            if (col >= N) continue;
+            const float bv = (EPI != EP_NONE && bias != nullptr) ? bias[col] : 0.0f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = brow + wr * 64 + ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                if (row >= M) continue;
+                float v = acc[ti][tj][r] + bv;
+                if (EPI == EP_COS) v = __cosf(v) * scale;
+                else if (EPI == EP_RELU) v = fmaxf(v - alpha, 0.0f) * scale;
+                C[(size_t)row * N + col] = v;
+            }
+        }
+    }
+}
+
+static int launch_gemm_fused(const float* A, const float* B, const float* bias,
+                             float* C, int M, int K, int N, float scale,
+                             float alpha, int epilogue, hipStream_t stream) {
+    const int nwg = ks_ceil_div(M, BM) * ks_ceil_div(N, BN);
+    dim3 grid(nwg), block(256);
+    switch (epilogue) {
+        case EP_NONE:
+            hipLaunchKernelGGL(gemm_fused_f32_kernel<EP_NONE>, grid, block, 0,
+                               stream, A, B, bias, C, M, K, N, scale, alpha);
+            break;
+        case EP_BIAS:
+            hipLaunchKernelGGL(gemm_fused_f32_kernel<EP_BIAS>, grid, block, 0,
+                               stream, A, B, bias, C, M, K, N, scale, alpha);
+            break;
+        case EP_COS:
+            hipLaunchKernelGGL(gemm_fused_f32_kernel<EP_COS>, grid, block, 0,
+                               stream, A, B, bias, C, M, K, N, scale, alpha);
+            break;
+        case EP_RELU:
+            hipLaunchKernelGGL(gemm_fused_f32_kernel<EP_RELU>, grid, block, 0,
+                               stream, A, B, bias, C, M, K, N, scale, alpha);
+            break;
+        default:
+            return (int)hipErrorInvalidValue;
+    }
+    return (int)hipGetLastError();
+}
+
+KS_EXPORT int ks_gemm_fused_f32(const float* A, const float* B,
+                                const float* bias, float* C, int M, int K,
+                                int N, float scale, float alpha, int epilogue,
+                                hipStream_t stream) {
+    return launch_gemm_fused(A, B, bias, C, M, K, N, scale, alpha, epilogue,
+                             stream);
+}

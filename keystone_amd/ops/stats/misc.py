@@ -67,6 +67,24 @@ class NormalizeRows(ElementwiseTransformer):
         return X / X.norm(dim=-1, keepdim=True).clamp_min(self.eps)
 
 
+# ---- fused GPU fast path: SignedHellinger -> NormalizeRows is ONE HIP
+# kernel (2 HBM passes instead of ~6; see hip/src/elementwise.hip)
+def _hellinger_l2_gpu(parts, X: torch.Tensor) -> torch.Tensor:
+    from ... import hip as hip_ops
+    if X.dtype == torch.float32:
+        return hip_ops.hellinger_l2(X, parts[1].eps)
+    out = torch.sign(X) * torch.sqrt(torch.abs(X))
+    return out / out.norm(dim=-1, keepdim=True).clamp_min(parts[1].eps)
+
+
+def _register_fused():
+    from ...workflow.optimizer import FUSED_GPU_PATTERNS
+    FUSED_GPU_PATTERNS[(SignedHellingerMapper, NormalizeRows)] = _hellinger_l2_gpu
+
+
+_register_fused()
+
+
 class Sampler(Transformer):
     """Sample up to ``size`` items of the local shard (ref: Sampler)."""
 

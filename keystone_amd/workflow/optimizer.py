@@ -117,6 +117,12 @@ class ElementwiseTransformer(Transformer):
     consecutive ones can run in one pass over the shard."""
 
 
+#: registry of fused-GPU fast paths: tuple of transformer types -> callable
+#: (parts, X) -> tensor.  Populated by ops modules (e.g. SignedHellinger +
+#: NormalizeRows -> one hellinger_l2 HIP kernel).
+FUSED_GPU_PATTERNS: Dict[tuple, Callable] = {}
+
+
 class FusedElementwise(ElementwiseTransformer):
     """A chain of elementwise transformers executed as one node."""
 
@@ -128,6 +134,11 @@ class FusedElementwise(ElementwiseTransformer):
         return "+".join(type(p).__name__ for p in self.parts)
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        if X.is_cuda:
+            key = tuple(type(p) for p in self.parts)
+            fn = FUSED_GPU_PATTERNS.get(key)
+            if fn is not None:
+                return fn(self.parts, X)
         for p in self.parts:
             X = p.apply_batch(X)
         return X

@@ -1,0 +1,119 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (GPU only).
+
+Per the test contract every HIP kernel is compared against a plain torch
+fp32 implementation of the same op (the CPU oracle pattern of SURVEY.md §4,
+run on-device)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def hip():
+    from keystone_amd import hip as h
+    assert h.available(), "HIP extension must be built and loadable"
+    return h
+
+
+def test_gemm_fused_matches_matmul(hip):
+    torch.manual_seed(0)
+    for M, K, N in [(128, 64, 128), (300, 177, 530), (1024, 512, 2048),
+                    (64, 32, 64), (129, 33, 257)]:
+        X = torch.randn(M, K, device="cuda")
+        W = torch.randn(K, N, device="cuda")
+        out = hip.gemm_fused(X, W, None, hip.EP_NONE)
+        ref = X @ W
+        # f32 MFMA is an exact fmaf chain — tolerance is fp32 roundoff class
+        err = (out - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err <= 1e-4 * max(scale, 1.0), (M, K, N, err)
+
+
+def test_gemm_fused_transpose_detecting(hip):
+    # asymmetric operands (guide §3: symmetric B passes a transposed C-write)
+    M, K, N = 96, 64, 160
+    X = torch.arange(M * K, device="cuda", dtype=torch.float32).reshape(M, K) / (M * K)
+    W = (torch.arange(K * N, device="cuda", dtype=torch.float32).reshape(K, N)
+         % 17 - 8) / 8
+    out = hip.gemm_fused(X, W, None, hip.EP_NONE)
+    ref = X @ W
+    assert torch.allclose(out, ref, atol=1e-3, rtol=1e-4)
+
+
+def test_gemm_cos_matches_torch(hip):
+    torch.manual_seed(1)
+    M, K, N = 513, 120, 777
+    X = torch.randn(M, K, device="cuda")
+    W = torch.randn(K, N, device="cuda") * 0.3
+    b = torch.rand(N, device="cuda") * (2 * math.pi)
+    scale = 0.123
+    out = hip.gemm_cos(X, W, b, scale)
+    ref = torch.cos(X @ W + b) * scale
+    assert (out - ref).abs().max().item() < 1e-3
+
+
+def test_gemm_bias(hip):
+    M, K, N = 128, 96, 256
+    X = torch.randn(M, K, device="cuda")
+    W = torch.randn(K, N, device="cuda")
+    b = torch.randn(N, device="cuda")
+    out = hip.gemm_fused(X, W, b, hip.EP_BIAS)
+    ref = X @ W + b
+    assert torch.allclose(out, ref, atol=1e-3, rtol=1e-4)
+
+
+def test_hellinger_l2_matches_torch(hip):
+    torch.manual_seed(2)
+    for rows, cols in [(8, 1000), (32, 4096), (5, 33)]:
+        X = torch.randn(rows, cols, device="cuda") * 3
+        out = hip.hellinger_l2(X)
+        y = torch.sign(X) * torch.sqrt(torch.abs(X))
+        ref = y / y.norm(dim=1, keepdim=True).clamp_min(1e-6)
+        assert (out - ref).abs().max().item() < 1e-5, (rows, cols)
+
+
+def test_sym_rectify_matches_torch(hip):
+    X = torch.randn(64, 500, device="cuda")
+    out = hip.sym_rectify(X, alpha=0.25)
+    ref = torch.cat([torch.clamp(X - 0.25, min=0.0),
+                     torch.clamp(-X - 0.25, min=0.0)], dim=1)
+    assert torch.equal(out, ref)
+
+
+def test_cosine_random_features_uses_hip_on_gpu(hip):
+    import keystone_amd as ka
+    from keystone_amd.ops.stats import CosineRandomFeatures
+    crf = CosineRandomFeatures(48, 512, gamma=0.5, seed=3)
+    X = torch.randn(256, 48, device="cuda")
+    out = crf.apply_batch(X)
+    W = crf.W.to("cuda")
+    b = crf.b.to("cuda")
+    ref = torch.cos(X @ W + b) * crf.scale
+    assert (out - ref).abs().max().item() < 1e-3
+
+
+def test_pipeline_end_to_end_gpu():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    from keystone_amd.ops.stats import CosineRandomFeatures, StandardScaler
+    from keystone_amd.ops.util import (ClassLabelIndicatorsFromIntLabels,
+                                       MaxClassifier)
+    torch.manual_seed(0)
+    n, d, D, K = 1024, 32, 512, 5
+    X = torch.randn(n, d, device="cuda")
+    y = torch.randint(0, K, (n,), device="cuda")
+    ds = ka.from_rows(X)
+    labels = ka.from_rows(
+        ClassLabelIndicatorsFromIntLabels(K).apply_batch(y).float())
+    pipe = (StandardScaler().with_data(ds)
+            .and_then(CosineRandomFeatures(d, D, 1.0))
+            .and_then(BlockLeastSquaresEstimator(block_size=128, num_iter=3,
+                                                 lam=1e-4), ds, labels)
+            .and_then(MaxClassifier()))
+    pred = pipe.apply(ds).get().data
+    acc = (pred == y).float().mean().item()
+    assert acc > 0.6
