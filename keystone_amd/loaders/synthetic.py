@@ -1,0 +1,112 @@
+"""Synthetic dataset generators for the five benchmark configs.
+
+BASELINE.json mandates synthetic data / random-init weights (no network
+for datasets), so every driver config has a generator here producing the
+same shapes as the real loaders:
+
+  1. 20-Newsgroups   -> random documents over a Zipf vocabulary, 20 classes
+  2. MNIST           -> [n, 784] fp32 rows, 10 classes
+  3. TIMIT           -> [n, 440] MFCC+context frames, 147 phone classes
+  4. Amazon reviews  -> longer documents, binary labels
+  5. ImageNet        -> 256x256 grayscale-ish images, 1000 classes
+
+All generators shard deterministically by rank: rank r generates rows
+[lo, hi) of the same global sequence, so N-GPU runs see the same global
+dataset as 1-GPU runs.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..parallel.dataset import ItemDataset, TensorDataset, from_items, from_rows
+from ..parallel.runtime import get_runtime
+
+_WORDS = None
+
+
+def _vocab(size: int) -> List[str]:
+    global _WORDS
+    if _WORDS is None or len(_WORDS) < size:
+        _WORDS = [f"w{i:05d}" for i in range(size)]
+    return _WORDS[:size]
+
+
+def synthetic_documents(n_docs: int, num_classes: int = 20,
+                        vocab_size: int = 5000, doc_len: int = 80,
+                        seed: int = 0, runtime=None
+                        ) -> Tuple[ItemDataset, TensorDataset]:
+    """Documents with class-dependent Zipf word distributions (so linear
+    models genuinely separate them) + int labels."""
+    rt = runtime or get_runtime()
+    lo, hi = rt.shard_bounds(n_docs)
+    words = _vocab(vocab_size)
+    # Zipf base distribution + per-class boosted sub-vocabulary
+    ranks = torch.arange(1, vocab_size + 1, dtype=torch.float64)
+    base = 1.0 / ranks
+    docs, labels = [], []
+    per_class = max(10, vocab_size // (4 * num_classes))
+    for i in range(lo, hi):
+        g = torch.Generator().manual_seed(seed * 1_000_003 + i)
+        c = int(torch.randint(0, num_classes, (1,), generator=g).item())
+        probs = base.clone()
+        probs[c * per_class % vocab_size:(c * per_class % vocab_size) + per_class] *= 40.0
+        probs /= probs.sum()
+        idx = torch.multinomial(probs, doc_len, replacement=True, generator=g)
+        docs.append(" ".join(words[j] for j in idx))
+        labels.append(c)
+    return (from_items(docs, rt),
+            from_rows(torch.tensor(labels, dtype=torch.long), rt))
+
+
+def synthetic_dense(n: int, d: int, num_classes: int, seed: int = 0,
+                    runtime=None, noise: float = 0.5,
+                    device=None) -> Tuple[TensorDataset, TensorDataset]:
+    """Gaussian class-cluster rows (MNIST/TIMIT shapes): class means drawn
+    once, rows = mean + noise."""
+    rt = runtime or get_runtime()
+    dev = device or rt.device
+    g0 = torch.Generator().manual_seed(seed)
+    means = torch.randn(num_classes, d, generator=g0)
+    lo, hi = rt.shard_bounds(n)
+    n_local = hi - lo
+    g = torch.Generator().manual_seed(seed + 17 + rt.rank)
+    labels = torch.randint(0, num_classes, (n_local,), generator=g)
+    X = means[labels] + noise * torch.randn(n_local, d, generator=g)
+    return (from_rows(X.to(dev), rt), from_rows(labels.to(dev), rt))
+
+
+def synthetic_mnist(n: int = 60000, seed: int = 0, runtime=None, device=None):
+    return synthetic_dense(n, 784, 10, seed, runtime, device=device)
+
+
+def synthetic_timit(n: int = 100000, context: int = 20, seed: int = 0,
+                    runtime=None, device=None):
+    """TIMIT MFCC frames with +-context splicing: 440 dims, 147 classes
+    (ref: TimitFeaturesDataLoader shapes)."""
+    d = 440
+    return synthetic_dense(n, d, 147, seed, runtime, device=device)
+
+
+def synthetic_images(n: int, h: int = 256, w: int = 256, num_classes: int = 1000,
+                     seed: int = 0, runtime=None, device=None,
+                     batched: bool = True):
+    """ImageNet-shaped synthetic grayscale images in [0,1]: smooth random
+    blobs (low-frequency noise) so SIFT-like featurizers see structure.
+    Returns ([n_local, h, w] TensorDataset, int labels)."""
+    rt = runtime or get_runtime()
+    dev = device or rt.device
+    lo, hi = rt.shard_bounds(n)
+    n_local = hi - lo
+    g = torch.Generator().manual_seed(seed + 23 + rt.rank)
+    # low-res noise upsampled = smooth structure, plus fine detail
+    coarse = torch.rand(n_local, 1, h // 8, w // 8, generator=g)
+    imgs = torch.nn.functional.interpolate(coarse, size=(h, w),
+                                           mode="bilinear",
+                                           align_corners=False).squeeze(1)
+    imgs = 0.8 * imgs + 0.2 * torch.rand(n_local, h, w, generator=g)
+    labels = torch.randint(0, num_classes, (n_local,), generator=g)
+    return (from_rows(imgs.to(dev), rt), from_rows(labels.to(dev), rt))
