@@ -1,0 +1,150 @@
+"""Dense multi-scale SIFT descriptors.
+
+ref: nodes.images.external.SIFTExtractor -> JNI
+``VLFeat.getSIFTs(step, binSize, numScales, img)`` (VLFeat dsift):
+Gaussian smoothing per scale, per-pixel gradient magnitude/orientation,
+soft orientation binning into 8 bins, triangular (bilinear) spatial
+pooling into a 4x4 grid of bins -> 128-dim descriptors on a dense grid,
+concatenated over ``num_scales`` scales with bin size doubling.
+
+This implementation is *batched over the whole shard*: every step is a
+tensor op ([B, H, W] in, [B, n_desc, 128] out), so the GPU path runs a
+fixed chain of MIOpen/rocBLAS/elementwise kernels per shard.  A fused
+single-kernel HIP implementation (per-image workgroups, LDS histograms)
+is the K1 upgrade; this version doubles as its numerics oracle.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from ...parallel.dataset import Dataset, TensorDataset
+from ...workflow.pipeline import Transformer
+
+N_ORI = 8
+N_SPATIAL = 4  # 4x4 spatial bins -> 128-dim
+
+
+def _gaussian_kernel1d(sigma: float, device, dtype) -> torch.Tensor:
+    radius = max(1, int(math.ceil(3.0 * sigma)))
+    x = torch.arange(-radius, radius + 1, device=device, dtype=dtype)
+    k = torch.exp(-0.5 * (x / sigma) ** 2)
+    return k / k.sum()
+
+
+def _smooth(imgs: torch.Tensor, sigma: float) -> torch.Tensor:
+    """Separable Gaussian blur on [B, H, W]."""
+    if sigma <= 0:
+        return imgs
+    k = _gaussian_kernel1d(sigma, imgs.device, imgs.dtype)
+    r = (k.numel() - 1) // 2
+    x = imgs.unsqueeze(1)
+    x = F.conv2d(F.pad(x, (r, r, 0, 0), mode="replicate"), k.reshape(1, 1, 1, -1))
+    x = F.conv2d(F.pad(x, (0, 0, r, r), mode="replicate"), k.reshape(1, 1, -1, 1))
+    return x.squeeze(1)
+
+
+def _triangle_kernel1d(bin_size: int, device, dtype) -> torch.Tensor:
+    """VLFeat dsift bilinear spatial window: support 2*bin_size-1."""
+    x = torch.arange(-(bin_size - 1), bin_size, device=device, dtype=dtype)
+    return 1.0 - x.abs() / bin_size
+
+
+class SIFTExtractor(Transformer):
+    def __init__(self, step: int = 4, bin_size: int = 4, num_scales: int = 2,
+                 scale_step: float = math.sqrt(2.0), clamp: float = 0.2,
+                 append_xy: bool = False):
+        self.step = step
+        self.bin_size = bin_size
+        self.num_scales = num_scales
+        self.scale_step = scale_step
+        self.clamp = clamp
+        self.append_xy = append_xy
+
+    # ------------------------------------------------------------------
+    def _one_scale(self, imgs: torch.Tensor, bin_size: int, sigma: float
+                   ) -> torch.Tensor:
+        """[B, H, W] -> [B, n, 128] descriptors at one scale."""
+        B, H, W = imgs.shape
+        dt = imgs.dtype
+        sm = _smooth(imgs, sigma)
+
+        # central-difference gradients
+        gx = torch.zeros_like(sm)
+        gy = torch.zeros_like(sm)
+        gx[:, :, 1:-1] = 0.5 * (sm[:, :, 2:] - sm[:, :, :-2])
+        gy[:, 1:-1, :] = 0.5 * (sm[:, 2:, :] - sm[:, :-2, :])
+        mag = torch.sqrt(gx * gx + gy * gy)
+        ang = torch.atan2(gy, gx) % (2 * math.pi)
+
+        # soft orientation binning -> [B, 8, H, W]
+        t = ang * (N_ORI / (2 * math.pi))
+        b0 = torch.floor(t)
+        frac = t - b0
+        b0 = b0.long() % N_ORI
+        b1 = (b0 + 1) % N_ORI
+        omap = torch.zeros(B, N_ORI, H, W, device=imgs.device, dtype=dt)
+        omap.scatter_add_(1, b0.unsqueeze(1), (mag * (1 - frac)).unsqueeze(1))
+        omap.scatter_add_(1, b1.unsqueeze(1), (mag * frac).unsqueeze(1))
+
+        # separable triangular spatial pooling per orientation channel
+        k = _triangle_kernel1d(bin_size, imgs.device, dt)
+        r = bin_size - 1
+        x = omap.reshape(B * N_ORI, 1, H, W)
+        x = F.conv2d(F.pad(x, (r, r, 0, 0)), k.reshape(1, 1, 1, -1))
+        x = F.conv2d(F.pad(x, (0, 0, r, r)), k.reshape(1, 1, -1, 1))
+        S = x.reshape(B, N_ORI, H, W)
+
+        # sample spatial-bin centers on the dense grid: descriptor anchored
+        # at top-left bin center; bins at offsets j*bin_size
+        span = (N_SPATIAL - 1) * bin_size
+        ys = torch.arange(bin_size // 2, H - span - bin_size // 2, self.step,
+                          device=imgs.device)
+        xs = torch.arange(bin_size // 2, W - span - bin_size // 2, self.step,
+                          device=imgs.device)
+        if len(ys) == 0 or len(xs) == 0:
+            return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
+                               device=imgs.device, dtype=dt)
+        parts = []
+        for j in range(N_SPATIAL):
+            for i in range(N_SPATIAL):
+                sub = S[:, :, ys + j * bin_size][:, :, :, xs + i * bin_size]
+                parts.append(sub)  # [B, 8, ny, nx]
+        D = torch.stack(parts, dim=2)  # [B, 8, 16, ny, nx]
+        D = D.permute(0, 3, 4, 2, 1).reshape(B, len(ys) * len(xs),
+                                             N_SPATIAL * N_SPATIAL * N_ORI)
+
+        # SIFT normalization: L2 -> clamp 0.2 -> L2
+        D = D / D.norm(dim=2, keepdim=True).clamp_min(1e-10)
+        D = D.clamp_max(self.clamp)
+        D = D / D.norm(dim=2, keepdim=True).clamp_min(1e-10)
+
+        if self.append_xy:
+            yy = (ys.to(dt) / H - 0.5).repeat_interleave(len(xs))
+            xx = (xs.to(dt) / W - 0.5).repeat(len(ys))
+            pos = torch.stack([xx, yy], dim=1).unsqueeze(0).expand(B, -1, -1)
+            D = torch.cat([D, pos], dim=2)
+        return D
+
+    def apply_batch(self, imgs: torch.Tensor) -> torch.Tensor:
+        """[B, H, W] -> [B, n_total, 128(+2)] over all scales."""
+        if imgs.dim() == 2:
+            return self.apply_batch(imgs.unsqueeze(0)).squeeze(0)
+        outs = []
+        for s in range(self.num_scales):
+            bs = max(1, int(round(self.bin_size * (self.scale_step ** s))))
+            sigma = bs / 3.0  # VLFeat magnif=3 convention
+            outs.append(self._one_scale(imgs, bs, sigma))
+        return torch.cat(outs, dim=1)
+
+    def apply(self, img: torch.Tensor) -> torch.Tensor:
+        return self.apply_batch(img.unsqueeze(0)).squeeze(0)
+
+    def apply_dataset(self, ds: Dataset) -> Dataset:
+        if isinstance(ds, TensorDataset):
+            return ds.map_batch(self.apply_batch)
+        return ds.map(self.apply)
