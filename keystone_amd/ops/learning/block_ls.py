@@ -182,11 +182,17 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
     """
 
     def __init__(self, block_size: int = 4096, num_iter: int = 10,
-                 lam: float = 0.0, mixture_weight: float = 0.5):
+                 lam: float = 0.0, mixture_weight: float = 0.5,
+                 woodbury_threshold: float = 0.25):
         self.block_size = block_size
         self.num_iter = num_iter
         self.lam = lam
         self.mixture_weight = mixture_weight
+        # per-class correction rank is n_c; below threshold*d_b solve via
+        # one base Cholesky + rank-n_c Woodbury update (O(d^2 n_c)) instead
+        # of a fresh per-class d^3 factorization — at ImageNet scale
+        # (K=1000, n_c of a few) this is a ~100x solver-flop reduction.
+        self.woodbury_threshold = woodbury_threshold
 
     def _fit(self, data: Dataset, labels: TensorDataset) -> BlockLinearMapper:
         getters, rt, n_local = _as_block_getters(data, self.block_size)
@@ -243,24 +249,59 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
 
                 Wb = torch.zeros(db, K, dtype=dtype, device=Xb.device)
                 if own:
-                    # batched per-class weighted systems, all rank-local
-                    grams, rhss = [], []
-                    for c in own:
-                        m = own_masks[c]
-                        Xc = Xb[m]
-                        wc = alpha / counts[c].item()
-                        G = base_w * gram_all + wc * (Xc.T @ Xc)
-                        r = base_w * xtr_all[:, c] + wc * (Xc.T @ T[m, c])
-                        grams.append(G)
-                        rhss.append(r)
-                    Gb = torch.stack(grams).to(torch.float64)
-                    rb = torch.stack(rhss).to(torch.float64).unsqueeze(-1)
-                    Gb += self.lam * torch.eye(db, dtype=torch.float64,
-                                               device=Gb.device)
-                    L = torch.linalg.cholesky(Gb)
-                    sol = torch.cholesky_solve(rb, L).squeeze(-1).to(dtype)
+                    # ONE base factorization (shared across classes), then a
+                    # rank-n_c Woodbury update per class:
+                    #   (B + U U^T)^{-1} r = B^{-1}r - B^{-1}U (I + U^T B^{-1}U)^{-1} U^T B^{-1} r
+                    # with B = base_w*G_all + lam I and U = sqrt(w_c) X_c^T.
+                    # Classes with equal n_c are solved as ONE batched
+                    # update (K=1000 at ImageNet scale -> a handful of
+                    # batched rocSOLVER/GEMM calls, not 1000 Python steps).
+                    B64 = base_w * gram_all.to(torch.float64)
+                    B64 += self.lam * torch.eye(db, dtype=torch.float64,
+                                                device=Xb.device)
+                    L = torch.linalg.cholesky(B64)
+                    T64 = T.to(torch.float64)
+                    Xb64 = Xb.to(torch.float64)
+                    rhs = torch.empty(db, len(own), dtype=torch.float64,
+                                      device=Xb.device)
+                    wcs = {c: alpha / counts[c].item() for c in own}
                     for j, c in enumerate(own):
-                        Wb[:, c] = sol[j]
+                        m = own_masks[c]
+                        rhs[:, j] = base_w * xtr_all[:, c].to(torch.float64) \
+                            + wcs[c] * (Xb64[m].T @ T64[m, c])
+                    RB = torch.cholesky_solve(rhs, L)        # B^{-1} rhs
+                    col = {c: j for j, c in enumerate(own)}
+
+                    by_nc: dict = {}
+                    for c in own:
+                        by_nc.setdefault(int(own_masks[c].sum()), []).append(c)
+                    for n_c, cls_list in by_nc.items():
+                        js = [col[c] for c in cls_list]
+                        if n_c == 0:
+                            for c, j in zip(cls_list, js):
+                                Wb[:, c] = RB[:, j].to(dtype)
+                            continue
+                        if n_c <= self.woodbury_threshold * db:
+                            U = torch.stack([
+                                (wcs[c] ** 0.5) * Xb64[own_masks[c]].T
+                                for c in cls_list])              # [G, db, nc]
+                            UB = torch.cholesky_solve(U, L.unsqueeze(0))
+                            S = torch.eye(n_c, dtype=torch.float64,
+                                          device=Xb.device) \
+                                + U.transpose(1, 2) @ UB         # [G, nc, nc]
+                            r = RB[:, js].T.unsqueeze(2)         # [G, db, 1]
+                            t = U.transpose(1, 2) @ r            # [G, nc, 1]
+                            corr = (UB @ torch.linalg.solve(S, t)).squeeze(2)
+                            sol = (RB[:, js].T - corr).to(dtype)  # [G, db]
+                            for g, c in enumerate(cls_list):
+                                Wb[:, c] = sol[g]
+                        else:
+                            for c, j in zip(cls_list, js):
+                                G = B64 + wcs[c] * (
+                                    Xb64[own_masks[c]].T @ Xb64[own_masks[c]])
+                                Lc = torch.linalg.cholesky(G)
+                                Wb[:, c] = torch.cholesky_solve(
+                                    rhs[:, j:j + 1], Lc).squeeze(1).to(dtype)
                 if rt.is_distributed:
                     rt.all_reduce_(Wb)   # disjoint columns -> sum == gather
                 R = T - Xb @ Wb

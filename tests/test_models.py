@@ -1,0 +1,99 @@
+"""End-to-end pipeline tests for the five driver configs (small CPU sizes).
+ref: SURVEY.md §4 — the driver configs are the integration benchmarks."""
+
+import pytest
+import torch
+
+import keystone_amd as ka
+from keystone_amd.evaluation import (BinaryClassifierEvaluator,
+                                     MeanAveragePrecisionEvaluator,
+                                     MulticlassClassifierEvaluator)
+from keystone_amd.loaders import (synthetic_dense, synthetic_documents,
+                                  synthetic_images)
+from keystone_amd.models import (AmazonConfig, CifarConfig, ImageNetFVConfig,
+                                 MnistConfig, TimitConfig, VOCConfig,
+                                 build_amazon, build_mnist_randomfft,
+                                 build_newsgroups, build_timit,
+                                 fit_cifar_patches, fit_imagenet_fv, fit_voc,
+                                 NewsgroupsConfig)
+
+torch.manual_seed(0)
+
+
+def test_config1_newsgroups():
+    docs, labels = synthetic_documents(150, num_classes=4, vocab_size=400,
+                                       doc_len=30, seed=2)
+    pipe = build_newsgroups(docs, labels, NewsgroupsConfig(
+        num_classes=4, ngrams=2, common_features=400))
+    pred = pipe.apply(docs).get()
+    m = MulticlassClassifierEvaluator(4).evaluate(pred, labels)
+    assert m.total_accuracy > 0.9
+
+
+def test_config2_mnist_randomfft():
+    train, labels = synthetic_dense(300, 64, 10, seed=3, noise=0.8)
+    pipe = build_mnist_randomfft(train, labels, MnistConfig(
+        input_dim=64, num_ffts=2, num_classes=10, block_size=32, num_iter=3))
+    pred = pipe.apply(train).get()
+    m = MulticlassClassifierEvaluator(10).evaluate(pred, labels)
+    assert m.total_accuracy > 0.9
+
+
+def test_config3_timit():
+    train, labels = synthetic_dense(400, 80, 12, seed=4, noise=1.0)
+    pipe = build_timit(train, labels, TimitConfig(
+        input_dim=80, num_features=512, gamma=0.02, num_classes=12,
+        block_size=128, num_iter=3, lam=1e-4))
+    pred = pipe.apply(train).get()
+    m = MulticlassClassifierEvaluator(12).evaluate(pred, labels)
+    assert m.total_accuracy > 0.85
+
+
+def test_config4_amazon():
+    docs, labels = synthetic_documents(200, num_classes=2, vocab_size=300,
+                                       doc_len=40, seed=5)
+    pipe = build_amazon(docs, labels, AmazonConfig(
+        ngrams=2, common_features=300, num_iters=15))
+    pred = pipe.apply(docs).get()
+    m = BinaryClassifierEvaluator().evaluate(pred, labels)
+    assert m.accuracy > 0.9
+
+
+def test_config5_imagenet_fv_small():
+    imgs, labels = synthetic_images(12, h=48, w=48, num_classes=4, seed=6)
+    conf = ImageNetFVConfig(num_classes=4, sift_step=6, sift_bin=4,
+                            sift_scales=1, lcs_step=6, lcs_subpatch=4,
+                            pca_dims=16, gmm_k=4, n_sample=2000,
+                            block_size=128, solver_epochs=2, lam=1e-4,
+                            top_k=1)
+    pipe, info = fit_imagenet_fv(imgs, labels, conf)
+    assert info["feature_dim"] == 2 * 4 * 16 * 2  # 2KD per branch, 2 branches
+    pred = pipe.apply(imgs).get().data.squeeze(1)
+    # tiny synthetic set: the solver should at least fit its own train set
+    acc = (pred == labels.data).float().mean().item()
+    assert acc > 0.5
+
+
+def test_cifar_patches_pipeline():
+    imgs = ka.from_rows(torch.rand(40, 3, 16, 16))
+    labels = ka.from_rows(torch.randint(0, 3, (40,)))
+    pipe = fit_cifar_patches(imgs, labels, CifarConfig(
+        num_filters=16, patch_size=4, patch_sample=500, pool_size=6,
+        pool_stride=6, lam=1.0, num_classes=3))
+    pred = pipe.apply(imgs).get()
+    m = MulticlassClassifierEvaluator(3).evaluate(pred, labels)
+    assert m.total_accuracy > 0.5  # memorize small train set
+
+
+def test_voc_map_pipeline():
+    imgs, _ = synthetic_images(10, h=40, w=40, num_classes=3, seed=7)
+    Y = (torch.rand(imgs.data.shape[0], 3) > 0.5).float()
+    Y[:, 0] = 1.0  # ensure a positive class
+    labels = ka.from_rows(Y)
+    pipe = fit_voc(imgs, labels, VOCConfig(
+        num_classes=3, sift_step=6, sift_bin=4, sift_scales=1, pca_dims=8,
+        gmm_k=2, n_sample=1000, block_size=16, solver_epochs=2))
+    scores = pipe.apply(imgs).get()
+    res = MeanAveragePrecisionEvaluator(3).evaluate(scores, labels)
+    assert 0.0 <= res["mAP"] <= 1.0
+    assert res["mAP"] > 0.5  # train-set ranking should be decent

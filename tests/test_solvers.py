@@ -273,3 +273,26 @@ def test_least_squares_dispatcher():
     chosen3 = LeastSquaresEstimator().optimize(
         ka.from_rows(X.to_sparse_csr()), ys)
     assert isinstance(chosen3, SparseLBFGSwithL2)
+
+
+def test_block_weighted_woodbury_path_matches_direct():
+    # n_c << d_b forces the Woodbury rank-update path; compare against the
+    # per-column weighted normal equations oracle
+    torch.manual_seed(7)
+    n, d, K = 40, 32, 8          # n_c ~ 5 rows << d_b = 32
+    X = torch.randn(n, d, dtype=torch.float64)
+    y = torch.arange(n) % K
+    Y = torch.full((n, K), -1.0, dtype=torch.float64)
+    Y[torch.arange(n), y] = 1.0
+    alpha, lam = 0.4, 1e-3
+    m = BlockWeightedLeastSquaresEstimator(
+        block_size=d, num_iter=25, lam=lam, mixture_weight=alpha,
+        woodbury_threshold=0.5).fit(ka.from_rows(X), ka.from_rows(Y))
+    W = torch.cat(m.blocks, dim=0)
+    counts = torch.bincount(y, minlength=K).to(torch.float64)
+    for c in range(K):
+        w = torch.full((n,), (1 - alpha) / n, dtype=torch.float64)
+        w[y == c] += alpha / counts[c]
+        G = X.T @ (w.unsqueeze(1) * X) + lam * torch.eye(d, dtype=torch.float64)
+        ref = torch.linalg.solve(G, X.T @ (w * Y[:, c]))
+        assert torch.allclose(W[:, c], ref, atol=1e-6), f"class {c}"
