@@ -1,0 +1,135 @@
+"""Flagship benchmark: ImageNet SIFT+LCS Fisher-Vector pipeline fit.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W`` (for
+N>1 the driver launches via torch.distributed.run, one rank per GPU over
+RCCL).  One *step* = one complete pipeline fit on a fixed per-GPU batch of
+synthetic ImageNet-shaped images (weak scaling): SIFT+LCS dense
+featurization -> distributed PCA (TSQR) -> GMM (distributed EM) -> Fisher
+Vector encoding -> signed-sqrt + L2 -> BlockWeightedLeastSquares (Gram
+all-reduce + class-sharded Woodbury solves).  The metric is the whole-job
+images/second of pipeline fitting (BASELINE.json: "pipeline fit time (s) +
+images/sec, ImageNet FV pipeline").
+
+Rank 0 prints ONE JSON line; the timed region is bracketed by
+barrier + torch.cuda.synchronize on both sides and the per-step time is
+the MAX over ranks.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=2)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--images-per-gpu", type=int, default=256)
+    p.add_argument("--image-size", type=int, default=256)
+    p.add_argument("--num-classes", type=int, default=1000)
+    p.add_argument("--gmm-k", type=int, default=64)
+    p.add_argument("--pca-dims", type=int, default=64)
+    p.add_argument("--sift-step", type=int, default=6)
+    p.add_argument("--solver-epochs", type=int, default=1)
+    p.add_argument("--no-lcs", action="store_true")
+    p.add_argument("--cpu-smoke", action="store_true",
+                   help="tiny CPU run for plumbing checks")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from keystone_amd.parallel.runtime import init_runtime
+    rt = init_runtime()
+    dev = rt.device
+
+    if args.cpu_smoke:
+        args.images_per_gpu, args.image_size = 8, 64
+        args.num_classes, args.gmm_k, args.pca_dims = 8, 4, 16
+
+    from keystone_amd.loaders.synthetic import synthetic_images
+    from keystone_amd.models.imagenet_fv import (ImageNetFVConfig,
+                                                 fit_imagenet_fv)
+
+    n_total = args.images_per_gpu * rt.world_size
+    imgs, labels = synthetic_images(
+        n_total, h=args.image_size, w=args.image_size,
+        num_classes=args.num_classes, seed=0, runtime=rt, device=dev)
+
+    conf = ImageNetFVConfig(
+        num_classes=args.num_classes, sift_step=args.sift_step,
+        sift_bin=6, sift_scales=2, lcs_step=args.sift_step, lcs_subpatch=6,
+        pca_dims=args.pca_dims, gmm_k=args.gmm_k,
+        n_sample=min(100000, 50 * args.images_per_gpu * 36),
+        use_lcs=not args.no_lcs, block_size=4096,
+        solver_epochs=args.solver_epochs, lam=1e-5, mixture_weight=0.5)
+
+    def step():
+        pipe, info = fit_imagenet_fv(imgs, labels, conf)
+        return info
+
+    # ---- warmup
+    for _ in range(args.warmup):
+        info = step()
+    rt.barrier()
+    rt.synchronize()
+
+    # ---- timed
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        info = step()
+    rt.synchronize()
+    rt.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    rt.all_reduce_(t, op="max")
+    elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    images_per_sec = n_total * args.steps / elapsed
+
+    if rt.rank == 0:
+        out = {
+            "metric": "imagenet_fv_fit_images_per_sec",
+            "value": images_per_sec,
+            "unit": "images/s",
+            "n_gpus": rt.world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "imagenet_sift_lcs_fv",
+                "pipeline": "SIFT+LCS->PCA->GMM->FisherVector->"
+                            "SignedHellinger+L2->BlockWeightedLeastSquares",
+                "global_batch": n_total,
+                "images_per_gpu": args.images_per_gpu,
+                "image_size": args.image_size,
+                "num_classes": args.num_classes,
+                "gmm_k": args.gmm_k,
+                "pca_dims": args.pca_dims,
+                "feature_dim": info.get("feature_dim"),
+                "fit_time_s_per_step": ms_per_step / 1000.0,
+                "parallelism": f"dp{rt.world_size}",
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
