@@ -282,16 +282,27 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                 Wb[:, c] = RB[:, j].to(dtype)
                             continue
                         if n_c <= self.woodbury_threshold * db:
-                            U = torch.stack([
+                            G_n = len(cls_list)
+                            # flat [db, G*nc] solve against the single L
+                            # (broadcasted batched cholesky_solve is
+                            # unreliable on ROCm), then per-class algebra
+                            # as batched GEMMs on views.
+                            U = torch.cat([
                                 (wcs[c] ** 0.5) * Xb64[own_masks[c]].T
-                                for c in cls_list])              # [G, db, nc]
-                            UB = torch.cholesky_solve(U, L.unsqueeze(0))
+                                for c in cls_list], dim=1)       # [db, G*nc]
+                            UBf = torch.cholesky_solve(U, L)     # [db, G*nc]
+                            Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
+                            UBb = UBf.T.reshape(G_n, n_c, db)
                             S = torch.eye(n_c, dtype=torch.float64,
                                           device=Xb.device) \
-                                + U.transpose(1, 2) @ UB         # [G, nc, nc]
+                                + Ub @ UBb.transpose(1, 2)       # [G, nc, nc]
                             r = RB[:, js].T.unsqueeze(2)         # [G, db, 1]
-                            t = U.transpose(1, 2) @ r            # [G, nc, 1]
-                            corr = (UB @ torch.linalg.solve(S, t)).squeeze(2)
+                            t = Ub @ r                           # [G, nc, 1]
+                            if n_c == 1:
+                                z = t / S
+                            else:
+                                z = torch.linalg.solve(S, t)
+                            corr = (UBb.transpose(1, 2) @ z).squeeze(2)
                             sol = (RB[:, js].T - corr).to(dtype)  # [G, db]
                             for g, c in enumerate(cls_list):
                                 Wb[:, c] = sol[g]
