@@ -99,6 +99,11 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     images_per_sec = n_total * args.steps / elapsed
 
+    if os.environ.get("KEYSTONE_BENCH_STAGES") and rt.rank == 0:
+        stages = {k: round(v, 4) for k, v in info.items()
+                  if k.startswith("t_")}
+        print(json.dumps({"stage_seconds": stages}), file=sys.stderr)
+
     if rt.rank == 0:
         out = {
             "metric": "imagenet_fv_fit_images_per_sec",
