@@ -230,7 +230,14 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
 
         own = [c for c in range(K) if (c % max(1, rt.world_size)) == rt.rank] \
             if rt.is_distributed else list(range(K))
-        own_masks = {c: (cls == c) for c in own}
+        # sort local rows by class ONCE: every per-class access below is a
+        # contiguous slice (no boolean-mask partition kernels per class)
+        sort_order = torch.argsort(cls, stable=True)
+        X_full, Y, cls = X_full[sort_order], Y[sort_order], cls[sort_order]
+        counts_local = torch.bincount(cls, minlength=K)
+        ends = counts_local.cumsum(0)
+        starts = ends - counts_local
+        cls_slice = {c: (int(starts[c]), int(ends[c])) for c in own}
 
         blocks = list(torch.split(X_full, self.block_size, dim=1))
         nb = len(blocks)
@@ -260,21 +267,26 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     B64 += self.lam * torch.eye(db, dtype=torch.float64,
                                                 device=Xb.device)
                     L = torch.linalg.cholesky(B64)
-                    T64 = T.to(torch.float64)
                     Xb64 = Xb.to(torch.float64)
-                    rhs = torch.empty(db, len(own), dtype=torch.float64,
-                                      device=Xb.device)
                     wcs = {c: alpha / counts[c].item() for c in own}
-                    for j, c in enumerate(own):
-                        m = own_masks[c]
-                        rhs[:, j] = base_w * xtr_all[:, c].to(torch.float64) \
-                            + wcs[c] * (Xb64[m].T @ T64[m, c])
+                    # rhs for ALL classes in two fused passes: the class-c
+                    # correction X_c^T T[rows_c, c] is a segmented
+                    # reduction over class-sorted rows (one index_add_).
+                    v = T.gather(1, cls.unsqueeze(1)).to(torch.float64)  # [n,1]
+                    seg = torch.zeros(K, db, dtype=torch.float64,
+                                      device=Xb.device)
+                    seg.index_add_(0, cls, Xb64 * v)
+                    wc_t = torch.tensor([wcs[c] for c in own],
+                                        dtype=torch.float64, device=Xb.device)
+                    rhs = base_w * xtr_all[:, own].to(torch.float64) \
+                        + seg[own].T * wc_t
                     RB = torch.cholesky_solve(rhs, L)        # B^{-1} rhs
                     col = {c: j for j, c in enumerate(own)}
 
                     by_nc: dict = {}
                     for c in own:
-                        by_nc.setdefault(int(own_masks[c].sum()), []).append(c)
+                        s, e = cls_slice[c]
+                        by_nc.setdefault(e - s, []).append(c)
                     for n_c, cls_list in by_nc.items():
                         js = [col[c] for c in cls_list]
                         if n_c == 0:
@@ -288,7 +300,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             # unreliable on ROCm), then per-class algebra
                             # as batched GEMMs on views.
                             U = torch.cat([
-                                (wcs[c] ** 0.5) * Xb64[own_masks[c]].T
+                                (wcs[c] ** 0.5)
+                                * Xb64[cls_slice[c][0]:cls_slice[c][1]].T
                                 for c in cls_list], dim=1)       # [db, G*nc]
                             UBf = torch.cholesky_solve(U, L)     # [db, G*nc]
                             Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
@@ -308,8 +321,9 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                 Wb[:, c] = sol[g]
                         else:
                             for c, j in zip(cls_list, js):
-                                G = B64 + wcs[c] * (
-                                    Xb64[own_masks[c]].T @ Xb64[own_masks[c]])
+                                s, e = cls_slice[c]
+                                Xc64 = Xb64[s:e]
+                                G = B64 + wcs[c] * (Xc64.T @ Xc64)
                                 Lc = torch.linalg.cholesky(G)
                                 Wb[:, c] = torch.cholesky_solve(
                                     rhs[:, j:j + 1], Lc).squeeze(1).to(dtype)
