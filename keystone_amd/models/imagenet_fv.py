@@ -60,17 +60,32 @@ class ImageNetFVConfig:
 
 
 def _fit_branch(extractor: PipelineBase, train_imgs: TensorDataset,
-                conf: ImageNetFVConfig, seed: int
+                conf: ImageNetFVConfig, seed: int,
+                timings: Optional[dict] = None, tag: str = ""
                 ) -> Tuple[Pipeline, TensorDataset]:
     """Fit PCA+GMM for one descriptor branch; returns (fitted transformer
     chain for inference, encoded training features)."""
+    import time
+
+    def tick(name, t0):
+        if timings is not None:
+            if train_imgs.data.is_cuda:
+                torch.cuda.synchronize()
+            timings[f"t_{tag}{name}"] = round(time.perf_counter() - t0, 4)
+        return time.perf_counter()
+
+    t = time.perf_counter()
     # descriptors once per shard: [B, n_desc, d] resident in HBM
     desc = extractor.apply(train_imgs).get()
+    t = tick("extract", t)
     sample = ColumnSampler(conf.n_sample, seed=seed).apply_dataset(desc)
+    t = tick("sample", t)
     pca = DistributedPCAEstimator(conf.pca_dims).fit(sample)   # TSQR + SVD
+    t = tick("pca", t)
     proj_sample = TensorDataset(pca.apply_batch(sample.data), sample.runtime)
     gmm = GaussianMixtureModelEstimator(conf.gmm_k, max_iterations=20,
                                         seed=seed).fit(proj_sample)
+    t = tick("gmm", t)
     fv = FisherVector(gmm)
     post = SignedHellingerMapper().and_then(NormalizeRows())
 
@@ -79,6 +94,7 @@ def _fit_branch(extractor: PipelineBase, train_imgs: TensorDataset,
     feats = post.apply(
         TensorDataset(fv.apply_batch(pca.apply_batch(desc.data)),
                       desc.runtime)).get()
+    tick("fv", t)
     return chain, feats
 
 
@@ -91,9 +107,12 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
     info = {}
     t0 = time.perf_counter()
 
+    import os
+    tm = info if os.environ.get("KEYSTONE_BENCH_STAGES") else None
     sift = GrayScaler().and_then(
         SIFTExtractor(conf.sift_step, conf.sift_bin, conf.sift_scales))
-    sift_chain, sift_feats = _fit_branch(sift, train_imgs, conf, conf.seed)
+    sift_chain, sift_feats = _fit_branch(sift, train_imgs, conf, conf.seed,
+                                         tm, "sift_")
     info["t_sift_branch"] = time.perf_counter() - t0
 
     branches = [sift_chain]
@@ -103,7 +122,7 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
         lcs = GrayScaler().and_then(
             LCSExtractor(conf.lcs_step, conf.lcs_subpatch))
         lcs_chain, lcs_feats = _fit_branch(lcs, train_imgs, conf,
-                                           conf.seed + 1)
+                                           conf.seed + 1, tm, "lcs_")
         branches.append(lcs_chain)
         feats = torch.cat([feats, lcs_feats.data], dim=1)
         info["t_lcs_branch"] = time.perf_counter() - t1

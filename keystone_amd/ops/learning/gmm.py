@@ -99,9 +99,16 @@ class GaussianMixtureModelEstimator(Estimator):
             lp = model.log_prob(X)                       # [n, K]
             lse = torch.logsumexp(lp, dim=1)
             gamma = torch.exp(lp - lse.unsqueeze(1))     # responsibilities
+            # sufficient statistics: fp64 for fp64 inputs (oracle parity);
+            # fp32 GEMMs (the MFMA fast path) otherwise, promoted to fp64
+            # only for the (tiny) reduced accumulators
             Nk = gamma.sum(0, dtype=torch.float64)                       # [K]
-            Sx = gamma.T.to(torch.float64) @ X.to(torch.float64)         # [K,D]
-            Sxx = gamma.T.to(torch.float64) @ (X.to(torch.float64) ** 2)
+            if X.dtype == torch.float64:
+                Sx = gamma.T @ X
+                Sxx = gamma.T @ (X * X)
+            else:
+                Sx = (gamma.T @ X).to(torch.float64)
+                Sxx = (gamma.T @ (X * X)).to(torch.float64)
             ll = torch.tensor([float(lse.sum(dtype=torch.float64))],
                               dtype=torch.float64, device=dev)
             rt.all_reduce_many_([Nk, Sx, Sxx, ll])

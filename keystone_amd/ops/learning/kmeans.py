@@ -54,10 +54,12 @@ class KMeansPlusPlusEstimator(Estimator):
         K = self.num_clusters
         dev, dt = X.device, X.dtype
 
-        # --- k-means++ init on a gathered sample (deterministic, replicated)
+        # --- k-means++ init on a gathered sample (deterministic, replicated;
+        # the K-step greedy loop runs on CPU — tiny data, and per-step GPU
+        # launches + .item() syncs would dominate)
         sample = data.sample_local(max(8 * K // max(1, rt.world_size), K),
                                    self.seed)
-        S = sample.collect().to(torch.float64)
+        S = sample.collect().to(torch.float64).cpu()
         g = torch.Generator().manual_seed(self.seed)
         centers = torch.empty(K, X.shape[1], dtype=torch.float64, device=S.device)
         first = int(torch.randint(0, S.shape[0], (1,), generator=g).item())
