@@ -36,7 +36,7 @@ import torch
 from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.pipeline import (BatchTransformer, LabelEstimator,
                                   Transformer, TupleDataset)
-from .linear import _solve_spd
+from .linear import _solve_dtype, _solve_spd
 
 __all__ = ["BlockLinearMapper", "BlockLeastSquaresEstimator",
            "BlockWeightedLeastSquaresEstimator"]
@@ -263,22 +263,23 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     # Classes with equal n_c are solved as ONE batched
                     # update (K=1000 at ImageNet scale -> a handful of
                     # batched rocSOLVER/GEMM calls, not 1000 Python steps).
-                    B64 = base_w * gram_all.to(torch.float64)
-                    B64 += self.lam * torch.eye(db, dtype=torch.float64,
+                    sdt = _solve_dtype(dtype)
+                    B64 = base_w * gram_all.to(sdt)
+                    B64 += self.lam * torch.eye(db, dtype=sdt,
                                                 device=Xb.device)
                     L = torch.linalg.cholesky(B64)
-                    Xb64 = Xb.to(torch.float64)
+                    Xb64 = Xb.to(sdt)
                     wcs = {c: alpha / counts[c].item() for c in own}
                     # rhs for ALL classes in two fused passes: the class-c
                     # correction X_c^T T[rows_c, c] is a segmented
                     # reduction over class-sorted rows (one index_add_).
-                    v = T.gather(1, cls.unsqueeze(1)).to(torch.float64)  # [n,1]
-                    seg = torch.zeros(K, db, dtype=torch.float64,
+                    v = T.gather(1, cls.unsqueeze(1)).to(sdt)  # [n,1]
+                    seg = torch.zeros(K, db, dtype=sdt,
                                       device=Xb.device)
                     seg.index_add_(0, cls, Xb64 * v)
                     wc_t = torch.tensor([wcs[c] for c in own],
-                                        dtype=torch.float64, device=Xb.device)
-                    rhs = base_w * xtr_all[:, own].to(torch.float64) \
+                                        dtype=sdt, device=Xb.device)
+                    rhs = base_w * xtr_all[:, own].to(sdt) \
                         + seg[own].T * wc_t
                     RB = torch.cholesky_solve(rhs, L)        # B^{-1} rhs
                     col = {c: j for j, c in enumerate(own)}
@@ -306,7 +307,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             UBf = torch.cholesky_solve(U, L)     # [db, G*nc]
                             Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
                             UBb = UBf.T.reshape(G_n, n_c, db)
-                            S = torch.eye(n_c, dtype=torch.float64,
+                            S = torch.eye(n_c, dtype=sdt,
                                           device=Xb.device) \
                                 + Ub @ UBb.transpose(1, 2)       # [G, nc, nc]
                             r = RB[:, js].T.unsqueeze(2)         # [G, db, 1]

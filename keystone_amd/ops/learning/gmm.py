@@ -117,8 +117,10 @@ class GaussianMixtureModelEstimator(Estimator):
             var64 = (Sxx / Nk_c - mu64 ** 2).clamp_min(self.min_variance)
             model = GaussianMixtureModel(mu64.to(dt), var64.to(dt),
                                          (Nk / n_tot).to(dt))
-            ll_now = ll.item() / n_tot
-            if abs(ll_now - prev_ll) < self.tol * max(1.0, abs(prev_ll)):
-                break
-            prev_ll = ll_now
+            # convergence check every 4 iters: .item() is a device sync
+            if (_it & 3) == 3:
+                ll_now = ll.item() / n_tot
+                if abs(ll_now - prev_ll) < self.tol * max(1.0, abs(prev_ll)):
+                    break
+                prev_ll = ll_now
         return model
