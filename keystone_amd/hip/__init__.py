@@ -87,14 +87,12 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     """argtypes for kernels added after v1 (ignore if absent)."""
     c_p, c_i, c_f = ctypes.c_void_p, ctypes.c_int, ctypes.c_float
     try:
-        lib.ks_fisher_vector_f32.argtypes = [c_p, c_p, c_p, c_p, c_p, c_i,
-                                             c_i, c_i, c_i, c_p]
+        lib.ks_fisher_vector_f32.argtypes = [c_p] * 8 + [c_i] * 4 + [c_p]
         lib.ks_fisher_vector_f32.restype = c_i
     except AttributeError:
         pass
     try:
-        lib.ks_gmm_posteriors_f32.argtypes = [c_p, c_p, c_p, c_p, c_p, c_p,
-                                              c_i, c_i, c_i, c_p]
+        lib.ks_gmm_posteriors_f32.argtypes = [c_p] * 6 + [c_i] * 3 + [c_p]
         lib.ks_gmm_posteriors_f32.restype = c_i
     except AttributeError:
         pass
@@ -172,6 +170,68 @@ def hellinger_l2(X: torch.Tensor, eps: float = 1e-12) -> torch.Tensor:
     _check(lib.ks_hellinger_l2_f32(
         ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         rows, cols, eps, _stream()), "ks_hellinger_l2_f32")
+    return out
+
+
+def _gmm_coeffs(mu: torch.Tensor, var: torch.Tensor, w: torch.Tensor):
+    """Host-precomputed expanded-Gaussian coefficients (see gmm_fv.hip)."""
+    import math
+    a = (0.5 / var).contiguous()
+    b = (mu / var).contiguous()
+    c = ((mu * mu * a).sum(1) + 0.5 * var.log().sum(1)
+         + 0.5 * mu.shape[1] * math.log(2 * math.pi)
+         - w.clamp_min(1e-30).log()).contiguous()
+    return a, b, c
+
+
+def gmm_supported(D: int, K: int) -> bool:
+    return D <= 64 and K <= 128
+
+
+def fv_supported(D: int, K: int) -> bool:
+    if D > 64 or K > 128 or K % 16 != 0 or (K // 4) not in (4, 8, 12, 16, 24, 32):
+        return False
+    return (2 * K * D + K + 128 * D + 128 * K) * 4 <= 160 * 1024
+
+
+def gmm_posteriors(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
+                   w: torch.Tensor, want_ll: bool = False):
+    """gamma = softmax_k log N(x|k) + log w_k; optional summed logsumexp."""
+    lib = _load()
+    X = _prep(X, "gmm_posteriors.X")
+    n, D = X.shape
+    K = mu.shape[0]
+    a, b, c = _gmm_coeffs(_prep(mu, "mu"), _prep(var, "var"),
+                          _prep(w, "w"))
+    gamma = torch.empty(n, K, dtype=torch.float32, device=X.device)
+    ll = torch.zeros(1, dtype=torch.float32, device=X.device) if want_ll else None
+    _check(lib.ks_gmm_posteriors_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(a.data_ptr()),
+        ctypes.c_void_p(b.data_ptr()), ctypes.c_void_p(c.data_ptr()),
+        ctypes.c_void_p(gamma.data_ptr()),
+        ctypes.c_void_p(ll.data_ptr()) if ll is not None else None,
+        n, D, K, _stream()), "ks_gmm_posteriors_f32")
+    return (gamma, ll) if want_ll else gamma
+
+
+def fisher_vector(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
+                  w: torch.Tensor) -> torch.Tensor:
+    """Improved FV per image: X [B, n, D] -> [B, 2*K*D]."""
+    lib = _load()
+    X = _prep(X, "fisher_vector.X")
+    B, n, D = X.shape
+    K = mu.shape[0]
+    mu = _prep(mu, "mu")
+    var = _prep(var, "var")
+    w = _prep(w, "w")
+    a, b, c = _gmm_coeffs(mu, var, w)
+    out = torch.empty(B, 2 * K * D, dtype=torch.float32, device=X.device)
+    _check(lib.ks_fisher_vector_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(a.data_ptr()),
+        ctypes.c_void_p(b.data_ptr()), ctypes.c_void_p(c.data_ptr()),
+        ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
+        ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        B, n, D, K, _stream()), "ks_fisher_vector_f32")
     return out
 
 

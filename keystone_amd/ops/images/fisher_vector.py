@@ -49,9 +49,9 @@ class FisherVector(Transformer):
         w = g.weights.to(device=X.device, dtype=X.dtype)       # [K]
         K = mu.shape[0]
 
-        if X.is_cuda and X.dtype == torch.float32:
+        if X.is_cuda and X.dtype == torch.float32 and self.improved_scaling:
             from ... import hip as hip_ops
-            if hasattr(hip_ops._load(), "ks_fisher_vector_f32"):
+            if hip_ops.fv_supported(D, K):
                 out = hip_ops.fisher_vector(X.contiguous(), mu, var, w)
                 return out.squeeze(0) if squeeze else out
 

@@ -94,11 +94,24 @@ class GaussianMixtureModelEstimator(Estimator):
         w = torch.full((K,), 1.0 / K, dtype=dt, device=dev)
 
         model = GaussianMixtureModel(mu, var, w)
+        use_hip = False
+        if X.is_cuda and X.dtype == torch.float32:
+            from ... import hip as hip_ops
+            use_hip = hip_ops.gmm_supported(D, K) and hip_ops.available()
         prev_ll = -float("inf")
         for _it in range(self.max_iterations):
-            lp = model.log_prob(X)                       # [n, K]
-            lse = torch.logsumexp(lp, dim=1)
-            gamma = torch.exp(lp - lse.unsqueeze(1))     # responsibilities
+            if use_hip:
+                # K2: fused logprob+softmax(+logsumexp) in one HIP launch
+                from ... import hip as hip_ops
+                gamma, ll_t = hip_ops.gmm_posteriors(
+                    X, model.means, model.variances, model.weights,
+                    want_ll=True)
+                lse_sum = ll_t.to(torch.float64)
+            else:
+                lp = model.log_prob(X)                   # [n, K]
+                lse = torch.logsumexp(lp, dim=1)
+                gamma = torch.exp(lp - lse.unsqueeze(1))  # responsibilities
+                lse_sum = None
             # sufficient statistics: fp64 for fp64 inputs (oracle parity);
             # fp32 GEMMs (the MFMA fast path) otherwise, promoted to fp64
             # only for the (tiny) reduced accumulators
@@ -109,8 +122,10 @@ class GaussianMixtureModelEstimator(Estimator):
             else:
                 Sx = (gamma.T @ X).to(torch.float64)
                 Sxx = (gamma.T @ (X * X)).to(torch.float64)
-            ll = torch.tensor([float(lse.sum(dtype=torch.float64))],
-                              dtype=torch.float64, device=dev)
+            if lse_sum is not None:
+                ll = lse_sum.reshape(1)
+            else:
+                ll = lse.sum(dtype=torch.float64).reshape(1)
             rt.all_reduce_many_([Nk, Sx, Sxx, ll])
             Nk_c = Nk.clamp_min(1e-10).unsqueeze(1)
             mu64 = Sx / Nk_c

@@ -18,6 +18,28 @@ import torch.distributed as dist
 import torch.multiprocessing as mp
 
 
+def _to_plain(x):
+    """Detach results from torch's shared-memory IPC: tensors are sent by
+    value (the child exits before the parent could detach a shared FD)."""
+    if isinstance(x, torch.Tensor):
+        return ("__tensor__", x.numpy().copy())
+    if isinstance(x, (list, tuple)):
+        return type(x)(_to_plain(v) for v in x)
+    if isinstance(x, dict):
+        return {k: _to_plain(v) for k, v in x.items()}
+    return x
+
+
+def _from_plain(x):
+    if isinstance(x, tuple) and len(x) == 2 and x[0] == "__tensor__":
+        return torch.from_numpy(x[1])
+    if isinstance(x, (list, tuple)):
+        return type(x)(_from_plain(v) for v in x)
+    if isinstance(x, dict):
+        return {k: _from_plain(v) for k, v in x.items()}
+    return x
+
+
 def _worker(rank: int, world: int, store_path: str, fn, args, q):
     try:
         store = dist.FileStore(store_path, world)
@@ -28,7 +50,7 @@ def _worker(rank: int, world: int, store_path: str, fn, args, q):
         set_runtime(Runtime(rank=rank, world_size=world,
                             device=torch.device("cpu"), backend="gloo"))
         result = fn(rank, world, *args)
-        q.put((rank, "ok", result))
+        q.put((rank, "ok", _to_plain(result)))
     except Exception:
         q.put((rank, "err", traceback.format_exc()))
     finally:
@@ -55,7 +77,7 @@ def run_distributed(fn, world_size: int = 2, args: tuple = ()):
                 for p in procs:
                     p.terminate()
                 raise RuntimeError(f"rank {rank} failed:\n{payload}")
-            results[rank] = payload
+            results[rank] = _from_plain(payload)
         for p in procs:
             p.join(timeout=60)
     return [results[r] for r in range(world_size)]
