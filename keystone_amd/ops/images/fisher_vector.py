@@ -55,9 +55,11 @@ class FisherVector(Transformer):
                 out = hip_ops.fisher_vector(X.contiguous(), mu, var, w)
                 return out.squeeze(0) if squeeze else out
 
-        # log N(x|k) via the expanded-GEMM form (see gmm.log_prob)
+        # log N(x|k) via the expanded-GEMM form (see gmm.log_prob);
+        # X^2 hoisted (shared by the logit GEMM and S2)
+        X2 = X * X
         inv2v = 0.5 / var
-        a = (X * X) @ inv2v.T                                   # [B, n, K]
+        a = X2 @ inv2v.T                                        # [B, n, K]
         b = X @ (mu / var).T
         c = (mu * mu * inv2v).sum(1) + 0.5 * var.log().sum(1)
         lp = -a + b - c + w.clamp_min(1e-30).log()
@@ -65,7 +67,7 @@ class FisherVector(Transformer):
 
         S0 = gamma.sum(dim=1)                                   # [B, K]
         S1 = gamma.transpose(1, 2) @ X                          # [B, K, D]
-        S2 = gamma.transpose(1, 2) @ (X * X)                    # [B, K, D]
+        S2 = gamma.transpose(1, 2) @ X2                         # [B, K, D]
 
         sig = var.sqrt()
         s0 = S0.unsqueeze(2)
