@@ -49,7 +49,13 @@ class FisherVector(Transformer):
         w = g.weights.to(device=X.device, dtype=X.dtype)       # [K]
         K = mu.shape[0]
 
-        if X.is_cuda and X.dtype == torch.float32 and self.improved_scaling:
+        # The fused per-image kernel (hip/src/gmm_fv.hip) is measured 16x
+        # SLOWER than this batched-GEMM formulation at FV-pipeline shapes
+        # (rocBLAS owns these GEMMs; the kernel's posterior phase is
+        # LDS-read-issue bound) — opt-in only, kept for small-batch regimes.
+        import os
+        if (X.is_cuda and X.dtype == torch.float32 and self.improved_scaling
+                and os.environ.get("KEYSTONE_FV_KERNEL")):
             from ... import hip as hip_ops
             if hip_ops.fv_supported(D, K):
                 out = hip_ops.fisher_vector(X.contiguous(), mu, var, w)
