@@ -70,6 +70,19 @@ class GaussianMixtureModelEstimator(Estimator):
         self.init = init
 
     def _fit(self, data: TensorDataset) -> GaussianMixtureModel:
+        import os
+        import time
+        dbg = os.environ.get("KEYSTONE_GMM_DEBUG")
+
+        def tick(name, t0):
+            if dbg:
+                if data.data.is_cuda:
+                    torch.cuda.synchronize()
+                print(f"[gmm] {name}: {time.perf_counter() - t0:.4f}s",
+                      flush=True)
+            return time.perf_counter()
+
+        t = time.perf_counter()
         X = data.data
         rt = data.runtime
         K, D = self.k, X.shape[1]
@@ -79,6 +92,7 @@ class GaussianMixtureModelEstimator(Estimator):
         if self.init == "kmeans":
             km = KMeansPlusPlusEstimator(K, max_iterations=5, seed=self.seed)._fit(data)
             mu = km.centers.to(dt)
+            t = tick("kmeans_init", t)
         else:
             S = data.sample_local(max(2 * K // max(1, rt.world_size), K),
                                   self.seed).collect()
@@ -93,6 +107,7 @@ class GaussianMixtureModelEstimator(Estimator):
         var = gvar.unsqueeze(0).expand(K, D).to(dt).contiguous()
         w = torch.full((K,), 1.0 / K, dtype=dt, device=dev)
 
+        t = tick("global_stats", t)
         model = GaussianMixtureModel(mu, var, w)
         # X^2 is reused by the logit GEMM and the Sxx statistic every
         # iteration — hoist it out of the EM loop (one extra resident
@@ -139,4 +154,5 @@ class GaussianMixtureModelEstimator(Estimator):
                 if abs(ll_now - prev_ll) < self.tol * max(1.0, abs(prev_ll)):
                     break
                 prev_ll = ll_now
+        tick(f"em_loop({_it + 1} iters)", t)
         return model
