@@ -72,13 +72,18 @@ class KMeansPlusPlusEstimator(Estimator):
             d2 = torch.minimum(d2, ((S - centers[k]) ** 2).sum(1))
         C = centers.to(device=dev, dtype=dt)
 
-        # --- Lloyd iterations
+        # --- Lloyd iterations.  Per-centroid (sum, count) via a one-hot
+        # GEMM (assignment matrix^T @ X) — a large fp64 index_add_ runs at
+        # ~5 GB/s on ROCm while the GEMM runs at the MFMA rate.
+        xsq = (X * X).sum(1, keepdim=True)
+        acc_dt = torch.float64 if dt == torch.float64 else torch.float32
         for _it in range(self.max_iterations):
-            d2m = (X * X).sum(1, keepdim=True) - 2.0 * (X @ C.T) + (C * C).sum(1)
+            d2m = xsq - 2.0 * (X @ C.T) + (C * C).sum(1)
             idx = d2m.argmin(dim=1)
-            sums = torch.zeros(K, X.shape[1], dtype=torch.float64, device=dev)
-            sums.index_add_(0, idx, X.to(torch.float64))
-            cnt = torch.bincount(idx, minlength=K).to(torch.float64)
+            onehot = torch.zeros(X.shape[0], K, dtype=acc_dt, device=dev)
+            onehot.scatter_(1, idx.unsqueeze(1), 1.0)
+            sums = (onehot.T @ X.to(acc_dt)).to(torch.float64)
+            cnt = onehot.sum(0, dtype=torch.float64)
             rt.all_reduce_many_([sums, cnt])
             newC = torch.where(cnt.unsqueeze(1) > 0,
                                sums / cnt.clamp_min(1.0).unsqueeze(1),
