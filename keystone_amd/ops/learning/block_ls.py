@@ -227,6 +227,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         rt.all_reduce_many_([n, counts])
         n_tot = n.item()
         counts = counts.clamp_min(1.0)
+        counts_cpu = counts.cpu()
 
         own = [c for c in range(K) if (c % max(1, rt.world_size)) == rt.rank] \
             if rt.is_distributed else list(range(K))
@@ -235,8 +236,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         sort_order = torch.argsort(cls, stable=True)
         X_full, Y, cls = X_full[sort_order], Y[sort_order], cls[sort_order]
         counts_local = torch.bincount(cls, minlength=K)
-        ends = counts_local.cumsum(0)
-        starts = ends - counts_local
+        ends = counts_local.cumsum(0).cpu()
+        starts = (ends - counts_local.cpu())
         cls_slice = {c: (int(starts[c]), int(ends[c])) for c in own}
 
         blocks = list(torch.split(X_full, self.block_size, dim=1))
@@ -269,7 +270,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                                 device=Xb.device)
                     L = torch.linalg.cholesky(B64)
                     Xb64 = Xb.to(sdt)
-                    wcs = {c: alpha / counts[c].item() for c in own}
+                    wcs = {c: alpha / float(counts_cpu[c]) for c in own}
                     # rhs for ALL classes in two fused passes: the class-c
                     # correction X_c^T T[rows_c, c] is a segmented
                     # reduction over class-sorted rows (one index_add_).
@@ -291,8 +292,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     for n_c, cls_list in by_nc.items():
                         js = [col[c] for c in cls_list]
                         if n_c == 0:
-                            for c, j in zip(cls_list, js):
-                                Wb[:, c] = RB[:, j].to(dtype)
+                            Wb[:, torch.tensor(cls_list, device=Wb.device)] \
+                                = RB[:, js].to(dtype)
                             continue
                         if n_c <= self.woodbury_threshold * db:
                             G_n = len(cls_list)
@@ -318,8 +319,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                 z = torch.linalg.solve(S, t)
                             corr = (UBb.transpose(1, 2) @ z).squeeze(2)
                             sol = (RB[:, js].T - corr).to(dtype)  # [G, db]
-                            for g, c in enumerate(cls_list):
-                                Wb[:, c] = sol[g]
+                            Wb[:, torch.tensor(cls_list,
+                                               device=Wb.device)] = sol.T
                         else:
                             for c, j in zip(cls_list, js):
                                 s, e = cls_slice[c]

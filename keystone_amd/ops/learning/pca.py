@@ -91,10 +91,16 @@ class TSQR:
 
     @staticmethod
     def factor(data: TensorDataset) -> torch.Tensor:
-        X = data.data.to(torch.float64)
+        # factor dtype follows the input: fp64 shards keep the fp64 oracle
+        # path; fp32 shards use the ~10x-faster fp32 geqrf (QR avoids the
+        # condition-squaring of normal equations, so fp32 is stable here);
+        # the tiny stacked-R combine is always fp64.
+        X = data.data
+        if X.dtype not in (torch.float32, torch.float64):
+            X = X.to(torch.float32)
         rt = data.runtime
         # local QR of the HBM shard
-        R = torch.linalg.qr(X, mode="r").R                     # [d, d]
+        R = torch.linalg.qr(X, mode="r").R.to(torch.float64)   # [d, d]
         if rt.is_distributed:
             Rs = rt.all_gather(R.contiguous())
             R = torch.linalg.qr(torch.cat(Rs, dim=0), mode="r").R
