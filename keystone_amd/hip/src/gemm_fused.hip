@@ -34,8 +34,13 @@ void gemm_fused_f32_kernel(const float* __restrict__ A,  // [M, K] row-major
                            const float* __restrict__ bias,  // [N] or null
                            float* __restrict__ C,        // [M, N]
                            int M, int K, int N, float scale, float alpha) {
-    __shared__ float As[BM][BK + APAD];
-    __shared__ float Bs[BK][BN];
+    // double-buffered LDS, 2-phase pipeline (guide T3 minimum form +
+    // T14 reg-staging): per K-tile, issue next tile's global loads into
+    // registers FIRST, run the MFMA phase on the current buffer (hides
+    // the ~900-cycle HBM latency), then ds_write the landed registers
+    // into the other buffer; ONE barrier per tile.
+    __shared__ float As[2][BM][BK + APAD];
+    __shared__ float Bs[2][BK][BN];
 
     const unsigned nbn = (unsigned)ks_ceil_div(N, BN);
     const unsigned nwg = nbn * (unsigned)ks_ceil_div(M, BM);
@@ -50,75 +55,90 @@ void gemm_fused_f32_kernel(const float* __restrict__ A,  // [M, K] row-major
     const int l31 = lane & 31;
     const int lk = lane >> 5;                  // 0/1: k offset inside MFMA
 
+    // per-thread staging geometry (same for every tile)
+    const int a_kv = (tid & 7) * 4;            // A: k offset 0..28
+    const int a_r0 = tid >> 3;                 // A: row 0..31 (x4 passes)
+    const int b_nv = (tid & 31) * 4;           // B: col offset 0..124
+    const int b_k0 = tid >> 5;                 // B: k row 0..7 (x4 passes)
+
+    float4 ra[4], rb[4];
+
+    auto stage_load = [&](int k0) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int gr = brow + a_r0 + p * 32;
+            const int gk = k0 + a_kv;
+            float4 v = {0, 0, 0, 0};
+            if (gr < M) {
+                if (gk + 3 < K) {
+                    v = *reinterpret_cast<const float4*>(&A[(size_t)gr * K + gk]);
+                } else {
+#pragma unroll
+                    for (int q = 0; q < 4; ++q)
+                        if (gk + q < K) ((float*)&v)[q] = A[(size_t)gr * K + gk + q];
+                }
+            }
+            ra[p] = v;
+        }
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int gk = k0 + b_k0 + p * 8;
+            const int gn = bcol + b_nv;
+            float4 v = {0, 0, 0, 0};
+            if (gk < K) {
+                if (gn + 3 < N) {
+                    v = *reinterpret_cast<const float4*>(&B[(size_t)gk * N + gn]);
+                } else {
+#pragma unroll
+                    for (int q = 0; q < 4; ++q)
+                        if (gn + q < N) ((float*)&v)[q] = B[(size_t)gk * N + gn + q];
+                }
+            }
+            rb[p] = v;
+        }
+    };
+
+    auto stage_write = [&](int buf) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int row = a_r0 + p * 32;
+            As[buf][row][a_kv + 0] = ra[p].x;
+            As[buf][row][a_kv + 1] = ra[p].y;
+            As[buf][row][a_kv + 2] = ra[p].z;
+            As[buf][row][a_kv + 3] = ra[p].w;
+            *reinterpret_cast<float4*>(&Bs[buf][b_k0 + p * 8][b_nv]) = rb[p];
+        }
+    };
+
     f32x16 acc[2][2];
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.0f);
 
-    for (int k0 = 0; k0 < K; k0 += BK) {
-        // ---- stage A: 128 rows x 32 k, float4 per thread x 4 passes
-        {
-            const int kv = (tid & 7) * 4;        // k offset (0..28 step 4)
-            const int r0 = tid >> 3;             // row 0..31
-#pragma unroll
-            for (int p = 0; p < 4; ++p) {
-                const int row = r0 + p * 32;
-                const int gr = brow + row;
-                float4 v = {0, 0, 0, 0};
-                if (gr < M) {
-                    const int gk = k0 + kv;
-                    if (gk + 3 < K) {
-                        v = *reinterpret_cast<const float4*>(&A[(size_t)gr * K + gk]);
-                    } else {
-#pragma unroll
-                        for (int q = 0; q < 4; ++q)
-                            if (gk + q < K) ((float*)&v)[q] = A[(size_t)gr * K + gk + q];
-                    }
-                }
-                As[row][kv + 0] = v.x;
-                As[row][kv + 1] = v.y;
-                As[row][kv + 2] = v.z;
-                As[row][kv + 3] = v.w;
-            }
-        }
-        // ---- stage B: 32 k x 128 n
-        {
-            const int nv = (tid & 31) * 4;       // col offset 0..124
-            const int kr0 = tid >> 5;            // k row 0..7
-#pragma unroll
-            for (int p = 0; p < 4; ++p) {
-                const int kr = kr0 + p * 8;
-                const int gk = k0 + kr;
-                const int gn = bcol + nv;
-                float4 v = {0, 0, 0, 0};
-                if (gk < K) {
-                    if (gn + 3 < N) {
-                        v = *reinterpret_cast<const float4*>(&B[(size_t)gk * N + gn]);
-                    } else {
-#pragma unroll
-                        for (int q = 0; q < 4; ++q)
-                            if (gn + q < N) ((float*)&v)[q] = B[(size_t)gk * N + gn + q];
-                    }
-                }
-                *reinterpret_cast<float4*>(&Bs[kr][nv]) = v;
-            }
-        }
-        __syncthreads();
+    const int ntiles = ks_ceil_div(K, BK);
+    stage_load(0);
+    stage_write(0);
+    __syncthreads();
 
+    int cur = 0;
+    for (int t = 0; t < ntiles; ++t) {
+        if (t + 1 < ntiles) stage_load((t + 1) * BK);  // async under MFMA
         // ---- MFMA inner loop: 16 k-steps of K=2
 #pragma unroll
         for (int kk = 0; kk < BK; kk += 2) {
-            float a0 = As[wr * 64 + 0 + l31][kk + lk];
-            float a1 = As[wr * 64 + 32 + l31][kk + lk];
-            float b0 = Bs[kk + lk][wc * 64 + 0 + l31];
-            float b1 = Bs[kk + lk][wc * 64 + 32 + l31];
+            float a0 = As[cur][wr * 64 + 0 + l31][kk + lk];
+            float a1 = As[cur][wr * 64 + 32 + l31][kk + lk];
+            float b0 = Bs[cur][kk + lk][wc * 64 + 0 + l31];
+            float b1 = Bs[cur][kk + lk][wc * 64 + 32 + l31];
             acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
             acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
             acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
             acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
         }
+        if (t + 1 < ntiles) stage_write(cur ^ 1);      // vmcnt at first use
         __syncthreads();
+        cur ^= 1;
     }
 
     // ---- epilogue: C/D layout col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)

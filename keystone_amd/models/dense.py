@@ -48,6 +48,22 @@ def build_mnist_randomfft(train, labels_int, conf: MnistConfig = MnistConfig()
             .and_then(MaxClassifier()))
 
 
+def build_mnist_random_cosine(train, labels_int,
+                              conf: MnistConfig = MnistConfig(),
+                              num_features: int = 4096,
+                              gamma: float = 2e-3) -> Pipeline:
+    """Driver config 2: MNIST RandomCosine features + BlockLeastSquares."""
+    ind = ClassLabelIndicatorsFromIntLabels(conf.num_classes)
+    labels = TensorDataset(ind.apply_batch(labels_int.data).float(),
+                           labels_int.runtime)
+    return (CosineRandomFeatures(conf.input_dim, num_features, gamma,
+                                 seed=conf.seed)
+            .and_then(BlockLeastSquaresEstimator(conf.block_size,
+                                                 conf.num_iter, conf.lam),
+                      train, labels)
+            .and_then(MaxClassifier()))
+
+
 @dataclass
 class TimitConfig:
     input_dim: int = 440

@@ -97,3 +97,14 @@ def test_voc_map_pipeline():
     res = MeanAveragePrecisionEvaluator(3).evaluate(scores, labels)
     assert 0.0 <= res["mAP"] <= 1.0
     assert res["mAP"] > 0.5  # train-set ranking should be decent
+
+
+def test_config2_mnist_random_cosine():
+    from keystone_amd.models import build_mnist_random_cosine
+    train, labels = synthetic_dense(300, 64, 10, seed=8, noise=0.8)
+    pipe = build_mnist_random_cosine(train, labels, MnistConfig(
+        input_dim=64, num_classes=10, block_size=128, num_iter=3, lam=1e-4),
+        num_features=512, gamma=0.02)
+    pred = pipe.apply(train).get()
+    m = MulticlassClassifierEvaluator(10).evaluate(pred, labels)
+    assert m.total_accuracy > 0.9
