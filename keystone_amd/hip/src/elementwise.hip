@@ -90,3 +90,46 @@ KS_EXPORT int ks_sym_rectify_f32(const float* X, float* out, int rows,
                        X, out, rows, cols, alpha);
     return (int)hipGetLastError();
 }
+
+// In-place fused bias + cos epilogue: X[r, c] = cos(X[r, c] + b[c]) * scale.
+// Pairs with a library GEMM (hipBLASLt ~150 TF f32) to form the fastest
+// random-cosine-features path: GEMM + ONE elementwise pass instead of
+// eager's 3-4 passes.
+__global__ __launch_bounds__(256)
+void bias_cos_kernel(float* __restrict__ X, const float* __restrict__ b,
+                     int rows, int cols, float scale) {
+    const size_t n = (size_t)rows * cols;
+    size_t i = ((size_t)blockIdx.x * 256 + threadIdx.x) * 4;
+    const size_t stride = (size_t)gridDim.x * 256 * 4;
+    for (; i + 3 < n; i += stride) {
+        float4 v = *reinterpret_cast<float4*>(&X[i]);
+        const int c = (int)(i % cols);
+        // cols is a multiple of 4 in practice; handle wrap generically
+        if (c + 3 < cols) {
+            v.x = __cosf(v.x + b[c]) * scale;
+            v.y = __cosf(v.y + b[c + 1]) * scale;
+            v.z = __cosf(v.z + b[c + 2]) * scale;
+            v.w = __cosf(v.w + b[c + 3]) * scale;
+            *reinterpret_cast<float4*>(&X[i]) = v;
+        } else {
+            for (int q = 0; q < 4; ++q)
+                X[i + q] = __cosf(X[i + q] + b[(i + q) % cols]) * scale;
+        }
+    }
+    if (blockIdx.x == 0 && threadIdx.x < 4) {
+        for (size_t j = n - (n & 3) + threadIdx.x; j < n; j += 4)
+            X[j] = __cosf(X[j] + b[j % cols]) * scale;
+    }
+}
+
+KS_EXPORT int ks_bias_cos_f32(float* X, const float* b, int rows, int cols,
+                              float scale, hipStream_t stream) {
+    if (rows == 0 || cols == 0) return 0;
+    const size_t n = (size_t)rows * cols;
+    int blocks = (int)((n / 4 + 255) / 256);
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(bias_cos_kernel, dim3(blocks), dim3(256), 0, stream,
+                       X, b, rows, cols, scale);
+    return (int)hipGetLastError();
+}

@@ -309,11 +309,11 @@ static int launch_gemm_fused(const float* A, const float* B, const float* bias,
                              float* C, int M, int K, int N, float scale,
                              float alpha, int epilogue, hipStream_t stream) {
     // big shapes go to the 256^2 tile (2x arithmetic intensity; the 128^2
-    // tile is HBM-bound); small/edge shapes keep the 128^2 tile.
-    const bool big = (M >= 512 && N >= 512);
-    const int nwg = big
-        ? ks_ceil_div(M, BM2) * ks_ceil_div(N, BN2)
-        : ks_ceil_div(M, BM) * ks_ceil_div(N, BN);
+    // tile is HBM-bound) — but only when the 256-tile grid still fills the
+    // 256-CU chip (a launch needs >= #CUs workgroups).
+    const int nwg256 = ks_ceil_div(M, BM2) * ks_ceil_div(N, BN2);
+    const bool big = (M >= 512 && N >= 512 && nwg256 >= 256);
+    const int nwg = big ? nwg256 : ks_ceil_div(M, BM) * ks_ceil_div(N, BN);
     dim3 grid(nwg), block(big ? 512 : 256);
 #define KS_LAUNCH(EPI_)                                                     \
     do {                                                                    \

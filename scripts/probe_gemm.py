@@ -42,9 +42,11 @@ def main():
     W = torch.randn(K, N, device=dev) * 0.1
     b = torch.rand(N, device=dev)
     t = bench(lambda: hip.gemm_cos(X, W, b, 0.1))
+    import os; os.environ["KEYSTONE_FUSED_GEMM_COS"]="1"
+    t_full = bench(lambda: hip.gemm_cos(X, W, b, 0.1)); del os.environ["KEYSTONE_FUSED_GEMM_COS"]
     tf = 2 * M * K * N / t / 1e12
     t_eager = bench(lambda: torch.cos(X @ W + b) * 0.1)
-    print(f"gemm_cos {M}x{K}x{N}: {t*1e3:.2f} ms = {tf:.1f} TF "
+    print(f"gemm_cos {M}x{K}x{N}: hybrid {t*1e3:.2f} ms, fullfused {t_full*1e3:.2f} ms = {tf:.1f} TF "
           f"(eager matmul+cos {t_eager*1e3:.2f} ms, "
           f"speedup {t_eager/t:.2f}x)")
 
