@@ -165,3 +165,11 @@ def test_fisher_vector_transformer_uses_kernel_on_gpu(hip):
     out = fv.apply_batch(X)
     ref = fv.apply_batch(X.cpu())
     assert (out.cpu() - ref).abs().max().item() < 5e-4
+
+
+def test_bias_cos_matches_torch(hip):
+    X = torch.randn(130, 777, device="cuda")
+    b = torch.rand(777, device="cuda")
+    ref = torch.cos(X + b) * 0.5
+    out = hip.bias_cos_(X.clone(), b, 0.5)
+    assert (out - ref).abs().max().item() < 1e-4
