@@ -117,16 +117,28 @@ def _as_block_getters(data: Dataset, block_size: int):
 
 
 class BlockLeastSquaresEstimator(LabelEstimator):
-    """ref: BlockLeastSquaresEstimator(blockSize, numIter, lambda)."""
+    """ref: BlockLeastSquaresEstimator(blockSize, numIter, lambda).
+
+    ``checkpoint_dir`` enables epoch-boundary solver checkpointing (aux
+    subsystem): each rank writes (epoch, W blocks, residual)
+    after every epoch and a re-run with the same dir resumes from the last
+    completed epoch — failure recovery without Spark lineage."""
 
     def __init__(self, block_size: int = 4096, num_iter: int = 10,
-                 lam: float = 0.0, fit_intercept: bool = True):
+                 lam: float = 0.0, fit_intercept: bool = True,
+                 checkpoint_dir: Optional[str] = None):
         self.block_size = block_size
         self.num_iter = num_iter
         self.lam = lam
         self.fit_intercept = fit_intercept
+        self.checkpoint_dir = checkpoint_dir
+
+    def _ckpt_path(self, rank: int) -> str:
+        import os
+        return os.path.join(self.checkpoint_dir, f"bls_rank{rank}.pt")
 
     def _fit(self, data: Dataset, labels: TensorDataset) -> BlockLinearMapper:
+        import os
         getters, rt, n_local = _as_block_getters(data, self.block_size)
         Y = labels.data
         if Y.dim() == 1:
@@ -148,7 +160,14 @@ class BlockLeastSquaresEstimator(LabelEstimator):
         nb = len(getters)
         Ws = [None] * nb
         R = Y.clone()  # residual Y - X W, local shard rows
-        for _epoch in range(self.num_iter):
+        start_epoch = 0
+        if self.checkpoint_dir and os.path.exists(self._ckpt_path(rt.rank)):
+            st = torch.load(self._ckpt_path(rt.rank), map_location=Y.device)
+            start_epoch = st["epoch"]
+            Ws = [w.to(Y.device) if w is not None else None
+                  for w in st["Ws"]]
+            R = st["R"].to(Y.device)
+        for _epoch in range(start_epoch, self.num_iter):
             for b in range(nb):
                 Xb = getters[b]()
                 if x_means is not None:
@@ -161,6 +180,15 @@ class BlockLeastSquaresEstimator(LabelEstimator):
                 Wb = _solve_spd(gram, rhs, self.lam)
                 R = T - Xb @ Wb
                 Ws[b] = Wb
+            if self.checkpoint_dir:
+                import os
+                os.makedirs(self.checkpoint_dir, exist_ok=True)
+                tmp = self._ckpt_path(rt.rank) + ".tmp"
+                torch.save({"epoch": _epoch + 1,
+                            "Ws": [w.cpu() if w is not None else None
+                                   for w in Ws],
+                            "R": R.cpu()}, tmp)
+                os.replace(tmp, self._ckpt_path(rt.rank))
 
         intercept = None
         if self.fit_intercept:

@@ -1,0 +1,47 @@
+"""Runtime configuration (config/flag system).
+
+The reference used per-app scopt configs + Spark conf; here each app has a
+dataclass config (models/*.py) and the runtime knobs live in ONE
+``RuntimeConfig`` populated from ``KEYSTONE_*`` environment variables:
+
+  KEYSTONE_HBM_BUDGET_GB   memory budget the optimizer plans against
+  KEYSTONE_BLOCK_SIZE      default solver block size
+  KEYSTONE_TRACE           JSONL trace output path (utils.tracing)
+  KEYSTONE_ALLOW_EAGER     permit eager fallback where a HIP kernel exists
+  KEYSTONE_COLL_TIMEOUT_S  RCCL/gloo collective timeout (abort cleanly)
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+
+
+@dataclass
+class RuntimeConfig:
+    hbm_budget_gb: float = 288.0
+    block_size: int = 4096
+    trace_path: str | None = None
+    allow_eager: bool = False
+    collective_timeout_s: int = 600
+
+    @classmethod
+    def from_env(cls) -> "RuntimeConfig":
+        return cls(
+            hbm_budget_gb=float(os.environ.get("KEYSTONE_HBM_BUDGET_GB", 288.0)),
+            block_size=int(os.environ.get("KEYSTONE_BLOCK_SIZE", 4096)),
+            trace_path=os.environ.get("KEYSTONE_TRACE"),
+            allow_eager=bool(os.environ.get("KEYSTONE_ALLOW_EAGER")),
+            collective_timeout_s=int(os.environ.get("KEYSTONE_COLL_TIMEOUT_S",
+                                                    600)),
+        )
+
+
+_CONFIG: RuntimeConfig | None = None
+
+
+def get_config() -> RuntimeConfig:
+    global _CONFIG
+    if _CONFIG is None:
+        _CONFIG = RuntimeConfig.from_env()
+    return _CONFIG

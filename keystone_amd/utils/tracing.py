@@ -43,6 +43,27 @@ def trace_event(kind: str, **fields) -> None:
         _trace_file.flush()
 
 
+def export_chrome_trace(jsonl_path: str, out_path: str) -> None:
+    """Convert a JSONL op trace into chrome://tracing format."""
+    events = []
+    with open(jsonl_path) as f:
+        for line in f:
+            rec = json.loads(line)
+            if rec.get("kind") != "op":
+                continue
+            events.append({
+                "name": rec.get("name", "op"),
+                "ph": "X",
+                "ts": rec["t"] * 1e6,
+                "dur": rec.get("seconds", 0) * 1e6,
+                "pid": 0,
+                "tid": 0,
+                "args": {"bytes": rec.get("bytes", 0)},
+            })
+    with open(out_path, "w") as f:
+        json.dump({"traceEvents": events}, f)
+
+
 # auto-enable from env
 if os.environ.get("KEYSTONE_TRACE"):
     enable_trace()

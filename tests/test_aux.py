@@ -1,0 +1,97 @@
+"""Aux subsystem tests: checkpoint/resume, determinism,
+tracing, config, CLI plumbing."""
+
+import json
+import os
+import subprocess
+import sys
+
+import torch
+
+import keystone_amd as ka
+from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+from keystone_amd.utils.determinism import model_fingerprint
+from keystone_amd.utils import tracing
+
+torch.manual_seed(0)
+
+
+def test_solver_checkpoint_resume(tmp_path):
+    X = torch.randn(100, 12, dtype=torch.float64)
+    Y = torch.randn(100, 2, dtype=torch.float64)
+    ds, ys = ka.from_rows(X), ka.from_rows(Y)
+    # full run: 6 epochs straight
+    full = BlockLeastSquaresEstimator(4, 6, 0.1).fit(ds, ys)
+    # interrupted run: 3 epochs with checkpointing, then resume to 6
+    ck = str(tmp_path / "ck")
+    BlockLeastSquaresEstimator(4, 3, 0.1, checkpoint_dir=ck).fit(ds, ys)
+    assert os.path.exists(os.path.join(ck, "bls_rank0.pt"))
+    resumed = BlockLeastSquaresEstimator(4, 6, 0.1, checkpoint_dir=ck).fit(ds, ys)
+    for a, b in zip(full.blocks, resumed.blocks):
+        assert torch.allclose(a, b, atol=1e-12)
+
+
+def test_model_fingerprint_deterministic():
+    X = torch.randn(50, 8)
+    Y = torch.randn(50, 2)
+    from keystone_amd.ops.learning import LinearMapEstimator
+    m1 = LinearMapEstimator(0.1).fit(ka.from_rows(X), ka.from_rows(Y))
+    m2 = LinearMapEstimator(0.1).fit(ka.from_rows(X), ka.from_rows(Y))
+    assert model_fingerprint(m1) == model_fingerprint(m2)
+    m3 = LinearMapEstimator(0.2).fit(ka.from_rows(X), ka.from_rows(Y))
+    assert model_fingerprint(m1) != model_fingerprint(m3)
+
+
+def test_refit_determinism_pipeline():
+    """Same seed => bitwise-identical fitted pipeline (re-run check)."""
+    from keystone_amd.ops.stats import CosineRandomFeatures
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    X = torch.randn(60, 10)
+    Y = torch.randn(60, 3)
+    ds, ys = ka.from_rows(X), ka.from_rows(Y)
+
+    def fit_once():
+        pipe = (CosineRandomFeatures(10, 64, 0.5, seed=7)
+                .and_then(BlockLeastSquaresEstimator(16, 3, 1e-3), ds, ys))
+        return pipe.fit()
+
+    f1, f2 = fit_once(), fit_once()
+    fps1 = [model_fingerprint(t) for t in f1.transformers()]
+    fps2 = [model_fingerprint(t) for t in f2.transformers()]
+    assert sorted(fps1) == sorted(fps2)
+
+
+def test_trace_jsonl_and_chrome(tmp_path):
+    trace_file = str(tmp_path / "trace.jsonl")
+    tracing.enable_trace(trace_file)
+    try:
+        from keystone_amd.ops.stats import NormalizeRows
+        ds = ka.from_rows(torch.randn(5, 4))
+        NormalizeRows().and_then(ka.Identity()).apply(ds).get()
+    finally:
+        tracing.disable_trace()
+    lines = [json.loads(l) for l in open(trace_file)]
+    assert any(l["kind"] == "op" for l in lines)
+    chrome = str(tmp_path / "trace.json")
+    tracing.export_chrome_trace(trace_file, chrome)
+    data = json.load(open(chrome))
+    assert data["traceEvents"]
+
+
+def test_runtime_config_env(monkeypatch):
+    from keystone_amd.utils import config
+    monkeypatch.setenv("KEYSTONE_BLOCK_SIZE", "1234")
+    monkeypatch.setenv("KEYSTONE_HBM_BUDGET_GB", "100.5")
+    cfg = config.RuntimeConfig.from_env()
+    assert cfg.block_size == 1234
+    assert cfg.hbm_budget_gb == 100.5
+
+
+def test_cli_mnist_smoke():
+    out = subprocess.run(
+        [sys.executable, "bin/run_pipeline.py", "mnist", "--n", "300",
+         "--num-features", "256"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "accuracy" in out.stdout
