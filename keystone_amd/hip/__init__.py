@@ -99,9 +99,12 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
-        lib.ks_sift_f32.argtypes = [c_p, c_p, c_i, c_i, c_i, c_i, c_i, c_i,
-                                    c_p]
-        lib.ks_sift_f32.restype = c_i
+        lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
+                                         c_p]
+        lib.ks_sift_omap_f32.restype = c_i
+        lib.ks_sift_norm_f32.argtypes = [c_p, ctypes.c_long, c_i, c_f, c_f,
+                                         c_p]
+        lib.ks_sift_norm_f32.restype = c_i
     except AttributeError:
         pass
 
@@ -255,6 +258,45 @@ def fisher_vector(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
         ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         B, n, D, K, _stream()), "ks_fisher_vector_f32")
     return out
+
+
+_gauss_cache = {}
+
+
+def sift_omap(imgs: torch.Tensor, sigma: float) -> torch.Tensor:
+    """Fused blur+gradient+orientation-binning: [B, H, W] -> [B, 8, H, W]."""
+    import math as _m
+    lib = _load()
+    imgs = _prep(imgs, "sift_omap.imgs")
+    B, H, W = imgs.shape
+    key = (round(sigma, 6), str(imgs.device))
+    gk = _gauss_cache.get(key)
+    if gk is None:
+        radius = max(1, int(_m.ceil(3.0 * sigma)))
+        radius = min(radius, 12)
+        x = torch.arange(-radius, radius + 1, dtype=torch.float32)
+        k = torch.exp(-0.5 * (x / sigma) ** 2)
+        gk = (k / k.sum()).to(imgs.device)
+        _gauss_cache[key] = gk
+    radius = (gk.numel() - 1) // 2
+    out = torch.empty(B, 8, H, W, dtype=torch.float32, device=imgs.device)
+    _check(lib.ks_sift_omap_f32(
+        ctypes.c_void_p(imgs.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_void_p(gk.data_ptr()), radius, B, H, W, _stream()),
+        "ks_sift_omap_f32")
+    return out
+
+
+def sift_norm_(D: torch.Tensor, clamp: float = 0.2,
+               eps: float = 1e-20) -> torch.Tensor:
+    """In-place SIFT descriptor normalization (L2 -> clamp -> L2) per row."""
+    lib = _load()
+    D = _prep(D, "sift_norm.D")
+    rows, dim = D.shape
+    _check(lib.ks_sift_norm_f32(
+        ctypes.c_void_p(D.data_ptr()), ctypes.c_long(rows), dim, clamp, eps,
+        _stream()), "ks_sift_norm_f32")
+    return D
 
 
 def sym_rectify(X: torch.Tensor, alpha: float = 0.0) -> torch.Tensor:

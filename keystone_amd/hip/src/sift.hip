@@ -1,0 +1,150 @@
+// K1: dense-SIFT orientation-map kernel (gfx950).
+//
+// ref: VLFeat dense SIFT (JNI VLFeat.getSIFTs) — the
+// Gaussian-blur -> gradient -> soft orientation-binning front half, fused
+// into ONE kernel with LDS halo tiles:
+//
+//   omap[b, o, y, x] = mag(y,x) * softbin_o(angle(y,x))     (8 bins)
+//
+// where the image is first blurred with a separable Gaussian (replicate
+// boundary, radius r).  The eager torch chain for this stage is 2 banded
+// GEMMs + ~8 elementwise/scatter kernels re-reading HBM; this kernel
+// reads the image once and writes the 8-plane map once (the remaining
+// SIFT stages — triangular spatial pooling at the descriptor grid and
+// descriptor normalization — are sampling-banded GEMMs (hipBLASLt) and
+// the ks_sift_norm kernel).
+//
+// Tile: 16 rows x 64 cols of output per 256-thread block; LDS stages
+// raw -> h-blur -> v-blur with the halos each stage needs.
+
+#include "common.h"
+#include <math.h>
+
+#define OTH 16   // output tile rows
+#define OTW 64   // output tile cols
+#define MAXR 12  // max Gaussian radius
+
+__global__ __launch_bounds__(256)
+void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
+                      float* __restrict__ omap,       // [B, 8, H, W]
+                      const float* __restrict__ gk,   // [2r+1]
+                      int r, int H, int W) {
+    const int b = blockIdx.z;
+    const int oy0 = blockIdx.y * OTH;
+    const int ox0 = blockIdx.x * OTW;
+    const int tid = threadIdx.x;
+
+    // LDS geometry (worst case r=MAXR)
+    const int RH = OTH + 2 + 2 * r;          // raw/hblur rows
+    const int RW = OTW + 2 + 2 * r;          // raw cols
+    const int HW = OTW + 2;                  // hblur/vblur cols
+    __shared__ float raw[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * MAXR];
+    __shared__ float hb[OTH + 2 + 2 * MAXR][OTW + 2];
+    __shared__ float vb[OTH + 2][OTW + 2];
+
+    const float* im = img + (size_t)b * H * W;
+
+    // ---- load raw tile + halo (replicate boundary)
+    for (int i = tid; i < RH * RW; i += 256) {
+        int ry = i / RW, rx = i % RW;
+        int gy = oy0 - 1 - r + ry;
+        int gx = ox0 - 1 - r + rx;
+        gy = min(max(gy, 0), H - 1);
+        gx = min(max(gx, 0), W - 1);
+        raw[ry][rx] = im[(size_t)gy * W + gx];
+    }
+    __syncthreads();
+
+    // ---- horizontal blur: rows [0, RH), cols [0, OTW+2)
+    for (int i = tid; i < RH * HW; i += 256) {
+        int ry = i / HW, cx = i % HW;
+        float s = 0.0f;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry][cx + t];
+        hb[ry][cx] = s;
+    }
+    __syncthreads();
+
+    // ---- vertical blur: rows [0, OTH+2)
+    for (int i = tid; i < (OTH + 2) * HW; i += 256) {
+        int vy = i / HW, cx = i % HW;
+        float s = 0.0f;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[vy + t][cx];
+        vb[vy][cx] = s;
+    }
+    __syncthreads();
+
+    // ---- gradient + orientation soft-binning + 8-plane write
+    const float inv_step = 8.0f / (2.0f * (float)M_PI);
+    for (int i = tid; i < OTH * OTW; i += 256) {
+        int ty = i / OTW, tx = i % OTW;
+        const int gy = oy0 + ty, gx = ox0 + tx;
+        if (gy >= H || gx >= W) continue;
+        float gxv = 0.0f, gyv = 0.0f;
+        if (gx >= 1 && gx + 1 < W)
+            gxv = 0.5f * (vb[ty + 1][tx + 2] - vb[ty + 1][tx]);
+        if (gy >= 1 && gy + 1 < H)
+            gyv = 0.5f * (vb[ty + 2][tx + 1] - vb[ty][tx + 1]);
+        const float mag = sqrtf(gxv * gxv + gyv * gyv);
+        float ang = atan2f(gyv, gxv);
+        if (ang < 0.0f) ang += 2.0f * (float)M_PI;
+        float t = ang * inv_step;
+        int b0 = (int)t;
+        float frac = t - (float)b0;
+        b0 &= 7;
+        const int b1 = (b0 + 1) & 7;
+        float vals[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        vals[b0] = mag * (1.0f - frac);
+        vals[b1] = mag * frac;
+        float* ob = omap + ((size_t)b * 8) * H * W + (size_t)gy * W + gx;
+#pragma unroll
+        for (int o = 0; o < 8; ++o) ob[(size_t)o * H * W] = vals[o];
+    }
+}
+
+KS_EXPORT int ks_sift_omap_f32(const float* img, float* omap,
+                               const float* gauss, int radius, int B, int H,
+                               int W, hipStream_t stream) {
+    if (radius > MAXR || radius < 0) return (int)hipErrorInvalidValue;
+    dim3 grid(ks_ceil_div(W, OTW), ks_ceil_div(H, OTH), B);
+    hipLaunchKernelGGL(sift_omap_kernel, grid, dim3(256), 0, stream, img,
+                       omap, gauss, radius, H, W);
+    return (int)hipGetLastError();
+}
+
+// SIFT descriptor normalization: per row (128-dim), L2 -> clamp -> L2.
+__global__ __launch_bounds__(64)
+void sift_norm_kernel(float* __restrict__ D, long rows, int dim,
+                      float clampv, float eps) {
+    const long row = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+    if (row >= rows) return;
+    const int lane = threadIdx.x & 63;
+    float* d = D + row * dim;
+    float ss = 0.0f;
+    for (int c = lane; c < dim; c += 64) ss += d[c] * d[c];
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss += __shfl_down(ss, off, 64);
+    ss = __shfl(ss, 0, 64);
+    float inv = rsqrtf(fmaxf(ss, eps));
+    float ss2 = 0.0f;
+    for (int c = lane; c < dim; c += 64) {
+        float v = fminf(d[c] * inv, clampv);
+        d[c] = v;
+        ss2 += v * v;
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss2 += __shfl_down(ss2, off, 64);
+    ss2 = __shfl(ss2, 0, 64);
+    float inv2 = rsqrtf(fmaxf(ss2, eps));
+    for (int c = lane; c < dim; c += 64) d[c] *= inv2;
+}
+
+KS_EXPORT int ks_sift_norm_f32(float* D, long rows, int dim, float clampv,
+                               float eps, hipStream_t stream) {
+    if (rows == 0) return 0;
+    const int waves_per_block = 4;
+    long blocks = (rows + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(sift_norm_kernel, dim3((unsigned)blocks),
+                       dim3(64 * waves_per_block), 0, stream, D, rows, dim,
+                       clampv, eps);
+    return (int)hipGetLastError();
+}

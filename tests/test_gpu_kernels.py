@@ -173,3 +173,38 @@ def test_bias_cos_matches_torch(hip):
     ref = torch.cos(X + b) * 0.5
     out = hip.bias_cos_(X.clone(), b, 0.5)
     assert (out - ref).abs().max().item() < 1e-4
+
+
+def test_sift_omap_kernel_matches_torch(hip):
+    from keystone_amd.ops.images.sift import SIFTExtractor
+    torch.manual_seed(9)
+    imgs = torch.rand(3, 96, 80, device="cuda")
+    s = SIFTExtractor(step=4, bin_size=4, num_scales=1)
+    out = hip.sift_omap(imgs, sigma=4 / 3.0)
+    ref = s._orientation_maps(imgs.cpu(), 4 / 3.0)
+    err = (out.cpu() - ref).abs().max().item()
+    assert err < 2e-3, err  # atan2f / blur roundoff at bin boundaries
+
+
+def test_sift_descriptors_gpu_close_to_cpu(hip):
+    from keystone_amd.ops.images.sift import SIFTExtractor
+    torch.manual_seed(10)
+    imgs = torch.rand(2, 64, 64, device="cuda")
+    s = SIFTExtractor(step=6, bin_size=4, num_scales=2)
+    Dg = s.apply_batch(imgs)
+    Dc = s.apply_batch(imgs.cpu())
+    assert Dg.shape == Dc.shape
+    # descriptors are unit-normalized; small angle roundoffs shift a tiny
+    # fraction of mass between adjacent bins
+    diff = (Dg.cpu() - Dc).abs()
+    assert float(diff.mean()) < 2e-3
+    assert float(diff.max()) < 0.08
+
+
+def test_sift_norm_kernel(hip):
+    D = torch.rand(100, 128, device="cuda") * 2
+    ref = D / D.norm(dim=1, keepdim=True).clamp_min(1e-10)
+    ref = ref.clamp_max(0.2)
+    ref = ref / ref.norm(dim=1, keepdim=True).clamp_min(1e-10)
+    out = hip.sift_norm_(D.clone(), 0.2)
+    assert (out - ref).abs().max().item() < 1e-5
