@@ -112,7 +112,8 @@ KS_EXPORT int ks_sift_omap_f32(const float* img, float* omap,
 }
 
 // SIFT descriptor normalization: per row (128-dim), L2 -> clamp -> L2.
-__global__ __launch_bounds__(64)
+// 4 waves per block, one row per wave.
+__global__ __launch_bounds__(256)
 void sift_norm_kernel(float* __restrict__ D, long rows, int dim,
                       float clampv, float eps) {
     const long row = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
