@@ -130,8 +130,10 @@ class DistributedPCAEstimator(Estimator):
             mean = (s / n.item())
             data = TensorDataset((X.to(torch.float64) - mean).to(X.dtype), rt)
         R = TSQR.factor(data)
-        _U, _S, Vh = torch.linalg.svd(R)
-        V = _sign_fix(Vh.T[:, : self.dims]).to(X.dtype)
+        # d x d R factor is tiny: CPU SVD is microseconds and avoids
+        # sporadic multi-ms rocSOLVER gesvd launches
+        _U, _S, Vh = torch.linalg.svd(R.cpu())
+        V = _sign_fix(Vh.T[:, : self.dims]).to(X.dtype).to(X.device)
         return PCATransformer(V, mean.to(X.dtype) if mean is not None else None)
 
 
