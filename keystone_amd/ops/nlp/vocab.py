@@ -68,15 +68,19 @@ class SparseFeatureVectorizer(Transformer):
         return TensorDataset(csr.to(ds.runtime.device), ds.runtime)
 
     def get_config(self):
-        # keys may be NGram tuples; serialize as joined strings
-        return {"num_features": self.num_features,
-                "terms": [" ".join(t) if isinstance(t, tuple) else str(t)
-                          for t in self.index],
+        # keys may be NGram tuples (n-gram terms) or plain strings; encode
+        # tuples as lists so the load path rebuilds equal-hashing keys
+        terms = [list(t) if isinstance(t, tuple) else t for t in self.index]
+        return {"num_features": self.num_features, "terms": terms,
                 "indices": list(self.index.values())}
 
     @classmethod
     def from_saved(cls, config, state):
-        index = {t: i for t, i in zip(config["terms"], config["indices"])}
+        from .ngrams import NGram
+        index = {}
+        for t, i in zip(config["terms"], config["indices"]):
+            key = NGram(t) if isinstance(t, list) else t
+            index[key] = i
         return cls(index, config["num_features"])
 
 
