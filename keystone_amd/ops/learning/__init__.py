@@ -13,6 +13,7 @@ from .pca import (PCAEstimator, PCATransformer, DistributedPCAEstimator,
 from .whitening import ZCAWhitener, ZCAWhitenerEstimator, LinearDiscriminantAnalysis
 from .krr import (GaussianKernelGenerator, KernelRidgeRegression,
                   KernelBlockLinearMapper)
+from .kernel_matrix import KernelMatrix, BlockKernelMatrix
 
 __all__ = [
     "LinearMapper", "LinearMapEstimator", "LocalLeastSquaresEstimator",
@@ -28,5 +29,5 @@ __all__ = [
     "ApproximatePCAEstimator", "TSQR",
     "ZCAWhitener", "ZCAWhitenerEstimator", "LinearDiscriminantAnalysis",
     "GaussianKernelGenerator", "KernelRidgeRegression",
-    "KernelBlockLinearMapper",
+    "KernelBlockLinearMapper", "KernelMatrix", "BlockKernelMatrix",
 ]

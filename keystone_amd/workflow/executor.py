@@ -94,9 +94,17 @@ class GraphExecutor:
             from .optimizer import Optimizable
             if isinstance(est, Optimizable):
                 est = est.optimize(*inputs)
-            if len(inputs) == 1:
-                return est.fit(inputs[0])
-            return est.fit(inputs[0], inputs[1])
+            # fitted-state reuse (ref: SavedStateLoadRule): key by estimator
+            # class + hyperparameters when a state cache is enabled
+            from .state_cache import save_fitted, state_cache_dir, try_load
+            if state_cache_dir() is not None:
+                cached = try_load(est)
+                if cached is not None:
+                    return cached
+            fitted = est.fit(*inputs[:2])
+            if state_cache_dir() is not None:
+                save_fitted(est, fitted)
+            return fitted
         raise RuntimeError(f"unknown operator {op!r}")
 
     @staticmethod

@@ -95,3 +95,46 @@ def test_cli_mnist_smoke():
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert out.returncode == 0, out.stderr[-2000:]
     assert "accuracy" in out.stdout
+
+
+def test_state_cache_reuses_fit(tmp_path):
+    from keystone_amd.workflow import state_cache
+    from keystone_amd.ops.learning import LinearMapEstimator
+
+    calls = []
+
+    class CountingEst(LinearMapEstimator):
+        def _fit(self, data, labels):
+            calls.append(1)
+            return super()._fit(data, labels)
+
+    X = torch.randn(30, 5)
+    Y = torch.randn(30, 2)
+    ds, ys = ka.from_rows(X), ka.from_rows(Y)
+    state_cache.enable_state_cache(str(tmp_path / "sc"))
+    try:
+        p1 = ka.Identity().and_then(CountingEst(0.1), ds, ys).fit()
+        p2 = ka.Identity().and_then(CountingEst(0.1), ds, ys).fit()
+        assert len(calls) == 1  # second fit loaded from the state cache
+        a = p1.apply(ds).get().data
+        b = p2.apply(ds).get().data
+        assert torch.allclose(a, b, atol=1e-6)
+    finally:
+        state_cache.disable_state_cache()
+
+
+def test_shuffler_preserves_rows():
+    from keystone_amd.ops.util import Shuffler
+    X = torch.arange(20, dtype=torch.float32).reshape(10, 2)
+    out = Shuffler(seed=1).apply_dataset(ka.from_rows(X))
+    assert out.data.shape == X.shape
+    assert torch.equal(out.data.sum(0), X.sum(0))
+
+
+def test_kernel_matrix_blocks():
+    from keystone_amd.ops.learning import BlockKernelMatrix, GaussianKernelGenerator
+    X = torch.randn(30, 4)
+    km = BlockKernelMatrix(GaussianKernelGenerator(0.5), ka.from_rows(X), 8)
+    full = GaussianKernelGenerator(0.5)(X, X)
+    got = torch.cat([blk for _lo, _hi, blk in km], dim=0)
+    assert torch.allclose(got, full, atol=1e-6)
