@@ -67,12 +67,22 @@ class Runtime:
 
     # --------------------------------------------------------- collectives
     # Spark treeReduce(_+_) / treeAggregate  ->  RCCL all-reduce over xGMI.
+    # RCCL only moves GPU tensors: small CPU-side metadata tensors (counts,
+    # shapes) are bounced through HBM transparently on the nccl backend.
+    def _coll_dev(self, t: torch.Tensor) -> torch.Tensor:
+        if self.backend == "nccl" and not t.is_cuda:
+            return t.to(self.device)
+        return t
+
     def all_reduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         """In-place sum/max/min across ranks. No-op when single-process."""
         if self.is_distributed:
             red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
                    "min": dist.ReduceOp.MIN}[op]
-            dist.all_reduce(t, op=red)
+            tc = self._coll_dev(t)
+            dist.all_reduce(tc, op=red)
+            if tc is not t:
+                t.copy_(tc.cpu())
         return t
 
     def all_reduce_many_(self, tensors: List[torch.Tensor], op: str = "sum") -> None:
@@ -80,19 +90,24 @@ class Runtime:
         call; stripes better across the 7 xGMI links than many small calls)."""
         if not self.is_distributed or not tensors:
             return
-        flat = torch.cat([t.reshape(-1) for t in tensors])
+        dev = tensors[0].device
+        flat = torch.cat([t.reshape(-1).to(dev) for t in tensors])
         self.all_reduce_(flat, op)
+        flat = flat.to(dev)
         off = 0
         for t in tensors:
             n = t.numel()
-            t.copy_(flat[off:off + n].view_as(t))
+            t.copy_(flat[off:off + n].view_as(t).to(t.device))
             off += n
 
     def all_gather(self, t: torch.Tensor) -> List[torch.Tensor]:
         if not self.is_distributed:
             return [t]
-        outs = [torch.empty_like(t) for _ in range(self.world_size)]
-        dist.all_gather(outs, t.contiguous())
+        tc = self._coll_dev(t.contiguous())
+        outs = [torch.empty_like(tc) for _ in range(self.world_size)]
+        dist.all_gather(outs, tc)
+        if tc is not t and not t.is_cuda:
+            outs = [o.cpu() for o in outs]
         return outs
 
     def all_gather_rows(self, t: torch.Tensor) -> torch.Tensor:
@@ -100,9 +115,7 @@ class Runtime:
         if not self.is_distributed:
             return t
         n_local = torch.tensor([t.shape[0]], dtype=torch.int64)
-        counts = [torch.zeros(1, dtype=torch.int64) for _ in range(self.world_size)]
-        dist.all_gather(counts, n_local)
-        counts = [int(c.item()) for c in counts]
+        counts = [int(c.item()) for c in self.all_gather(n_local)]
         n_max = max(counts)
         pad = t
         if t.shape[0] < n_max:
@@ -135,8 +148,11 @@ class Runtime:
             dist.all_gather_object(gathered, [t.cpu() for t in send])
             return [gathered[src][self.rank].to(send[0].device)
                     for src in range(self.world_size)]
-        sizes = [torch.tensor([s.shape[0]], dtype=torch.int64) for s in send]
-        recv_sizes = [torch.zeros(1, dtype=torch.int64) for _ in range(self.world_size)]
+        sdev = self.device if self.backend == "nccl" else torch.device("cpu")
+        sizes = [torch.tensor([s.shape[0]], dtype=torch.int64, device=sdev)
+                 for s in send]
+        recv_sizes = [torch.zeros(1, dtype=torch.int64, device=sdev)
+                      for _ in range(self.world_size)]
         dist.all_to_all(recv_sizes, sizes)
         trailing = send[0].shape[1:]
         recv = [torch.empty((int(n.item()),) + trailing, dtype=send[0].dtype,
