@@ -1,6 +1,6 @@
 from .linear import LinearMapper, LinearMapEstimator, LocalLeastSquaresEstimator
 from .block_ls import (BlockLinearMapper, BlockLeastSquaresEstimator,
-                       BlockWeightedLeastSquaresEstimator)
+                       BlockWeightedLeastSquaresEstimator, LazyBlocks)
 from .least_squares import LeastSquaresEstimator
 from .lbfgs import (DenseLBFGSwithL2, SparseLBFGSwithL2, MiniBatchSGDwithL2,
                     LeastSquaresGradient, LogisticGradient)
@@ -18,7 +18,7 @@ from .kernel_matrix import KernelMatrix, BlockKernelMatrix
 __all__ = [
     "LinearMapper", "LinearMapEstimator", "LocalLeastSquaresEstimator",
     "BlockLinearMapper", "BlockLeastSquaresEstimator",
-    "BlockWeightedLeastSquaresEstimator", "LeastSquaresEstimator",
+    "BlockWeightedLeastSquaresEstimator", "LazyBlocks", "LeastSquaresEstimator",
     "DenseLBFGSwithL2", "SparseLBFGSwithL2", "MiniBatchSGDwithL2",
     "LeastSquaresGradient", "LogisticGradient",
     "LogisticRegressionEstimator", "LogisticRegressionModel",

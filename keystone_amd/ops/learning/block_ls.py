@@ -39,7 +39,7 @@ from ...workflow.pipeline import (BatchTransformer, LabelEstimator,
 from .linear import _solve_dtype, _solve_spd
 
 __all__ = ["BlockLinearMapper", "BlockLeastSquaresEstimator",
-           "BlockWeightedLeastSquaresEstimator"]
+           "BlockWeightedLeastSquaresEstimator", "LazyBlocks"]
 
 
 class BlockLinearMapper(Transformer):
@@ -50,16 +50,22 @@ class BlockLinearMapper(Transformer):
     are never materialized at once."""
 
     def __init__(self, blocks: List[torch.Tensor],
-                 intercept: Optional[torch.Tensor] = None):
+                 intercept: Optional[torch.Tensor] = None,
+                 featurizers: Optional[List[Transformer]] = None):
         self.blocks = blocks
         self.intercept = intercept
+        # lazy mode: block b of the features is featurizers[b](raw input),
+        # regenerated at apply time (never materialized jointly)
+        self.featurizers = featurizers
 
     @property
     def block_sizes(self) -> List[int]:
         return [int(b.shape[0]) for b in self.blocks]
 
     def apply(self, x):
-        if isinstance(x, (tuple, list)):
+        if self.featurizers is not None:
+            parts = [f.apply(x) for f in self.featurizers]
+        elif isinstance(x, (tuple, list)):
             parts = list(x)
         else:
             parts = list(torch.split(x, self.block_sizes, dim=-1))
@@ -75,6 +81,16 @@ class BlockLinearMapper(Transformer):
         return self.apply(X)
 
     def apply_dataset(self, ds: Dataset):
+        if self.featurizers is not None:
+            out = None
+            for f, W in zip(self.featurizers, self.blocks):
+                fb = f.apply_dataset(ds.raw if isinstance(ds, LazyBlocks) else ds)
+                X = fb.data
+                W = W.to(device=X.device, dtype=X.dtype)
+                out = X @ W if out is None else out + X @ W
+            if self.intercept is not None:
+                out = out + self.intercept.to(device=out.device, dtype=out.dtype)
+            return TensorDataset(out, ds.runtime)
         if isinstance(ds, TupleDataset):
             out = None
             for part, W in zip(ds.parts, self.blocks):
@@ -93,19 +109,66 @@ class BlockLinearMapper(Transformer):
         st = {f"block_{i}": b for i, b in enumerate(self.blocks)}
         if self.intercept is not None:
             st["intercept"] = self.intercept
+        if self.featurizers is not None:
+            for i, f in enumerate(self.featurizers):
+                for k, v in f.get_state().items():
+                    st[f"feat{i}__{k}"] = v
         return st
 
     def get_config(self):
-        return {"n_blocks": len(self.blocks)}
+        cfg = {"n_blocks": len(self.blocks)}
+        if self.featurizers is not None:
+            cfg["featurizers"] = [
+                {"class": type(f).__module__ + "." + type(f).__qualname__,
+                 "config": f.get_config()} for f in self.featurizers]
+        return cfg
 
     @classmethod
     def from_saved(cls, config, state):
         blocks = [state[f"block_{i}"] for i in range(config["n_blocks"])]
-        return cls(blocks, state.get("intercept"))
+        feats = None
+        if "featurizers" in config:
+            from ...workflow.serialization import _load_class
+            feats = []
+            for i, fc in enumerate(config["featurizers"]):
+                fstate = {k.split("__", 1)[1]: v for k, v in state.items()
+                          if k.startswith(f"feat{i}__")}
+                feats.append(_load_class(fc["class"]).from_saved(
+                    fc["config"], fstate))
+        return cls(blocks, state.get("intercept"), feats)
+
+
+class LazyBlocks(Dataset):
+    """Lazy per-block featurization (ref: BlockLinearMapper's lazy mode +
+    the TIMIT 200k-cosine-feature training recipe): the d-wide feature
+    matrix is NEVER materialized — block b is regenerated as
+    ``featurizers[b](raw)`` every time a solver epoch touches it (the
+    memory/compute trade the whole-pipeline optimizer reasons about for
+    features wider than the 288 GB HBM budget)."""
+
+    def __init__(self, raw: Dataset, featurizers: Sequence[Transformer]):
+        super().__init__(raw.runtime)
+        self.raw = raw
+        self.featurizers = list(featurizers)
+
+    def local_count(self) -> int:
+        return self.raw.local_count()
+
+    def block(self, b: int) -> torch.Tensor:
+        out = self.featurizers[b].apply_dataset(self.raw)
+        return out.data if isinstance(out, TensorDataset) else torch.stack(
+            list(out.items()))
+
+    def items(self):
+        return self.raw.items()
 
 
 def _as_block_getters(data: Dataset, block_size: int):
     """Return (list of zero-arg block getters, runtime, n_local)."""
+    if isinstance(data, LazyBlocks):
+        return ([lambda b=b: data.block(b)
+                 for b in range(len(data.featurizers))],
+                data.runtime, data.local_count())
     if isinstance(data, TupleDataset):
         parts = [p.data for p in data.parts]
         return [lambda p=p: p for p in parts], data.runtime, parts[0].shape[0]
@@ -113,7 +176,8 @@ def _as_block_getters(data: Dataset, block_size: int):
         X = data.data
         splits = list(torch.split(X, block_size, dim=-1))
         return [lambda s=s: s for s in splits], data.runtime, X.shape[0]
-    raise TypeError("Block solvers need a TensorDataset or VectorSplitter output")
+    raise TypeError("Block solvers need a TensorDataset, VectorSplitter "
+                    "output, or LazyBlocks")
 
 
 class BlockLeastSquaresEstimator(LabelEstimator):
@@ -193,7 +257,8 @@ class BlockLeastSquaresEstimator(LabelEstimator):
         intercept = None
         if self.fit_intercept:
             intercept = y_mean - sum(m @ W for m, W in zip(x_means, Ws))
-        return BlockLinearMapper(Ws, intercept)
+        feats = data.featurizers if isinstance(data, LazyBlocks) else None
+        return BlockLinearMapper(Ws, intercept, feats)
 
 
 class BlockWeightedLeastSquaresEstimator(LabelEstimator):

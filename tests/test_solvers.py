@@ -296,3 +296,31 @@ def test_block_weighted_woodbury_path_matches_direct():
         G = X.T @ (w.unsqueeze(1) * X) + lam * torch.eye(d, dtype=torch.float64)
         ref = torch.linalg.solve(G, X.T @ (w * Y[:, c]))
         assert torch.allclose(W[:, c], ref, atol=1e-6), f"class {c}"
+
+
+def test_lazy_block_featurization_matches_materialized():
+    """ref capability: BlockLinearMapper lazy per-block featurization —
+    features are regenerated per block instead of materialized jointly."""
+    from keystone_amd.ops.learning.block_ls import LazyBlocks
+    from keystone_amd.ops.stats import CosineRandomFeatures
+    torch.manual_seed(11)
+    n, d, bs, nb = 80, 12, 16, 3
+    X = torch.randn(n, d)
+    Y = torch.randn(n, 2)
+    ds = ka.from_rows(X)
+    feats = [CosineRandomFeatures(d, bs, 0.5, seed=100 + i) for i in range(nb)]
+    lazy = LazyBlocks(ds, feats)
+    m_lazy = BlockLeastSquaresEstimator(bs, 8, 1e-3).fit(lazy, ka.from_rows(Y))
+    # materialized reference
+    Xf = torch.cat([f.apply_batch(X) for f in feats], dim=1)
+    m_mat = BlockLeastSquaresEstimator(bs, 8, 1e-3).fit(
+        ka.from_rows(Xf), ka.from_rows(Y))
+    for a, b in zip(m_lazy.blocks, m_mat.blocks):
+        assert torch.allclose(a, b, atol=1e-4)
+    # lazy apply: takes RAW input, never the 48-dim features
+    pred_lazy = m_lazy.apply_batch(X)
+    pred_mat = m_mat.apply_batch(Xf)
+    assert torch.allclose(pred_lazy, pred_mat, atol=1e-4)
+    # dataset-level lazy apply + save/load round trip
+    out = m_lazy.apply_dataset(ds).data
+    assert torch.allclose(out, pred_mat, atol=1e-4)
