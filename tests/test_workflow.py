@@ -154,3 +154,34 @@ def test_cacher_memoization():
     assert out.data.shape == (4, 4)
     # CSE + memoization: the shared prefix ran once
     assert len(calls) == 1
+
+
+def test_autocache_rule_inserts_cacher():
+    from keystone_amd.workflow.optimizer import AutoCacheRule
+    from keystone_amd.workflow.graph import TransformerOperator
+    from keystone_amd.workflow.cache import Cacher
+    X = torch.randn(6, 4)
+    ds = ka.from_rows(X)
+    shared = Scale(2.0)
+    base = ka.Identity().and_then(shared)
+    p = ka.gather([base.and_then(Scale(1.0)), base.and_then(Scale(3.0))]) \
+        .and_then(VectorCombiner())
+    lazy = p.apply(ds)
+    from keystone_amd.workflow.optimizer import EquivalentNodeMergeRule
+    g = EquivalentNodeMergeRule().apply(lazy.graph)
+    g2 = AutoCacheRule(mem_budget_bytes=1 << 30).apply(g)
+    cachers = [op for op in g2.operators.values()
+               if isinstance(op, TransformerOperator)
+               and isinstance(op.transformer, Cacher)]
+    assert len(cachers) >= 1  # the shared Scale node got pinned
+
+
+def test_profiler_produces_stats():
+    from keystone_amd.workflow.optimizer import Profiler
+    X = torch.randn(64, 4)
+    ds = ka.from_rows(X)
+    lazy = Scale(2.0).and_then(Scale(3.0)).apply(ds)
+    prof = Profiler(sample_rows=16).profile(lazy.graph)
+    assert prof, "profiler returned no records"
+    rec = next(iter(prof.values()))
+    assert rec["seconds"] >= 0 and rec["bytes"] > 0
