@@ -204,8 +204,14 @@ def init_runtime(device: Optional[str] = None, backend: Optional[str] = None,
         if not dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29571")
+            kwargs = {}
+            if backend == "nccl" and use_gpu:
+                # bind the communicator to this rank's device up front
+                # (avoids device-guess hangs on heterogeneous mappings)
+                kwargs["device_id"] = dev
             dist.init_process_group(backend=backend, rank=rank, world_size=world,
-                                    timeout=datetime.timedelta(seconds=timeout_s))
+                                    timeout=datetime.timedelta(seconds=timeout_s),
+                                    **kwargs)
     else:
         backend = None
 
