@@ -36,7 +36,7 @@ def parse_args():
     p.add_argument("--images-per-gpu", type=int, default=2048)
     p.add_argument("--image-size", type=int, default=256)
     p.add_argument("--num-classes", type=int, default=1000)
-    p.add_argument("--gmm-k", type=int, default=64)
+    p.add_argument("--gmm-k", type=int, default=256)
     p.add_argument("--pca-dims", type=int, default=64)
     p.add_argument("--sift-step", type=int, default=6)
     p.add_argument("--solver-epochs", type=int, default=1)
