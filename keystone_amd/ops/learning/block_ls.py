@@ -362,6 +362,12 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     B64 += self.lam * torch.eye(db, dtype=sdt,
                                                 device=Xb.device)
                     L = torch.linalg.cholesky(B64)
+                    # explicit inverse: the downstream B^{-1}-applications
+                    # become GEMMs (hipBLASLt ~150 TF) instead of TRSMs
+                    # (rocSOLVER trsm is several x slower at these shapes);
+                    # numerically fine here — lam-regularized SPD base and
+                    # the BCD outer loop re-derives residuals exactly.
+                    Binv = torch.cholesky_inverse(L)
                     Xb64 = Xb.to(sdt)
                     wcs = {c: alpha / float(counts_cpu[c]) for c in own}
                     # rhs for ALL classes in two fused passes: the class-c
@@ -375,7 +381,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                         dtype=sdt, device=Xb.device)
                     rhs = base_w * xtr_all[:, own].to(sdt) \
                         + seg[own].T * wc_t
-                    RB = torch.cholesky_solve(rhs, L)        # B^{-1} rhs
+                    RB = Binv @ rhs                          # B^{-1} rhs
                     col = {c: j for j, c in enumerate(own)}
 
                     by_nc: dict = {}
@@ -398,7 +404,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                                 (wcs[c] ** 0.5)
                                 * Xb64[cls_slice[c][0]:cls_slice[c][1]].T
                                 for c in cls_list], dim=1)       # [db, G*nc]
-                            UBf = torch.cholesky_solve(U, L)     # [db, G*nc]
+                            UBf = Binv @ U                       # [db, G*nc]
                             Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
                             UBb = UBf.T.reshape(G_n, n_c, db)
                             S = torch.eye(n_c, dtype=sdt,

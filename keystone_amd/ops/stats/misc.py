@@ -116,10 +116,14 @@ class ColumnSampler(Transformer):
 
     def apply_dataset(self, ds: Dataset) -> TensorDataset:
         per_rank = max(1, self.num_samples // max(1, ds.runtime.world_size))
-        mats = [x for x in ds.items()]
-        if not mats:
-            raise ValueError("ColumnSampler: empty local shard")
-        all_rows = torch.cat([m.reshape(-1, m.shape[-1]) for m in mats], dim=0)
+        if isinstance(ds, TensorDataset) and ds.data.dim() >= 2:
+            all_rows = ds.data.reshape(-1, ds.data.shape[-1])
+        else:
+            mats = [x for x in ds.items()]
+            if not mats:
+                raise ValueError("ColumnSampler: empty local shard")
+            all_rows = torch.cat([m.reshape(-1, m.shape[-1]) for m in mats],
+                                 dim=0)
         n = all_rows.shape[0]
         g = torch.Generator().manual_seed(self.seed + 7919 * ds.runtime.rank)
         idx = torch.randperm(n, generator=g)[:per_rank].to(all_rows.device)
