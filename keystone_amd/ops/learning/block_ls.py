@@ -396,14 +396,20 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             continue
                         if n_c <= self.woodbury_threshold * db:
                             G_n = len(cls_list)
-                            # flat [db, G*nc] solve against the single L
-                            # (broadcasted batched cholesky_solve is
-                            # unreliable on ROCm), then per-class algebra
-                            # as batched GEMMs on views.
-                            U = torch.cat([
-                                (wcs[c] ** 0.5)
-                                * Xb64[cls_slice[c][0]:cls_slice[c][1]].T
-                                for c in cls_list], dim=1)       # [db, G*nc]
+                            # one vectorized gather builds every class's
+                            # sqrt(w_c)-scaled rows (a per-class cat loop
+                            # costs ~1000 kernel launches per block)
+                            s_t = torch.tensor(
+                                [cls_slice[c][0] for c in cls_list],
+                                device=Xb.device)
+                            row_idx = (s_t.unsqueeze(1)
+                                       + torch.arange(n_c, device=Xb.device)
+                                       ).reshape(-1)            # [G*nc]
+                            w_row = torch.tensor(
+                                [wcs[c] for c in cls_list], dtype=sdt,
+                                device=Xb.device).sqrt().repeat_interleave(n_c)
+                            U = (Xb64[row_idx] * w_row.unsqueeze(1)).T
+                            # [db, G*nc], class-major blocks of nc columns
                             UBf = Binv @ U                       # [db, G*nc]
                             Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
                             UBb = UBf.T.reshape(G_n, n_c, db)

@@ -126,5 +126,10 @@ class ColumnSampler(Transformer):
                                  dim=0)
         n = all_rows.shape[0]
         g = torch.Generator().manual_seed(self.seed + 7919 * ds.runtime.rank)
-        idx = torch.randperm(n, generator=g)[:per_rank].to(all_rows.device)
+        if per_rank >= n:
+            return TensorDataset(all_rows, ds.runtime)
+        # with-replacement draw: a CPU randperm over millions of
+        # descriptors costs ~100 ms; duplicates are harmless for the
+        # PCA/GMM sample fits this feeds
+        idx = torch.randint(0, n, (per_rank,), generator=g).to(all_rows.device)
         return TensorDataset(all_rows[idx], ds.runtime)
