@@ -452,10 +452,14 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             Wb[:, torch.tensor(cls_list,
                                                device=Wb.device)] = sol.T
                         else:
+                            # large n_c: direct per-class factorization
+                            B_dir = base_w * gram_all.to(sdt) \
+                                + self.lam * torch.eye(db, dtype=sdt,
+                                                       device=Xb.device)
                             for c, j in zip(cls_list, js):
                                 s, e = cls_slice[c]
                                 Xc64 = Xb64[s:e]
-                                G = B64 + wcs[c] * (Xc64.T @ Xc64)
+                                G = B_dir + wcs[c] * (Xc64.T @ Xc64)
                                 Lc = torch.linalg.cholesky(G)
                                 Wb[:, c] = torch.cholesky_solve(
                                     rhs[:, j:j + 1], Lc).squeeze(1).to(dtype)
