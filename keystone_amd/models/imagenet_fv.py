@@ -57,6 +57,8 @@ class ImageNetFVConfig:
     mixture_weight: float = 0.5
     top_k: int = 5
     seed: int = 0
+    gmm_iters: int = 20
+    kmeans_iters: int = 5
 
 
 def _fit_branch(extractor: PipelineBase, train_imgs: TensorDataset,
@@ -83,7 +85,8 @@ def _fit_branch(extractor: PipelineBase, train_imgs: TensorDataset,
     pca = DistributedPCAEstimator(conf.pca_dims).fit(sample)   # TSQR + SVD
     t = tick("pca", t)
     proj_sample = TensorDataset(pca.apply_batch(sample.data), sample.runtime)
-    gmm = GaussianMixtureModelEstimator(conf.gmm_k, max_iterations=20,
+    gmm = GaussianMixtureModelEstimator(conf.gmm_k,
+                                        max_iterations=conf.gmm_iters,
                                         seed=seed).fit(proj_sample)
     t = tick("gmm", t)
     fv = FisherVector(gmm)
