@@ -177,15 +177,20 @@ _RUNTIME: Optional[Runtime] = None
 
 
 def init_runtime(device: Optional[str] = None, backend: Optional[str] = None,
-                 timeout_s: int = 600) -> Runtime:
+                 timeout_s: Optional[int] = None) -> Runtime:
     """Initialise the process's runtime from torchrun-style env vars.
 
     Single-process (no RANK in env): a trivial runtime, no process group.
     Multi-process: ``nccl`` (=RCCL) when CUDA is available, else ``gloo``.
+    A hung collective aborts the job after ``timeout_s``
+    (KEYSTONE_COLL_TIMEOUT_S, default 600) instead of deadlocking —
+    the failure-detection story for single-node SPMD.
     """
     global _RUNTIME
     if _RUNTIME is not None:
         return _RUNTIME
+    if timeout_s is None:
+        timeout_s = int(os.environ.get("KEYSTONE_COLL_TIMEOUT_S", "600"))
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
