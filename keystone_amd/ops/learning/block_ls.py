@@ -231,6 +231,33 @@ class BlockLeastSquaresEstimator(LabelEstimator):
             Ws = [w.to(Y.device) if w is not None else None
                   for w in st["Ws"]]
             R = st["R"].to(Y.device)
+
+        # ---- batched factorization pre-pass (same rationale as the
+        # weighted solver): block Grams + lam I don't depend on residuals,
+        # so factor every block in one batched Cholesky + inverse and turn
+        # the per-block solves into GEMMs.  Materialized uniform blocks
+        # only; lazy/ragged blocks keep the per-block _solve_spd path.
+        Binvs = None
+        if (isinstance(data, TensorDataset) and nb > 1
+                and data.data.shape[1] % self.block_size == 0):
+            sdt = _solve_dtype(dtype)
+            try:
+                n_loc = data.data.shape[0]
+                Xs = data.data.view(n_loc, nb, self.block_size) \
+                    .permute(1, 2, 0).contiguous().to(sdt)   # [nb, db, n]
+                if x_means is not None:
+                    means = torch.stack(x_means).to(sdt)     # [nb, db]
+                    Xs = Xs - means.unsqueeze(2)
+                grams = Xs @ Xs.transpose(1, 2)
+                rt.all_reduce_(grams)
+                eye = torch.eye(self.block_size, dtype=sdt,
+                                device=data.data.device)
+                Lb = torch.linalg.cholesky(grams + self.lam * eye)
+                Binvs = torch.cholesky_inverse(Lb)
+                del Xs, grams, Lb
+            except torch.linalg.LinAlgError:
+                Binvs = None
+
         for _epoch in range(start_epoch, self.num_iter):
             for b in range(nb):
                 Xb = getters[b]()
@@ -238,10 +265,14 @@ class BlockLeastSquaresEstimator(LabelEstimator):
                     Xb = Xb - x_means[b]
                 # add this block's current contribution back into the target
                 T = R if Ws[b] is None else R + Xb @ Ws[b]
-                gram = Xb.T @ Xb            # [db, db] local GEMM (hipBLASLt)
                 rhs = Xb.T @ T              # [db, k]
-                rt.all_reduce_many_([gram, rhs])   # the #1 comm hot spot
-                Wb = _solve_spd(gram, rhs, self.lam)
+                if Binvs is not None:
+                    rt.all_reduce_(rhs)
+                    Wb = (Binvs[b] @ rhs.to(Binvs.dtype)).to(dtype)
+                else:
+                    gram = Xb.T @ Xb        # [db, db] local GEMM (hipBLASLt)
+                    rt.all_reduce_many_([gram, rhs])  # the #1 comm hot spot
+                    Wb = _solve_spd(gram, rhs, self.lam)
                 R = T - Xb @ Wb
                 Ws[b] = Wb
             if self.checkpoint_dir:
