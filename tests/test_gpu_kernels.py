@@ -208,3 +208,34 @@ def test_sift_norm_kernel(hip):
     ref = ref / ref.norm(dim=1, keepdim=True).clamp_min(1e-10)
     out = hip.sift_norm_(D.clone(), 0.2)
     assert (out - ref).abs().max().item() < 1e-5
+
+
+def test_imagenet_fv_fit_and_serve_gpu():
+    import keystone_amd as ka
+    from keystone_amd.loaders import synthetic_images
+    from keystone_amd.models import ImageNetFVConfig, fit_imagenet_fv
+    imgs, labels = synthetic_images(16, h=64, w=64, num_classes=8, seed=12)
+    conf = ImageNetFVConfig(num_classes=8, sift_step=6, sift_bin=4,
+                            sift_scales=1, lcs_step=6, lcs_subpatch=4,
+                            pca_dims=16, gmm_k=16, n_sample=4000,
+                            block_size=128, solver_epochs=2, lam=1e-4,
+                            top_k=1, gmm_iters=10)
+    pipe, info = fit_imagenet_fv(imgs, labels, conf)
+    pred = pipe.apply(imgs).get().data.squeeze(1)
+    acc = (pred == labels.data).float().mean().item()
+    assert acc > 0.5
+
+
+def test_save_load_on_gpu(tmp_path_factory):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import LinearMapEstimator
+    tmp = tmp_path_factory.mktemp("gpu_ckpt")
+    X = torch.randn(40, 6, device="cuda")
+    Y = torch.randn(40, 2, device="cuda")
+    m = LinearMapEstimator(0.1).fit(ka.from_rows(X), ka.from_rows(Y))
+    pipe = ka.Identity().and_then(m)
+    pipe.save(str(tmp / "p"))
+    loaded = ka.Pipeline.load(str(tmp / "p"))
+    a = pipe.apply(ka.from_rows(X)).get().data
+    b = loaded.apply(ka.from_rows(X)).get().data
+    assert torch.allclose(a, b, atol=1e-5)
