@@ -239,3 +239,24 @@ def test_save_load_on_gpu(tmp_path_factory):
     a = pipe.apply(ka.from_rows(X)).get().data
     b = loaded.apply(ka.from_rows(X)).get().data
     assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_graph_captured_serving_matches_eager():
+    import keystone_amd as ka
+    from keystone_amd.workflow.capture import capture_apply
+    from keystone_amd.ops.stats import CosineRandomFeatures, NormalizeRows
+    from keystone_amd.ops.learning import LinearMapEstimator
+    torch.manual_seed(13)
+    X = torch.randn(256, 32, device="cuda")
+    Y = torch.randn(256, 4, device="cuda")
+    ds = ka.from_rows(X)
+    pipe = (CosineRandomFeatures(32, 128, 0.5)
+            .and_then(NormalizeRows())
+            .and_then(LinearMapEstimator(0.1), ds, ka.from_rows(Y))).fit()
+    eager = pipe.apply(ds).get().data
+    cap = capture_apply(pipe, X)
+    X2 = torch.randn(256, 32, device="cuda")
+    out2 = cap.apply_batch(X2)
+    ref2 = pipe.apply(ka.from_rows(X2)).get().data
+    assert torch.allclose(cap.apply_batch(X), eager, atol=1e-4)
+    assert torch.allclose(out2, ref2, atol=1e-4)
