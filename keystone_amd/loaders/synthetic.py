@@ -10,9 +10,9 @@ same shapes as the real loaders:
   4. Amazon reviews  -> longer documents, binary labels
   5. ImageNet        -> 256x256 grayscale-ish images, 1000 classes
 
-All generators shard deterministically by rank: rank r generates rows
-[lo, hi) of the same global sequence, so N-GPU runs see the same global
-dataset as 1-GPU runs.
+Sharding: documents are per-index deterministic (identical global corpus
+at any world size); the dense/image generators draw independent per-rank
+streams (weak-scaling benchmarks fix per-rank work, not the global set).
 """
 
 from __future__ import annotations

@@ -12,7 +12,6 @@ from __future__ import annotations
 import math
 
 import torch
-import torch.nn.functional as F
 
 from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.pipeline import Transformer

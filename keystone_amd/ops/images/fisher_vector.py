@@ -21,12 +21,9 @@ dispatched automatically for fp32 CUDA shards.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
-from ...parallel.dataset import Dataset, ItemDataset, TensorDataset
-from ...workflow.optimizer import Optimizable
+from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.pipeline import Transformer
 from ..learning.gmm import GaussianMixtureModel
 

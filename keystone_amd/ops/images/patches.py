@@ -11,7 +11,6 @@ from typing import List
 import torch
 
 from ...workflow.pipeline import Transformer
-from ...parallel.dataset import Dataset, ItemDataset
 
 
 def _crop(x: torch.Tensor, top: int, left: int, size: int) -> torch.Tensor:

@@ -17,10 +17,9 @@ is the K1 upgrade; this version doubles as its numerics oracle.
 from __future__ import annotations
 
 import math
-from typing import List, Tuple
+from typing import Tuple
 
 import torch
-import torch.nn.functional as F
 
 from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.pipeline import Transformer

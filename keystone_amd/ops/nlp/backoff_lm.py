@@ -14,7 +14,7 @@ import math
 from collections import Counter
 from typing import Dict, List, Sequence
 
-from ...parallel.dataset import Dataset, ItemDataset
+from ...parallel.dataset import ItemDataset
 from ...workflow.pipeline import Estimator, Transformer
 from .vocab import _merge_counters
 

@@ -9,9 +9,9 @@ all-gather-object exchange (host-side; n-gram keys are strings).
 from __future__ import annotations
 
 from collections import Counter
-from typing import Dict, List, Sequence, Tuple
+from typing import List, Sequence
 
-from ...parallel.dataset import Dataset, ItemDataset
+from ...parallel.dataset import Dataset
 from ...workflow.pipeline import Transformer
 
 

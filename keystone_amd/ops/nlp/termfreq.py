@@ -8,7 +8,7 @@ from __future__ import annotations
 
 import math
 from collections import Counter
-from typing import Callable, Dict, Sequence
+from typing import Dict, Sequence
 
 from ...workflow.pipeline import Transformer
 

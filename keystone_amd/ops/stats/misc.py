@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import torch
 
-from ...parallel.dataset import Dataset, ItemDataset, TensorDataset, from_rows
+from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.optimizer import ElementwiseTransformer
 from ...workflow.pipeline import Transformer
 

@@ -7,11 +7,10 @@ ref: nodes.util.ClassLabelIndicatorsFromIntLabels (label ->
 
 from __future__ import annotations
 
-from typing import Dict, List, Sequence
+from typing import Dict, Sequence
 
 import torch
 
-from ...parallel.dataset import Dataset, ItemDataset, TensorDataset
 from ...workflow.pipeline import BatchTransformer, Transformer
 
 

@@ -8,11 +8,11 @@ Densify / Sparsify, FloatToDouble (generalized to Cast).
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 
-from ...parallel.dataset import Dataset, ItemDataset, TensorDataset
+from ...parallel.dataset import Dataset, TensorDataset
 from ...workflow.pipeline import BatchTransformer, Transformer, TupleDataset
 
 
