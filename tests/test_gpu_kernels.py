@@ -260,3 +260,33 @@ def test_graph_captured_serving_matches_eager():
     ref2 = pipe.apply(ka.from_rows(X2)).get().data
     assert torch.allclose(cap.apply_batch(X), eager, atol=1e-4)
     assert torch.allclose(out2, ref2, atol=1e-4)
+
+
+def test_solvers_gpu_match_cpu_fp32():
+    """GPU fp32 solver results track the CPU fp32 oracle (fp32 algebra on
+    both sides; BCD self-correction keeps them close)."""
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import (BlockLeastSquaresEstimator,
+                                           BlockWeightedLeastSquaresEstimator)
+    torch.manual_seed(20)
+    n, d, K = 256, 64, 8
+    X = torch.randn(n, d)
+    y = torch.arange(n) % K
+    Y = torch.full((n, K), -1.0)
+    Y[torch.arange(n), y] = 1.0
+
+    m_cpu = BlockLeastSquaresEstimator(16, 5, 1e-3).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    m_gpu = BlockLeastSquaresEstimator(16, 5, 1e-3).fit(
+        ka.from_rows(X.cuda()), ka.from_rows(Y.cuda()))
+    W_c = torch.cat(m_cpu.blocks, 0)
+    W_g = torch.cat([b.cpu() for b in m_gpu.blocks], 0)
+    assert (W_c - W_g).abs().max().item() < 1e-2
+
+    w_cpu = BlockWeightedLeastSquaresEstimator(16, 5, 1e-3, 0.5).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    w_gpu = BlockWeightedLeastSquaresEstimator(16, 5, 1e-3, 0.5).fit(
+        ka.from_rows(X.cuda()), ka.from_rows(Y.cuda()))
+    Wc = torch.cat(w_cpu.blocks, 0)
+    Wg = torch.cat([b.cpu() for b in w_gpu.blocks], 0)
+    assert (Wc - Wg).abs().max().item() < 1e-2
