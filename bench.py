@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--sift-step", type=int, default=6)
     p.add_argument("--solver-epochs", type=int, default=1)
     p.add_argument("--no-lcs", action="store_true")
+    p.add_argument("--gray", action="store_true",
+                   help="grayscale images (default: color, 96-dim LCS)")
     p.add_argument("--cpu-smoke", action="store_true",
                    help="tiny CPU run for plumbing checks")
     return p.parse_args()
@@ -63,7 +65,8 @@ def main():
     n_total = args.images_per_gpu * rt.world_size
     imgs, labels = synthetic_images(
         n_total, h=args.image_size, w=args.image_size,
-        num_classes=args.num_classes, seed=0, runtime=rt, device=dev)
+        num_classes=args.num_classes, seed=0, runtime=rt, device=dev,
+        channels=1 if (args.gray or args.cpu_smoke) else 3)
 
     conf = ImageNetFVConfig(
         num_classes=args.num_classes, sift_step=args.sift_step,
@@ -152,6 +155,7 @@ def main():
                 "global_batch": n_total,
                 "images_per_gpu": args.images_per_gpu,
                 "image_size": args.image_size,
+                "color": not (args.gray or args.cpu_smoke),
                 "num_classes": args.num_classes,
                 "gmm_k": args.gmm_k,
                 "pca_dims": args.pca_dims,

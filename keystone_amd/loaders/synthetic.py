@@ -93,20 +93,23 @@ def synthetic_timit(n: int = 100000, context: int = 20, seed: int = 0,
 
 def synthetic_images(n: int, h: int = 256, w: int = 256, num_classes: int = 1000,
                      seed: int = 0, runtime=None, device=None,
-                     batched: bool = True):
-    """ImageNet-shaped synthetic grayscale images in [0,1]: smooth random
-    blobs (low-frequency noise) so SIFT-like featurizers see structure.
-    Returns ([n_local, h, w] TensorDataset, int labels)."""
+                     channels: int = 1):
+    """ImageNet-shaped synthetic images in [0,1]: smooth random blobs
+    (low-frequency noise) so SIFT-like featurizers see structure.
+    Returns ([n_local, h, w] gray or [n_local, C, h, w] color
+    TensorDataset, int labels)."""
     rt = runtime or get_runtime()
     dev = device or rt.device
     lo, hi = rt.shard_bounds(n)
     n_local = hi - lo
     g = torch.Generator().manual_seed(seed + 23 + rt.rank)
     # low-res noise upsampled = smooth structure, plus fine detail
-    coarse = torch.rand(n_local, 1, h // 8, w // 8, generator=g)
+    coarse = torch.rand(n_local, channels, h // 8, w // 8, generator=g)
     imgs = torch.nn.functional.interpolate(coarse, size=(h, w),
                                            mode="bilinear",
-                                           align_corners=False).squeeze(1)
-    imgs = 0.8 * imgs + 0.2 * torch.rand(n_local, h, w, generator=g)
+                                           align_corners=False)
+    imgs = 0.8 * imgs + 0.2 * torch.rand(n_local, channels, h, w, generator=g)
+    if channels == 1:
+        imgs = imgs.squeeze(1)
     labels = torch.randint(0, num_classes, (n_local,), generator=g)
     return (from_rows(imgs.to(dev), rt), from_rows(labels.to(dev), rt))

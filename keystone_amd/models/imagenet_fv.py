@@ -122,8 +122,13 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
     feats = sift_feats.data
     if conf.use_lcs:
         t1 = time.perf_counter()
-        lcs = GrayScaler().and_then(
-            LCSExtractor(conf.lcs_step, conf.lcs_subpatch))
+        # LCS consumes the COLOR image directly (ref: 96-dim color LCS);
+        # the extractor handles gray [B,H,W] shards too.  (Identity prefix
+        # lifts the bare transformer to a Pipeline: Transformer.apply is
+        # the datum-level method.)
+        from ..workflow.pipeline import Identity
+        lcs = Identity().and_then(LCSExtractor(conf.lcs_step,
+                                               conf.lcs_subpatch))
         lcs_chain, lcs_feats = _fit_branch(lcs, train_imgs, conf,
                                            conf.seed + 1, tm, "lcs_")
         branches.append(lcs_chain)

@@ -108,3 +108,19 @@ def test_config2_mnist_random_cosine():
     pred = pipe.apply(train).get()
     m = MulticlassClassifierEvaluator(10).evaluate(pred, labels)
     assert m.total_accuracy > 0.9
+
+
+def test_config5_imagenet_fv_color():
+    imgs, labels = synthetic_images(10, h=48, w=48, num_classes=4, seed=9,
+                                    channels=3)
+    assert imgs.data.dim() == 4
+    conf = ImageNetFVConfig(num_classes=4, sift_step=6, sift_bin=4,
+                            sift_scales=1, lcs_step=6, lcs_subpatch=4,
+                            pca_dims=16, gmm_k=4, n_sample=2000,
+                            block_size=128, solver_epochs=2, lam=1e-4,
+                            top_k=1)
+    pipe, info = fit_imagenet_fv(imgs, labels, conf)
+    # LCS branch sees color: 96-dim raw descriptors -> PCA 16
+    pred = pipe.apply(imgs).get().data.squeeze(1)
+    acc = (pred == labels.data).float().mean().item()
+    assert acc > 0.5
