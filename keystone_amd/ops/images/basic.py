@@ -23,15 +23,18 @@ class GrayScaler(BatchTransformer):
     """RGB -> luminance gray (ref: GrayScaler). [B?, 3, H, W] -> [B?, H, W];
     already-gray inputs pass through."""
 
-    _w = torch.tensor([0.299, 0.587, 0.114])
+    def __init__(self):
+        self._w = torch.tensor([0.299, 0.587, 0.114])
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
         if X.dim() == 3:  # [B, H, W] already gray
             return X
-        w = self._w.to(device=X.device, dtype=X.dtype)
+        # cache weights on the compute device (a per-call H2D upload would
+        # also break hipGraph capture of the serving path)
+        self._w = self._w.to(device=X.device, dtype=X.dtype)
         if X.shape[1] == 1:
             return X[:, 0]
-        return torch.einsum("bchw,c->bhw", X, w)
+        return torch.einsum("bchw,c->bhw", X, self._w)
 
     def apply(self, x):
         if x.dim() == 2:

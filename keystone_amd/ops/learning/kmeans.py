@@ -24,7 +24,8 @@ class KMeansModel(BatchTransformer):
         self.centers = centers  # [K, d]
 
     def assign(self, X: torch.Tensor) -> torch.Tensor:
-        C = self.centers.to(device=X.device, dtype=X.dtype)
+        self.centers = self.centers.to(device=X.device, dtype=X.dtype)
+        C = self.centers
         d2 = (X * X).sum(1, keepdim=True) - 2.0 * (X @ C.T) + (C * C).sum(1)
         return d2.argmin(dim=1)
 

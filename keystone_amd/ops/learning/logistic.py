@@ -77,8 +77,9 @@ class NaiveBayesModel(BatchTransformer):
         self.log_theta = log_theta   # [d, K]
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
-        lt = self.log_theta.to(device=X.device)
-        lp = self.log_prior.to(device=X.device)
+        self.log_theta = self.log_theta.to(device=X.device)
+        self.log_prior = self.log_prior.to(device=X.device)
+        lt, lp = self.log_theta, self.log_prior
         if X.layout == torch.sparse_csr or X.is_sparse:
             return torch.sparse.mm(X, lt.to(X.dtype)) + lp.to(X.dtype)
         return X @ lt.to(X.dtype) + lp.to(X.dtype)

@@ -37,10 +37,11 @@ class PCATransformer(BatchTransformer):
         self.mean = mean
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
-        V = self.components.to(device=X.device, dtype=X.dtype)
+        self.components = self.components.to(device=X.device, dtype=X.dtype)
         if self.mean is not None:
-            X = X - self.mean.to(device=X.device, dtype=X.dtype)
-        return X @ V
+            self.mean = self.mean.to(device=X.device, dtype=X.dtype)
+            X = X - self.mean
+        return X @ self.components
 
     def apply(self, x: torch.Tensor) -> torch.Tensor:
         return self.apply_batch(x) if x.dim() > 1 else \

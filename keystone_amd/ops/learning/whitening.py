@@ -25,9 +25,9 @@ class ZCAWhitener(BatchTransformer):
         self.mean = mean            # [d]
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
-        W = self.whitener.to(device=X.device, dtype=X.dtype)
-        mu = self.mean.to(device=X.device, dtype=X.dtype)
-        return (X - mu) @ W
+        self.whitener = self.whitener.to(device=X.device, dtype=X.dtype)
+        self.mean = self.mean.to(device=X.device, dtype=X.dtype)
+        return (X - self.mean) @ self.whitener
 
     @classmethod
     def from_saved(cls, config, state):

@@ -19,10 +19,11 @@ class StandardScalerModel(BatchTransformer):
         self.std = std
 
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
-        mean = self.mean.to(device=X.device, dtype=X.dtype)
+        self.mean = self.mean.to(device=X.device, dtype=X.dtype)
         if self.std is None:
-            return X - mean
-        return (X - mean) / self.std.to(device=X.device, dtype=X.dtype)
+            return X - self.mean
+        self.std = self.std.to(device=X.device, dtype=X.dtype)
+        return (X - self.mean) / self.std
 
     @classmethod
     def from_saved(cls, config, state):

@@ -264,8 +264,10 @@ def test_graph_captured_serving_matches_eager():
 
 def test_solvers_gpu_match_cpu_fp32():
     """GPU fp32 solver results track the CPU fp32 oracle (fp32 algebra on
-    both sides; BCD self-correction keeps them close)."""
+    both sides; BCD self-correction keeps them close).  The CPU fit pins
+    an explicit CPU runtime (from_rows would auto-move to the GPU)."""
     import keystone_amd as ka
+    from keystone_amd.parallel.runtime import Runtime
     from keystone_amd.ops.learning import (BlockLeastSquaresEstimator,
                                            BlockWeightedLeastSquaresEstimator)
     torch.manual_seed(20)
@@ -274,19 +276,20 @@ def test_solvers_gpu_match_cpu_fp32():
     y = torch.arange(n) % K
     Y = torch.full((n, K), -1.0)
     Y[torch.arange(n), y] = 1.0
+    cpu_rt = Runtime(rank=0, world_size=1, device=torch.device("cpu"))
 
     m_cpu = BlockLeastSquaresEstimator(16, 5, 1e-3).fit(
-        ka.from_rows(X), ka.from_rows(Y))
+        ka.TensorDataset(X, cpu_rt), ka.TensorDataset(Y, cpu_rt))
     m_gpu = BlockLeastSquaresEstimator(16, 5, 1e-3).fit(
         ka.from_rows(X.cuda()), ka.from_rows(Y.cuda()))
-    W_c = torch.cat(m_cpu.blocks, 0)
+    W_c = torch.cat(m_cpu.blocks, 0).cpu()
     W_g = torch.cat([b.cpu() for b in m_gpu.blocks], 0)
     assert (W_c - W_g).abs().max().item() < 1e-2
 
     w_cpu = BlockWeightedLeastSquaresEstimator(16, 5, 1e-3, 0.5).fit(
-        ka.from_rows(X), ka.from_rows(Y))
+        ka.TensorDataset(X, cpu_rt), ka.TensorDataset(Y, cpu_rt))
     w_gpu = BlockWeightedLeastSquaresEstimator(16, 5, 1e-3, 0.5).fit(
         ka.from_rows(X.cuda()), ka.from_rows(Y.cuda()))
-    Wc = torch.cat(w_cpu.blocks, 0)
+    Wc = torch.cat(w_cpu.blocks, 0).cpu()
     Wg = torch.cat([b.cpu() for b in w_gpu.blocks], 0)
     assert (Wc - Wg).abs().max().item() < 1e-2
