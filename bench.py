@@ -40,6 +40,9 @@ def parse_args():
     p.add_argument("--pca-dims", type=int, default=64)
     p.add_argument("--sift-step", type=int, default=6)
     p.add_argument("--solver-epochs", type=int, default=1)
+    p.add_argument("--n-sample", type=int, default=100000,
+                   help="descriptors sampled per branch for PCA/GMM fits")
+    p.add_argument("--block-size", type=int, default=4096)
     p.add_argument("--no-lcs", action="store_true")
     p.add_argument("--gray", action="store_true",
                    help="grayscale images (default: color, 96-dim LCS)")
@@ -72,8 +75,8 @@ def main():
         num_classes=args.num_classes, sift_step=args.sift_step,
         sift_bin=6, sift_scales=2, lcs_step=args.sift_step, lcs_subpatch=6,
         pca_dims=args.pca_dims, gmm_k=args.gmm_k,
-        n_sample=min(100000, 50 * args.images_per_gpu * 36),
-        use_lcs=not args.no_lcs, block_size=4096,
+        n_sample=min(args.n_sample, 50 * args.images_per_gpu * 36),
+        use_lcs=not args.no_lcs, block_size=args.block_size,
         solver_epochs=args.solver_epochs, lam=1e-5, mixture_weight=0.5)
 
     def step():
@@ -159,6 +162,8 @@ def main():
                 "num_classes": args.num_classes,
                 "gmm_k": args.gmm_k,
                 "pca_dims": args.pca_dims,
+                "gmm_pca_sample": conf.n_sample,
+                "solver_block_size": args.block_size,
                 "feature_dim": info.get("feature_dim"),
                 "fit_time_s_per_step": ms_per_step / 1000.0,
                 "serve_images_per_sec": serve_images_per_sec,
