@@ -202,50 +202,58 @@ void gemm_fused_f32_256_kernel(const float* __restrict__ A,
 
     float4 ra[4], rb[4];
 
+    auto load_a_piece = [&](int k0, int p) -> float4 {
+        const int gr = brow + a_r0 + p * 64;
+        const int gk = k0 + a_kv;
+        float4 v = {0, 0, 0, 0};
+        if (gr < M) {
+            if (gk + 3 < K) {
+                v = *reinterpret_cast<const float4*>(&A[(size_t)gr * K + gk]);
+            } else {
+#pragma unroll
+                for (int q = 0; q < 4; ++q)
+                    if (gk + q < K) ((float*)&v)[q] = A[(size_t)gr * K + gk + q];
+            }
+        }
+        return v;
+    };
+    auto load_b_piece = [&](int k0, int p) -> float4 {
+        const int gk = k0 + b_k0 + p * 8;
+        const int gn = bcol + b_nv;
+        float4 v = {0, 0, 0, 0};
+        if (gk < K) {
+            if (gn + 3 < N) {
+                v = *reinterpret_cast<const float4*>(&B[(size_t)gk * N + gn]);
+            } else {
+#pragma unroll
+                for (int q = 0; q < 4; ++q)
+                    if (gn + q < N) ((float*)&v)[q] = B[(size_t)gk * N + gn + q];
+            }
+        }
+        return v;
+    };
+    auto write_a_piece = [&](int buf, int p) {
+        const int row = a_r0 + p * 64;
+        As[buf][row][a_kv + 0] = ra[p].x;
+        As[buf][row][a_kv + 1] = ra[p].y;
+        As[buf][row][a_kv + 2] = ra[p].z;
+        As[buf][row][a_kv + 3] = ra[p].w;
+    };
+    auto write_b_piece = [&](int buf, int p) {
+        *reinterpret_cast<float4*>(&Bs[buf][b_k0 + p * 8][b_nv]) = rb[p];
+    };
     auto stage_load = [&](int k0) {
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
-            const int gr = brow + a_r0 + p * 64;
-            const int gk = k0 + a_kv;
-            float4 v = {0, 0, 0, 0};
-            if (gr < M) {
-                if (gk + 3 < K) {
-                    v = *reinterpret_cast<const float4*>(&A[(size_t)gr * K + gk]);
-                } else {
-#pragma unroll
-                    for (int q = 0; q < 4; ++q)
-                        if (gk + q < K) ((float*)&v)[q] = A[(size_t)gr * K + gk + q];
-                }
-            }
-            ra[p] = v;
-        }
-#pragma unroll
-        for (int p = 0; p < 4; ++p) {
-            const int gk = k0 + b_k0 + p * 8;
-            const int gn = bcol + b_nv;
-            float4 v = {0, 0, 0, 0};
-            if (gk < K) {
-                if (gn + 3 < N) {
-                    v = *reinterpret_cast<const float4*>(&B[(size_t)gk * N + gn]);
-                } else {
-#pragma unroll
-                    for (int q = 0; q < 4; ++q)
-                        if (gn + q < N) ((float*)&v)[q] = B[(size_t)gk * N + gn + q];
-                }
-            }
-            rb[p] = v;
+            ra[p] = load_a_piece(k0, p);
+            rb[p] = load_b_piece(k0, p);
         }
     };
-
     auto stage_write = [&](int buf) {
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
-            const int row = a_r0 + p * 64;
-            As[buf][row][a_kv + 0] = ra[p].x;
-            As[buf][row][a_kv + 1] = ra[p].y;
-            As[buf][row][a_kv + 2] = ra[p].z;
-            As[buf][row][a_kv + 3] = ra[p].w;
-            *reinterpret_cast<float4*>(&Bs[buf][b_k0 + p * 8][b_nv]) = rb[p];
+            write_a_piece(buf, p);
+            write_b_piece(buf, p);
         }
     };
 
@@ -260,27 +268,44 @@ void gemm_fused_f32_256_kernel(const float* __restrict__ A,
     stage_write(0);
     __syncthreads();
 
+    // phase-interleaved k-loop: the ablation probe (gemm_ablate.hip)
+    // measures the MFMA-only structure at 150 TF and the block staging at
+    // ~7x real-time, so the 120-TF full kernel loses ~20% purely to the
+    // staging segment draining the matrix pipe.  Spreading the 8 load
+    // pieces over phases 0-3 and the 8 write pieces over phases 4-7 (each
+    // beside 16 MFMAs) lets the co-resident wave's MFMAs cover them, and
+    // the load->write distance (~8 k-steps) covers the HBM latency.
     int cur = 0;
     for (int t = 0; t < ntiles; ++t) {
-        if (t + 1 < ntiles) stage_load((t + 1) * BK);
+        const bool more = t + 1 < ntiles;
 #pragma unroll
-        for (int kk = 0; kk < BK; kk += 2) {
-            float a0 = As[cur][wr * 128 + 0 + l31][kk + lk];
-            float a1 = As[cur][wr * 128 + 32 + l31][kk + lk];
-            float a2 = As[cur][wr * 128 + 64 + l31][kk + lk];
-            float a3 = As[cur][wr * 128 + 96 + l31][kk + lk];
-            float b0 = Bs[cur][kk + lk][wc * 64 + 0 + l31];
-            float b1 = Bs[cur][kk + lk][wc * 64 + 32 + l31];
-            acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
-            acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
-            acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
-            acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
-            acc[2][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a2, b0, acc[2][0], 0, 0, 0);
-            acc[2][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a2, b1, acc[2][1], 0, 0, 0);
-            acc[3][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b0, acc[3][0], 0, 0, 0);
-            acc[3][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b1, acc[3][1], 0, 0, 0);
+        for (int ph = 0; ph < 8; ++ph) {
+            if (more && ph < 4) {
+                ra[ph] = load_a_piece((t + 1) * BK, ph);
+                rb[ph] = load_b_piece((t + 1) * BK, ph);
+            }
+#pragma unroll
+            for (int kk = 4 * ph; kk < 4 * ph + 4; kk += 2) {
+                float a0 = As[cur][wr * 128 + 0 + l31][kk + lk];
+                float a1 = As[cur][wr * 128 + 32 + l31][kk + lk];
+                float a2 = As[cur][wr * 128 + 64 + l31][kk + lk];
+                float a3 = As[cur][wr * 128 + 96 + l31][kk + lk];
+                float b0 = Bs[cur][kk + lk][wc * 64 + 0 + l31];
+                float b1 = Bs[cur][kk + lk][wc * 64 + 32 + l31];
+                acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+                acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+                acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+                acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+                acc[2][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a2, b0, acc[2][0], 0, 0, 0);
+                acc[2][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a2, b1, acc[2][1], 0, 0, 0);
+                acc[3][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b0, acc[3][0], 0, 0, 0);
+                acc[3][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b1, acc[3][1], 0, 0, 0);
+            }
+            if (more && ph >= 4) {
+                write_a_piece(cur ^ 1, ph - 4);
+                write_b_piece(cur ^ 1, ph - 4);
+            }
         }
-        if (t + 1 < ntiles) stage_write(cur ^ 1);
         __syncthreads();
         cur ^= 1;
     }
