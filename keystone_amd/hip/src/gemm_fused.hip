@@ -284,9 +284,6 @@ void gemm_fused_f32_256_kernel(const float* __restrict__ A,
                 ra[ph] = load_a_piece((t + 1) * BK, ph);
                 rb[ph] = load_b_piece((t + 1) * BK, ph);
             }
-            // T5: favour the MFMA cluster in the SIMD's wave arbitration
-            // (pays only on phase-split schedules like this one)
-            __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int kk = 4 * ph; kk < 4 * ph + 4; kk += 2) {
                 float a0 = As[cur][wr * 128 + 0 + l31][kk + lk];
@@ -304,7 +301,6 @@ void gemm_fused_f32_256_kernel(const float* __restrict__ A,
                 acc[3][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b0, acc[3][0], 0, 0, 0);
                 acc[3][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a3, b1, acc[3][1], 0, 0, 0);
             }
-            __builtin_amdgcn_s_setprio(0);
             if (more && ph >= 4) {
                 write_a_piece(cur ^ 1, ph - 4);
                 write_b_piece(cur ^ 1, ph - 4);
