@@ -162,3 +162,72 @@ def test_evaluator_distributed_confusion():
     results = run_distributed(_eval_run, 2)
     for conf in results:
         assert torch.equal(conf, ref)
+
+
+def _kmeans_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import KMeansPlusPlusEstimator
+    ds = ka.from_rows(_shard(rank, world, X_ALL.float()))
+    return KMeansPlusPlusEstimator(4, max_iterations=10, seed=2).fit(ds).centers
+
+
+def test_kmeans_distributed_replicated():
+    results = run_distributed(_kmeans_fit, 2)
+    assert torch.equal(results[0], results[1])  # bitwise-replicated model
+
+
+def _gmm_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning.gmm import GaussianMixtureModelEstimator
+    ds = ka.from_rows(_shard(rank, world, X_ALL.float()))
+    m = GaussianMixtureModelEstimator(3, max_iterations=8, seed=1).fit(ds)
+    return m.means, m.variances, m.weights
+
+
+def test_gmm_distributed_replicated():
+    results = run_distributed(_gmm_fit, 2)
+    for a, b in zip(results[0], results[1]):
+        assert torch.equal(a, b)
+
+
+def _logistic_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import LogisticRegressionEstimator
+    y = (X_ALL.float() @ torch.ones(D)) > 0
+    ds = ka.from_rows(_shard(rank, world, X_ALL.float()))
+    ys = ka.from_rows(_shard(rank, world, y.long()))
+    return LogisticRegressionEstimator(2, 20, 1e-3).fit(ds, ys).W
+
+
+def test_logistic_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import LogisticRegressionEstimator
+    y = (X_ALL.float() @ torch.ones(D)) > 0
+    ref = LogisticRegressionEstimator(2, 20, 1e-3).fit(
+        ka.from_rows(X_ALL.float()), ka.from_rows(y.long())).W
+    results = run_distributed(_logistic_fit, 2)
+    for W in results:
+        assert torch.allclose(W, ref, atol=1e-4)
+
+
+def _krr_fit_predict(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import (GaussianKernelGenerator,
+                                           KernelRidgeRegression)
+    ds = ka.from_rows(_shard(rank, world, X_ALL.float()))
+    ys = ka.from_rows(_shard(rank, world, Y_ALL.float()))
+    m = KernelRidgeRegression(GaussianKernelGenerator(0.2), 1e-4, 16, 3).fit(ds, ys)
+    out = m.apply_dataset(ds)
+    return out.runtime.all_gather_rows(out.data)
+
+
+def test_krr_distributed_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import (GaussianKernelGenerator,
+                                           KernelRidgeRegression)
+    m = KernelRidgeRegression(GaussianKernelGenerator(0.2), 1e-4, 16, 3).fit(
+        ka.from_rows(X_ALL.float()), ka.from_rows(Y_ALL.float()))
+    ref = m.apply_batch(X_ALL.float())
+    results = run_distributed(_krr_fit_predict, 2)
+    for pred in results:
+        assert torch.allclose(pred, ref, atol=1e-3)
