@@ -93,9 +93,17 @@ def synthetic_timit(n: int = 100000, context: int = 20, seed: int = 0,
 
 def synthetic_images(n: int, h: int = 256, w: int = 256, num_classes: int = 1000,
                      seed: int = 0, runtime=None, device=None,
-                     channels: int = 1):
+                     channels: int = 1, class_structure: bool = False,
+                     pattern_seed: int = 1234):
     """ImageNet-shaped synthetic images in [0,1]: smooth random blobs
     (low-frequency noise) so SIFT-like featurizers see structure.
+
+    ``class_structure=True`` additionally stamps each image with its
+    class's fixed random texture patch at jittered positions — labels
+    become *visually learnable*, so the FV pipeline's test-set accuracy
+    measures real generalization (used by the statistical-parity tests;
+    the throughput bench keeps random labels).
+
     Returns ([n_local, h, w] gray or [n_local, C, h, w] color
     TensorDataset, int labels)."""
     rt = runtime or get_runtime()
@@ -109,7 +117,21 @@ def synthetic_images(n: int, h: int = 256, w: int = 256, num_classes: int = 1000
                                            mode="bilinear",
                                            align_corners=False)
     imgs = 0.8 * imgs + 0.2 * torch.rand(n_local, channels, h, w, generator=g)
+    labels = torch.randint(0, num_classes, (n_local,), generator=g)
+    if class_structure:
+        ps = max(8, h // 8)
+        # class patterns are tied to pattern_seed, NOT the data seed, so
+        # train/test splits drawn with different seeds share the classes
+        gc = torch.Generator().manual_seed(pattern_seed)
+        patterns = torch.rand(num_classes, channels, ps, ps, generator=gc)
+        n_stamps = max(3, (h // ps))
+        for i in range(n_local):
+            pat = patterns[labels[i]]
+            for _ in range(n_stamps):
+                y0 = int(torch.randint(0, h - ps + 1, (1,), generator=g))
+                x0 = int(torch.randint(0, w - ps + 1, (1,), generator=g))
+                imgs[i, :, y0:y0 + ps, x0:x0 + ps] = \
+                    0.35 * imgs[i, :, y0:y0 + ps, x0:x0 + ps] + 0.65 * pat
     if channels == 1:
         imgs = imgs.squeeze(1)
-    labels = torch.randint(0, num_classes, (n_local,), generator=g)
     return (from_rows(imgs.to(dev), rt), from_rows(labels.to(dev), rt))
