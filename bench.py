@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=2)
     p.add_argument("--warmup", type=int, default=1)
-    p.add_argument("--images-per-gpu", type=int, default=2048)
+    p.add_argument("--images-per-gpu", type=int, default=8192)
     p.add_argument("--image-size", type=int, default=256)
     p.add_argument("--num-classes", type=int, default=1000)
     p.add_argument("--gmm-k", type=int, default=256)
