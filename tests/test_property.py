@@ -1,0 +1,90 @@
+"""Property-based tests (hypothesis): solver-oracle equivalence over random
+shapes, and optimizer rewrites never changing pipeline semantics."""
+
+import hypothesis
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+import keystone_amd as ka
+from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+from keystone_amd.workflow.executor import GraphExecutor
+from keystone_amd.workflow.optimizer import default_optimizer
+
+
+class Affine(ka.BatchTransformer):
+    def __init__(self, a, b):
+        self.a = a
+        self.b = b
+
+    def apply_batch(self, X):
+        return X * self.a + self.b
+
+
+@settings(max_examples=15, deadline=None)
+@given(n=st.integers(8, 60), d=st.integers(2, 20),
+       k=st.integers(1, 4), bs=st.integers(1, 8),
+       lam=st.floats(0.05, 1.0), seed=st.integers(0, 10_000))
+def test_bcd_converges_to_ridge_oracle(n, d, k, bs, lam, seed):
+    # well-conditioned regime: block coordinate descent's linear rate
+    # degrades as lam -> 0 with n ~ d (math, not implementation)
+    hypothesis.assume(n >= 3 * d)
+    g = torch.Generator().manual_seed(seed)
+    X = torch.randn(n, d, generator=g, dtype=torch.float64)
+    Y = torch.randn(n, k, generator=g, dtype=torch.float64)
+    m = BlockLeastSquaresEstimator(bs, num_iter=120, lam=lam,
+                                   fit_intercept=False).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    W = torch.cat(m.blocks, dim=0)
+    G = X.T @ X + lam * torch.eye(d, dtype=torch.float64)
+    ref = torch.linalg.solve(G, X.T @ Y)
+    assert torch.allclose(W, ref, atol=1e-4, rtol=1e-4)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.data())
+def test_optimizer_rewrites_preserve_semantics(data):
+    """Random pipelines of chains and gathers: output with the full rule
+    set == output with no optimization."""
+    g = torch.Generator().manual_seed(data.draw(st.integers(0, 9999)))
+    X = torch.randn(6, 4, generator=g)
+    ds = ka.from_rows(X)
+
+    def build(depth):
+        coef = data.draw(st.floats(-2, 2, allow_nan=False))
+        off = data.draw(st.floats(-1, 1, allow_nan=False))
+        node = Affine(round(coef, 3), round(off, 3))
+        if depth <= 0:
+            return ka.Identity().and_then(node)
+        choice = data.draw(st.integers(0, 2))
+        if choice == 0:
+            return build(depth - 1).and_then(node)
+        if choice == 1:
+            shared = build(depth - 1)
+            from keystone_amd.ops.util import VectorCombiner
+            return ka.gather([shared, shared.and_then(node)]) \
+                .and_then(VectorCombiner())
+        return ka.Identity().and_then(node)
+
+    pipe = build(data.draw(st.integers(1, 3)))
+    lazy = pipe.apply(ds)
+    out_opt = GraphExecutor(lazy.graph.copy(), optimize=True).execute(lazy.sink)
+    out_raw = GraphExecutor(lazy.graph.copy(), optimize=False).execute(lazy.sink)
+    assert torch.allclose(out_opt.data, out_raw.data, atol=1e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(d=st.integers(1, 30), k=st.integers(1, 10), seed=st.integers(0, 999))
+def test_linear_mapper_roundtrip_property(d, k, seed, tmp_path_factory):
+    from keystone_amd.ops.learning import LinearMapper
+    g = torch.Generator().manual_seed(seed)
+    W = torch.randn(d, k, generator=g)
+    b = torch.randn(k, generator=g)
+    pipe = ka.Identity().and_then(LinearMapper(W, b))
+    tmp = tmp_path_factory.mktemp("prop")
+    pipe.save(str(tmp / "m"))
+    loaded = ka.Pipeline.load(str(tmp / "m"))
+    X = torch.randn(5, d, generator=g)
+    a = pipe.apply(ka.from_rows(X)).get().data
+    bb = loaded.apply(ka.from_rows(X)).get().data
+    assert torch.allclose(a, bb, atol=1e-6)
