@@ -115,33 +115,69 @@ class GaussianMixtureModelEstimator(Estimator):
         # per-iteration kernel launches), and X2 is reused by Sxx.
         X2 = X * X
         Xext = torch.cat([X2, X], dim=1)                 # [n, 2D]
+
+        # The per-iteration LOCAL compute (logit addmm -> softmax -> local
+        # sufficient statistics) is a fixed launch-bound kernel chain —
+        # capture it into ONE hipGraph and replay per iteration.  Only the
+        # collective-free part is captured (the all-reduce and M-step stay
+        # eager), so the behaviour is identical at every world size and no
+        # RCCL call is ever captured.  KEYSTONE_NO_GRAPHS=1 disables.
+        import os
+        P_s = torch.empty(K, 2 * D, dtype=dt, device=dev)
+        cc_s = torch.empty(K, dtype=dt, device=dev)
+
+        def estep_local():
+            lp = torch.addmm(-cc_s, Xext, P_s.T)
+            lse = torch.logsumexp(lp, dim=1)
+            gamma = torch.exp(lp - lse.unsqueeze(1))
+            Nk_l = gamma.sum(0, dtype=torch.float64)
+            # fp64 stats for fp64 inputs (oracle parity); fp32 GEMMs else
+            if X.dtype == torch.float64:
+                Sx_l = gamma.T @ X
+                Sxx_l = gamma.T @ X2
+            else:
+                Sx_l = (gamma.T @ X).to(torch.float64)
+                Sxx_l = (gamma.T @ X2).to(torch.float64)
+            ll_l = lse.sum(dtype=torch.float64).reshape(1)
+            return Nk_l, Sx_l, Sxx_l, ll_l
+
+        def fill_coeffs(mdl):
+            inv2v = 0.5 / mdl.variances
+            cc_s.copy_((mdl.means * mdl.means * inv2v).sum(1)
+                       + 0.5 * mdl.variances.log().sum(1)
+                       + 0.5 * D * math.log(2 * math.pi)
+                       - mdl.weights.clamp_min(1e-30).log())
+            P_s[:, :D].copy_(-inv2v)
+            P_s[:, D:].copy_(mdl.means / mdl.variances)
+
+        graph = None
+        if X.is_cuda and not os.environ.get("KEYSTONE_NO_GRAPHS"):
+            try:
+                fill_coeffs(model)
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(2):
+                        estep_local()
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    g_out = estep_local()
+            except Exception:
+                graph = None
+
         prev_ll = -float("inf")
         for _it in range(self.max_iterations):
-            mu_c = model.means
-            var_c = model.variances
-            inv2v = 0.5 / var_c
-            cc = (mu_c * mu_c * inv2v).sum(1) + 0.5 * var_c.log().sum(1) \
-                + 0.5 * D * math.log(2 * math.pi) \
-                - model.weights.clamp_min(1e-30).log()
-            P = torch.cat([-inv2v, mu_c / var_c], dim=1)  # [K, 2D]
-            lp = torch.addmm(-cc, Xext, P.T)
-            lse = torch.logsumexp(lp, dim=1)
-            gamma = torch.exp(lp - lse.unsqueeze(1))     # responsibilities
-            lse_sum = None
-            # sufficient statistics: fp64 for fp64 inputs (oracle parity);
-            # fp32 GEMMs (the MFMA fast path) otherwise, promoted to fp64
-            # only for the (tiny) reduced accumulators
-            Nk = gamma.sum(0, dtype=torch.float64)                       # [K]
-            if X.dtype == torch.float64:
-                Sx = gamma.T @ X
-                Sxx = gamma.T @ X2
+            fill_coeffs(model)
+            if graph is not None:
+                graph.replay()
+                Nk = g_out[0].clone()
+                Sx = g_out[1].clone()
+                Sxx = g_out[2].clone()
+                ll = g_out[3].clone()
             else:
-                Sx = (gamma.T @ X).to(torch.float64)
-                Sxx = (gamma.T @ X2).to(torch.float64)
-            if lse_sum is not None:
-                ll = lse_sum.reshape(1)
-            else:
-                ll = lse.sum(dtype=torch.float64).reshape(1)
+                Nk, Sx, Sxx, ll = estep_local()
             rt.all_reduce_many_([Nk, Sx, Sxx, ll])
             Nk_c = Nk.clamp_min(1e-10).unsqueeze(1)
             mu64 = Sx / Nk_c
