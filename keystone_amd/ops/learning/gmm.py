@@ -190,5 +190,15 @@ class GaussianMixtureModelEstimator(Estimator):
                 if abs(ll_now - prev_ll) < self.tol * max(1.0, abs(prev_ll)):
                     break
                 prev_ll = ll_now
+        if graph is not None:
+            # release the graph's private memory pool NOW: keeping it
+            # reserved starves the multi-GB allocations of later pipeline
+            # stages (measured: a leaked pool made the following FV encode
+            # ~50x slower at bench scale from allocator thrash)
+            del g_out
+            graph.reset()
+            del graph
+            torch.cuda.synchronize()
+            torch.cuda.empty_cache()
         tick(f"em_loop({_it + 1} iters)", t)
         return model
