@@ -150,8 +150,15 @@ class GaussianMixtureModelEstimator(Estimator):
             P_s[:, :D].copy_(-inv2v)
             P_s[:, D:].copy_(mdl.means / mdl.variances)
 
+        # OPT-IN (KEYSTONE_GMM_GRAPHS=1): the captured E-step replays 3x
+        # faster in isolation, but the graph's private memory pool
+        # interacts pathologically with the multi-GB allocations of later
+        # pipeline stages at large batches (FV encode measured 40-70x
+        # slower afterwards, even with reset+empty_cache) — so the default
+        # stays eager; revisit when torch's ROCm graph pools can be
+        # reliably returned to the main allocator.
         graph = None
-        if X.is_cuda and not os.environ.get("KEYSTONE_NO_GRAPHS"):
+        if X.is_cuda and os.environ.get("KEYSTONE_GMM_GRAPHS"):
             try:
                 fill_coeffs(model)
                 side = torch.cuda.Stream()
