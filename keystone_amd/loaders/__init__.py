@@ -1,7 +1,8 @@
 from .synthetic import (synthetic_dense, synthetic_documents, synthetic_images,
                         synthetic_mnist, synthetic_timit)
 from .files import (AmazonReviewsDataLoader, CifarLoader, CsvDataLoader,
-                    ImageNetLoader, LabeledData, NewsgroupsDataLoader,
+                    ImageExtractor, ImageNetLoader, LabeledData,
+                    LabelExtractor, NewsgroupsDataLoader,
                     TimitFeaturesDataLoader, VOCLoader)
 
 __all__ = [
@@ -9,5 +10,5 @@ __all__ = [
     "synthetic_mnist", "synthetic_timit",
     "CsvDataLoader", "CifarLoader", "NewsgroupsDataLoader",
     "AmazonReviewsDataLoader", "TimitFeaturesDataLoader", "ImageNetLoader",
-    "VOCLoader", "LabeledData",
+    "VOCLoader", "LabeledData", "ImageExtractor", "LabelExtractor",
 ]

@@ -188,3 +188,17 @@ def VOCLoader(path: str, num_classes: int = 20, runtime=None):
         labels.append(y)
     return (from_items(imgs, rt),
             from_rows(torch.stack(labels).to(rt.device), rt))
+
+
+class ImageExtractor:
+    """LabeledData -> images (ref: nodes.images.ImageExtractor)."""
+
+    def __call__(self, ld: LabeledData):
+        return ld.data
+
+
+class LabelExtractor:
+    """LabeledData -> labels (ref: nodes.images.LabelExtractor)."""
+
+    def __call__(self, ld: LabeledData):
+        return ld.labels
