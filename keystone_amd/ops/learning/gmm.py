@@ -205,7 +205,5 @@ class GaussianMixtureModelEstimator(Estimator):
             del g_out
             graph.reset()
             del graph
-            torch.cuda.synchronize()
-            torch.cuda.empty_cache()
         tick(f"em_loop({_it + 1} iters)", t)
         return model
