@@ -380,15 +380,20 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         Binvs = grams_pre = None
         if uniform and own:
             db0 = blocks[0].shape[1]
-            Xs = X_full.view(X_full.shape[0], nb, db0) \
-                .permute(1, 2, 0).contiguous().to(sdt)       # [nb, db, n]
-            grams_pre = Xs @ Xs.transpose(1, 2)              # [nb, db, db]
+            # per-block plain TN GEMMs into the stacked buffer: the
+            # permuted strided-batched bmm ran at ~25 TF vs ~150 for the
+            # standard layout (profiled), and this skips the [nb, db, n]
+            # contiguous copy of X entirely
+            grams_pre = torch.empty(nb, db0, db0, dtype=sdt,
+                                    device=X_full.device)
+            for b, Xb in enumerate(blocks):
+                torch.matmul(Xb.T.to(sdt), Xb.to(sdt), out=grams_pre[b])
             rt.all_reduce_(grams_pre)
             eye = torch.eye(db0, dtype=sdt, device=X_full.device)
             Bb = base_w * grams_pre + self.lam * eye
             Lb = torch.linalg.cholesky(Bb)
             Binvs = torch.cholesky_inverse(Lb)
-            del Xs, Bb, Lb
+            del Bb, Lb
 
         Ws: List[Optional[torch.Tensor]] = [None] * nb
         R = Y.clone()
