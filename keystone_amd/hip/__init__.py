@@ -99,6 +99,11 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_fv_mfma_f32.argtypes = [c_p] * 6 + [c_i] * 4 + [c_p]
+        lib.ks_fv_mfma_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
@@ -237,6 +242,36 @@ def gmm_posteriors(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
         ctypes.c_void_p(ll.data_ptr()) if ll is not None else None,
         n, D, K, _stream()), "ks_gmm_posteriors_f32")
     return (gamma, ll) if want_ll else gamma
+
+
+def fv_mfma_supported(D: int, K: int) -> bool:
+    return D <= 64 and K in (64, 128, 256)
+
+
+def fisher_vector_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
+                       w: torch.Tensor) -> torch.Tensor:
+    """Fused MFMA FV (fv_mfma.hip): X [B, n, D] -> [B, 2*K*D] without
+    materializing gamma.  D <= 64, K in {64, 128, 256}."""
+    lib = _load()
+    X = _prep(X, "fv_mfma.X")
+    B, n, D = X.shape
+    K = mu.shape[0]
+    mu = _prep(mu, "mu")
+    var = _prep(var, "var")
+    w = _prep(w, "w")
+    a, b, c = _gmm_coeffs(mu, var, w)
+    DT = 64
+    P = torch.zeros(2 * DT + 32, K, dtype=torch.float32, device=X.device)
+    P[:D] = -a.T
+    P[DT:DT + D] = b.T
+    P[2 * DT] = -c
+    out = torch.empty(B, 2 * K * D, dtype=torch.float32, device=X.device)
+    _check(lib.ks_fv_mfma_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.contiguous().data_ptr()),
+        ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
+        ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        B, n, D, K, _stream()), "ks_fv_mfma_f32")
+    return out
 
 
 def fisher_vector(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,

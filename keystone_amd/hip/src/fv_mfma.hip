@@ -1,0 +1,224 @@
+// K3 v2: Fisher Vector as a fused MFMA kernel (gfx950).
+//
+// The batched-GEMM FV path materializes gamma [B, n, K] (tens of GB at
+// benchmark scale) and re-reads it for the moment GEMMs; this kernel
+// keeps everything on-chip: per image (one workgroup), per 32-descriptor
+// chunk:
+//
+//   phase L: logits[32, K] = Xext @ P           (MFMA, acc in AGPRs)
+//            with Xext = [x^2 | x | 1 | 0-pad] (LDS, built while staging)
+//            and P = [-1/(2var) ; mu/var ; -c ; 0] (global, L2-resident)
+//   phase S: softmax rows in LDS (8 lanes per row)
+//   phase A: S[K, XC] += gamma^T @ Xext         (MFMA, accumulators held
+//            across ALL chunks of the image: S2 | S1 | S0 columns)
+//
+// The ones-column trick makes S0 just another GEMM column, and zero-padded
+// tail descriptors contribute nothing (their Xext rows are zero), so no
+// masking is needed anywhere.  Epilogue converts (S0, S1, S2) to the
+// improved-FV form and writes [2*K*D] per image.
+//
+// Supported: D <= 64 (DT=64 padded), K in {64, 128, 256}.  4 waves.
+
+#include "common.h"
+#include <math.h>
+
+#define DT 64
+#define XC (2 * DT + 2)      // 130 live columns of Xext
+#define XCP 160              // padded to 5 MFMA col-tiles
+#define XSTRIDE (XCP + 1)    // LDS row stride: (161*r + c) % 32 distinct
+#define CHUNK 32
+
+template <int KT>
+__global__ __launch_bounds__(256, 1)
+void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
+                    const float* __restrict__ P,    // [XCP, KT] padded
+                    const float* __restrict__ mu,   // [KT, D]
+                    const float* __restrict__ var,  // [KT, D]
+                    const float* __restrict__ w,    // [KT]
+                    float* __restrict__ out,        // [B, 2*KT*D]
+                    int n, int D) {
+    constexpr int KT32 = KT / 32;               // logits col-tiles
+    constexpr int LT = (KT32 + 3) / 4;          // logits tiles per wave (max)
+    constexpr int ST = 5 * ((KT32 + 3) / 4);    // stats tiles per wave (max)
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* Xs = (float*)smem;                   // [CHUNK][XSTRIDE]
+    float* Gs = Xs + CHUNK * XSTRIDE;           // [CHUNK][KT + 1] gamma
+    float* S0s = Gs + CHUNK * (KT + 1);         // [KT] broadcast
+
+    const int img = blockIdx.x;
+    const float* Xi = X + (size_t)img * n * D;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int l31 = lane & 31;
+    const int lk = lane >> 5;                   // 0/1
+
+    // stats accumulators: tile index t -> (ti, tj) = (t / 5, t % 5) with
+    // ti strided over waves: ti = wave + 4 * (t / 5) ... flattened below.
+    f32x16 sacc[ST];
+#pragma unroll
+    for (int t = 0; t < ST; ++t) sacc[t] = (f32x16)(0.0f);
+
+    for (int c0 = 0; c0 < n; c0 += CHUNK) {
+        const int rows = min(CHUNK, n - c0);
+        __syncthreads();
+        // ---- stage Xext chunk: [x^2 | x | 1 | 0]
+        for (int i = tid; i < CHUNK * XCP; i += 256) {
+            const int r = i / XCP, c = i % XCP;
+            float v = 0.0f;
+            if (r < rows) {
+                if (c < DT) {
+                    if (c < D) {
+                        float x = Xi[(size_t)(c0 + r) * D + c];
+                        v = x * x;
+                    }
+                } else if (c < 2 * DT) {
+                    const int d = c - DT;
+                    if (d < D) v = Xi[(size_t)(c0 + r) * D + d];
+                } else if (c == 2 * DT) {
+                    v = 1.0f;
+                }
+            }
+            Xs[r * XSTRIDE + c] = v;
+        }
+        __syncthreads();
+
+        // ---- phase L: logits = Xext @ P  (KT32 col-tiles over 4 waves)
+        f32x16 lacc[LT];
+#pragma unroll
+        for (int t = 0; t < LT; ++t) lacc[t] = (f32x16)(0.0f);
+#pragma unroll 1
+        for (int kk = 0; kk < XC; kk += 2) {
+            const float a = Xs[l31 * XSTRIDE + kk + lk];
+#pragma unroll
+            for (int t = 0; t < LT; ++t) {
+                const int ct = wave + 4 * t;
+                if (ct < KT32) {
+                    const float b = P[(size_t)(kk + lk) * KT + ct * 32 + l31];
+                    lacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                        a, b, lacc[t], 0, 0, 0);
+                }
+            }
+        }
+        // write logits into Gs (C layout: col = l31, row = (r&3)+8*(r>>2)+4*lk)
+#pragma unroll
+        for (int t = 0; t < LT; ++t) {
+            const int ct = wave + 4 * t;
+            if (ct < KT32) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                    Gs[row * (KT + 1) + ct * 32 + l31] = lacc[t][r];
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- phase S: row softmax, 8 lanes per row (rows = tid >> 3)
+        {
+            const int row = tid >> 3;
+            const int sub = tid & 7;
+            float m = -1e30f;
+            for (int k = sub; k < KT; k += 8)
+                m = fmaxf(m, Gs[row * (KT + 1) + k]);
+#pragma unroll
+            for (int off = 4; off; off >>= 1)
+                m = fmaxf(m, __shfl_xor(m, off, 8));
+            float s = 0.0f;
+            for (int k = sub; k < KT; k += 8) {
+                float e = __expf(Gs[row * (KT + 1) + k] - m);
+                Gs[row * (KT + 1) + k] = e;
+                s += e;
+            }
+#pragma unroll
+            for (int off = 4; off; off >>= 1)
+                s += __shfl_xor(s, off, 8);
+            const float inv = 1.0f / s;
+            for (int k = sub; k < KT; k += 8)
+                Gs[row * (KT + 1) + k] *= inv;
+        }
+        __syncthreads();
+
+        // ---- phase A: S += gamma^T @ Xext
+#pragma unroll 1
+        for (int kk = 0; kk < CHUNK; kk += 2) {
+#pragma unroll
+            for (int t = 0; t < ST; ++t) {
+                const int ti = wave + 4 * (t / 5);
+                const int tj = t % 5;
+                if (ti < KT32) {
+                    const float a = Gs[(kk + lk) * (KT + 1) + ti * 32 + l31];
+                    const float b = Xs[(kk + lk) * XSTRIDE + tj * 32 + l31];
+                    sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                        a, b, sacc[t], 0, 0, 0);
+                }
+            }
+        }
+    }
+    __syncthreads();
+
+    // ---- epilogue: S0 broadcast, then FV from (S0, S1, S2)
+    // tile (ti, tj=4) col 128 = S0 column (j = l31 + 128 -> l31 == 0)
+#pragma unroll
+    for (int t = 0; t < ST; ++t) {
+        const int ti = wave + 4 * (t / 5);
+        const int tj = t % 5;
+        if (ti < KT32 && tj == 4 && l31 == 0) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                S0s[kcomp] = sacc[t][r];
+            }
+        }
+    }
+    __syncthreads();
+
+    // S2 tiles: tj in {0, 1} (cols 0..63 = x^2) ; S1 tiles: tj in {2, 3}
+    // For output element (kcomp, d): S2 at tile tj = d/32, S1 at
+    // tj = 2 + d/32 — same lane (l31 = d%32), same r. Pair them up.
+    const float invn = 1.0f / (float)n;
+#pragma unroll
+    for (int t = 0; t < ST; ++t) {
+        const int ti = wave + 4 * (t / 5);
+        const int tj = t % 5;
+        if (ti >= KT32 || tj >= 2) continue;      // S2 tiles drive the pair
+        const int d = tj * 32 + l31;
+        if (d >= D) continue;
+        // find the matching S1 tile (tj + 2) in this wave's tile list:
+        // flattened index t + 2 (same ti, tj+2) — valid because tiles are
+        // laid out tj-major within each ti.
+        const f32x16 s2 = sacc[t];
+        const f32x16 s1 = sacc[t + 2];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+            const float s0 = S0s[kcomp];
+            const float m = mu[kcomp * D + d];
+            const float v = var[kcomp * D + d];
+            const float wk = w[kcomp];
+            const float sig = sqrtf(v);
+            const float gm = (s1[r] - m * s0) / (sig * (float)n * sqrtf(wk));
+            const float gs = ((s2[r] - 2.0f * m * s1[r] + m * m * s0) / v
+                              - s0) * invn / sqrtf(2.0f * wk);
+            float* ob = out + (size_t)img * (2 * KT * D);
+            ob[kcomp * D + d] = gm;
+            ob[KT * D + kcomp * D + d] = gs;
+        }
+    }
+}
+
+KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
+                             const float* var, const float* w, float* out,
+                             int B, int n, int D, int K, hipStream_t stream) {
+    if (D > DT || (K != 64 && K != 128 && K != 256))
+        return (int)hipErrorInvalidValue;
+    size_t lds = (size_t)(CHUNK * XSTRIDE + CHUNK * (K + 1) + K) * 4;
+    dim3 grid(B), block(256);
+    switch (K) {
+        case 64:  hipLaunchKernelGGL(fv_mfma_kernel<64>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 128: hipLaunchKernelGGL(fv_mfma_kernel<128>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 256: hipLaunchKernelGGL(fv_mfma_kernel<256>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+    }
+    return (int)hipGetLastError();
+}

@@ -293,3 +293,20 @@ def test_solvers_gpu_match_cpu_fp32():
     Wc = torch.cat(w_cpu.blocks, 0).cpu()
     Wg = torch.cat([b.cpu() for b in w_gpu.blocks], 0)
     assert (Wc - Wg).abs().max().item() < 1e-2
+
+
+def test_fv_mfma_kernel_matches_gemm_path(hip):
+    from keystone_amd.ops.images.fisher_vector import FisherVector
+    from keystone_amd.ops.learning.gmm import GaussianMixtureModel
+    torch.manual_seed(21)
+    for (B, n, D, K) in [(4, 300, 64, 256), (3, 97, 48, 64), (2, 64, 64, 128)]:
+        X = torch.randn(B, n, D, device="cuda")
+        mu = torch.randn(K, D, device="cuda")
+        var = torch.rand(K, D, device="cuda") + 0.5
+        w = torch.softmax(torch.randn(K, device="cuda"), 0)
+        out = hip.fisher_vector_mfma(X, mu, var, w)
+        fv = FisherVector(GaussianMixtureModel(mu.cpu(), var.cpu(), w.cpu()))
+        ref = fv.apply_batch(X.cpu())
+        err = (out.cpu() - ref).abs().max().item()
+        assert out.shape == ref.shape
+        assert err < 5e-4, (B, n, D, K, err)
