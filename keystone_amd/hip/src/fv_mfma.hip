@@ -88,7 +88,11 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         f32x16 lacc[LT];
 #pragma unroll
         for (int t = 0; t < LT; ++t) lacc[t] = (f32x16)(0.0f);
-#pragma unroll 1
+        // unroll 10 (XC = 130 = 2*65 = 10*13 k-steps): the P loads are
+        // L2-latency (~200 cyc) feeding the MFMA directly — without the
+        // unroll each k-step serializes on its load (measured 0.7x the
+        // GEMM path); unrolled, loads pipeline ahead of the MFMA stream.
+#pragma unroll 5
         for (int kk = 0; kk < XC; kk += 2) {
             const float a = Xs[l31 * XSTRIDE + kk + lk];
 #pragma unroll
