@@ -40,6 +40,10 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     constexpr int KT32 = KT / 32;               // logits col-tiles
     constexpr int LT = (KT32 + 3) / 4;          // logits tiles per wave (max)
     constexpr int ST = 5 * ((KT32 + 3) / 4);    // stats tiles per wave (max)
+    // when KT32 is a multiple of 4 every wave's tiles are valid: the tile
+    // guards below become compile-time (a RUNTIME guard around each load
+    // makes hipcc branch + wait vmcnt(0) per element — guide §5 trap 4c)
+    constexpr bool FULL = (KT32 % 4 == 0);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* Xs = (float*)smem;                   // [CHUNK][XSTRIDE]
@@ -98,7 +102,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
 #pragma unroll
             for (int t = 0; t < LT; ++t) {
                 const int ct = wave + 4 * t;
-                if (ct < KT32) {
+                if (FULL || ct < KT32) {
                     const float b = P[(size_t)(kk + lk) * KT + ct * 32 + l31];
                     lacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
                         a, b, lacc[t], 0, 0, 0);
@@ -109,7 +113,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
 #pragma unroll
         for (int t = 0; t < LT; ++t) {
             const int ct = wave + 4 * t;
-            if (ct < KT32) {
+            if (FULL || ct < KT32) {
 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
                     const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
@@ -151,7 +155,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             for (int t = 0; t < ST; ++t) {
                 const int ti = wave + 4 * (t / 5);
                 const int tj = t % 5;
-                if (ti < KT32) {
+                if (FULL || ti < KT32) {
                     const float a = Gs[(kk + lk) * (KT + 1) + ti * 32 + l31];
                     const float b = Xs[(kk + lk) * XSTRIDE + tj * 32 + l31];
                     sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
