@@ -46,15 +46,21 @@ class FisherVector(Transformer):
         w = g.weights.to(device=X.device, dtype=X.dtype)       # [K]
         K = mu.shape[0]
 
-        # The fused per-image kernel (hip/src/gmm_fv.hip) is measured 16x
-        # SLOWER than this batched-GEMM formulation at FV-pipeline shapes
-        # (rocBLAS owns these GEMMs; the kernel's posterior phase is
-        # LDS-read-issue bound) — opt-in only, kept for small-batch regimes.
         import os
-        if (X.is_cuda and X.dtype == torch.float32 and self.improved_scaling
-                and os.environ.get("KEYSTONE_FV_KERNEL")):
+        if X.is_cuda and X.dtype == torch.float32 and self.improved_scaling:
             from ... import hip as hip_ops
-            if hip_ops.fv_supported(D, K):
+            # K3 v2 (fv_mfma.hip): fused logits-MFMA -> softmax -> moment-
+            # MFMA per image; never materializes gamma (10x less HBM
+            # traffic) and measured FASTER than the batched-GEMM path at
+            # the K=128/256 FV-vocabulary sizes (profiles/README.md).
+            if (hip_ops.fv_mfma_supported(D, K) and K in (128, 256)
+                    and not os.environ.get("KEYSTONE_FV_GEMM")):
+                out = hip_ops.fisher_vector_mfma(X.contiguous(), mu, var, w)
+                return out.squeeze(0) if squeeze else out
+            # first-generation fused kernel: measured slower than the GEMM
+            # formulation (LDS-read-issue bound) — kept opt-in
+            if (os.environ.get("KEYSTONE_FV_KERNEL")
+                    and hip_ops.fv_supported(D, K)):
                 out = hip_ops.fisher_vector(X.contiguous(), mu, var, w)
                 return out.squeeze(0) if squeeze else out
 
