@@ -46,9 +46,9 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     constexpr bool FULL = (KT32 % 4 == 0);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* Xs = (float*)smem;                   // [CHUNK][XSTRIDE]
-    float* Gs = Xs + CHUNK * XSTRIDE;           // [CHUNK][KT + 1] gamma
-    float* S0s = Gs + CHUNK * (KT + 1);         // [KT] broadcast
+    float* Xs0 = (float*)smem;                  // [2][CHUNK][XSTRIDE]
+    float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [2][CHUNK][KT + 1]
+    float* S0s = Gs0 + 2 * CHUNK * (KT + 1);    // [KT] broadcast
 
     const int img = blockIdx.x;
     const float* Xi = X + (size_t)img * n * D;
@@ -64,38 +64,61 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
 #pragma unroll
     for (int t = 0; t < ST; ++t) sacc[t] = (f32x16)(0.0f);
 
-    for (int c0 = 0; c0 < n; c0 += CHUNK) {
-        const int rows = min(CHUNK, n - c0);
-        __syncthreads();
-        // ---- stage Xext chunk: [x^2 | x | 1 | 0]
-        for (int i = tid; i < CHUNK * XCP; i += 256) {
-            const int r = i / XCP, c = i % XCP;
-            float v = 0.0f;
-            if (r < rows) {
-                if (c < DT) {
-                    if (c < D) {
-                        float x = Xi[(size_t)(c0 + r) * D + c];
-                        v = x * x;
-                    }
-                } else if (c < 2 * DT) {
-                    const int d = c - DT;
-                    if (d < D) v = Xi[(size_t)(c0 + r) * D + d];
-                } else if (c == 2 * DT) {
-                    v = 1.0f;
-                }
-            }
-            Xs[r * XSTRIDE + c] = v;
-        }
-        __syncthreads();
+    // Pipelined 2-deep schedule: chunk c's logits-MFMA and chunk c-1's
+    // stats-MFMA form ONE joint MFMA span per iteration (the co-resident
+    // wave covers the non-MFMA gaps), chunk c+1's global X loads are
+    // issued ahead of the span (T14), and only 3 barriers/iteration
+    // remain (MFMA span -> LDS writes -> softmax).
+    const int nc = (n + CHUNK - 1) / CHUNK;
 
-        // ---- phase L: logits = Xext @ P  (KT32 col-tiles over 4 waves)
+    // zero the pad columns (129..XCP-1) once; never touched again
+    for (int buf = 0; buf < 2; ++buf)
+        for (int i = tid; i < CHUNK * (XCP - XC + 1); i += 256) {
+            const int r = i / (XCP - XC + 1), c = XC - 1 + i % (XCP - XC + 1);
+            Xs0[(buf * CHUNK + r) * XSTRIDE + c] = 0.0f;
+        }
+
+    // per-thread staging slots: 8 x-elements (i = tid + 256*q)
+    float xreg[8];
+    auto stage_load = [&](int c0, int rows) {
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+            const int i = tid + 256 * q;
+            const int r = i / DT, d = i % DT;
+            xreg[q] = (r < rows && d < D)
+                ? Xi[(size_t)(c0 + r) * D + d] : 0.0f;
+        }
+    };
+    auto stage_write = [&](float* Xs, int rows) {
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+            const int i = tid + 256 * q;
+            const int r = i / DT, d = i % DT;
+            Xs[r * XSTRIDE + d] = xreg[q] * xreg[q];
+            Xs[r * XSTRIDE + DT + d] = xreg[q];
+        }
+        if (tid < CHUNK)
+            Xs[tid * XSTRIDE + 2 * DT] = (tid < rows) ? 1.0f : 0.0f;
+    };
+
+    stage_load(0, min(CHUNK, n));
+    stage_write(Xs0, min(CHUNK, n));
+    __syncthreads();
+
+    for (int c = 0; c < nc; ++c) {
+        const int cur = c & 1;
+        float* Xs = Xs0 + cur * CHUNK * XSTRIDE;
+        float* Gs = Gs0 + cur * CHUNK * (KT + 1);
+        float* Xp = Xs0 + (cur ^ 1) * CHUNK * XSTRIDE;
+        float* Gp = Gs0 + (cur ^ 1) * CHUNK * (KT + 1);
+        const bool more = c + 1 < nc;
+        if (more)
+            stage_load((c + 1) * CHUNK, min(CHUNK, n - (c + 1) * CHUNK));
+
+        // ---- joint MFMA span: logits(c) + stats(c-1)
         f32x16 lacc[LT];
 #pragma unroll
         for (int t = 0; t < LT; ++t) lacc[t] = (f32x16)(0.0f);
-        // unroll 10 (XC = 130 = 2*65 = 10*13 k-steps): the P loads are
-        // L2-latency (~200 cyc) feeding the MFMA directly — without the
-        // unroll each k-step serializes on its load (measured 0.7x the
-        // GEMM path); unrolled, loads pipeline ahead of the MFMA stream.
 #pragma unroll 5
         for (int kk = 0; kk < XC; kk += 2) {
             const float a = Xs[l31 * XSTRIDE + kk + lk];
@@ -109,7 +132,25 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 }
             }
         }
-        // write logits into Gs (C layout: col = l31, row = (r&3)+8*(r>>2)+4*lk)
+        if (c > 0) {
+#pragma unroll 1
+            for (int kk = 0; kk < CHUNK; kk += 2) {
+#pragma unroll
+                for (int t = 0; t < ST; ++t) {
+                    const int ti = wave + 4 * (t / 5);
+                    const int tj = t % 5;
+                    if (FULL || ti < KT32) {
+                        const float a = Gp[(kk + lk) * (KT + 1) + ti * 32 + l31];
+                        const float b = Xp[(kk + lk) * XSTRIDE + tj * 32 + l31];
+                        sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                            a, b, sacc[t], 0, 0, 0);
+                    }
+                }
+            }
+        }
+        __syncthreads();   // stats done reading Xp/Gp; logits accs private
+
+        // ---- LDS writes: logits(c) -> Gs, staged chunk c+1 -> Xp slot
 #pragma unroll
         for (int t = 0; t < LT; ++t) {
             const int ct = wave + 4 * t;
@@ -121,9 +162,11 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 }
             }
         }
+        if (more)
+            stage_write(Xp, min(CHUNK, n - (c + 1) * CHUNK));
         __syncthreads();
 
-        // ---- phase S: row softmax, 8 lanes per row (rows = tid >> 3)
+        // ---- softmax(c) on Gs, 8 lanes per row
         {
             const int row = tid >> 3;
             const int sub = tid & 7;
@@ -147,8 +190,13 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 Gs[row * (KT + 1) + k] *= inv;
         }
         __syncthreads();
+    }
 
-        // ---- phase A: S += gamma^T @ Xext
+    // ---- trailing stats for the last chunk
+    {
+        const int cur = (nc - 1) & 1;
+        float* Xl = Xs0 + cur * CHUNK * XSTRIDE;
+        float* Gl = Gs0 + cur * CHUNK * (KT + 1);
 #pragma unroll 1
         for (int kk = 0; kk < CHUNK; kk += 2) {
 #pragma unroll
@@ -156,8 +204,8 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 const int ti = wave + 4 * (t / 5);
                 const int tj = t % 5;
                 if (FULL || ti < KT32) {
-                    const float a = Gs[(kk + lk) * (KT + 1) + ti * 32 + l31];
-                    const float b = Xs[(kk + lk) * XSTRIDE + tj * 32 + l31];
+                    const float a = Gl[(kk + lk) * (KT + 1) + ti * 32 + l31];
+                    const float b = Xl[(kk + lk) * XSTRIDE + tj * 32 + l31];
                     sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
                         a, b, sacc[t], 0, 0, 0);
                 }
@@ -221,7 +269,8 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              int B, int n, int D, int K, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(CHUNK * XSTRIDE + CHUNK * (K + 1) + K) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K) * 4;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(B), block(256);
     switch (K) {
         case 64:  hipLaunchKernelGGL(fv_mfma_kernel<64>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;

@@ -49,11 +49,11 @@ class FisherVector(Transformer):
         import os
         if X.is_cuda and X.dtype == torch.float32 and self.improved_scaling:
             from ... import hip as hip_ops
-            # K3 v2 (fv_mfma.hip): fused logits-MFMA -> softmax -> moment-
-            # MFMA per image; never materializes gamma (10x less HBM
-            # traffic) and measured FASTER than the batched-GEMM path at
-            # the K=128/256 FV-vocabulary sizes (profiles/README.md).
-            if (hip_ops.fv_mfma_supported(D, K) and K in (128, 256)
+            # K3 v2 (fv_mfma.hip): pipelined fused logits-MFMA -> softmax
+            # -> moment-MFMA per image; never materializes gamma (10x less
+            # HBM traffic), measured 1.6x the batched-GEMM path at K=256
+            # (profiles/README.md).
+            if (hip_ops.fv_mfma_supported(D, K)
                     and not os.environ.get("KEYSTONE_FV_GEMM")):
                 out = hip_ops.fisher_vector_mfma(X.contiguous(), mu, var, w)
                 return out.squeeze(0) if squeeze else out

@@ -23,7 +23,10 @@ for (B, n, D, K) in [(2048, 2900, 64, 256), (8192, 2900, 64, 256),
     w = torch.softmax(torch.randn(K, device="cuda"), 0)
     fv = FisherVector(GaussianMixtureModel(mu, var, w))
     t_mfma = bench(lambda: hip.fisher_vector_mfma(X, mu, var, w))
+    import os
+    os.environ["KEYSTONE_FV_GEMM"] = "1"
     t_gemm = bench(lambda: fv._encode(X), iters=5)
+    del os.environ["KEYSTONE_FV_GEMM"]
     print(f"B{B} n{n} D{D} K{K}: mfma {t_mfma*1e3:.1f} ms, gemm {t_gemm*1e3:.1f} ms, "
           f"speedup {t_gemm/t_mfma:.2f}x")
     del X
