@@ -104,6 +104,11 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_gmm_stats_f32.argtypes = [c_p] * 3 + [c_i] * 4 + [c_p]
+        lib.ks_gmm_stats_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
@@ -260,18 +265,70 @@ def fisher_vector_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
     var = _prep(var, "var")
     w = _prep(w, "w")
     a, b, c = _gmm_coeffs(mu, var, w)
-    DT = 64
-    P = torch.zeros(2 * DT + 32, K, dtype=torch.float32, device=X.device)
-    P[:D] = -a.T
-    P[DT:DT + D] = b.T
-    P[2 * DT] = -c
+    P = _fv_P(a, b, c, K, X.device)
     out = torch.empty(B, 2 * K * D, dtype=torch.float32, device=X.device)
     _check(lib.ks_fv_mfma_f32(
-        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.contiguous().data_ptr()),
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
         ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
         ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         B, n, D, K, _stream()), "ks_fv_mfma_f32")
     return out
+
+
+def _fv_P(a: torch.Tensor, b: torch.Tensor, c: torch.Tensor, K: int,
+          device) -> torch.Tensor:
+    """Padded [XCP=160, K] operand for fv_mfma: [-a.T ; b.T ; -c ; 0]."""
+    DT = 64
+    D = a.shape[1]
+    P = torch.zeros(2 * DT + 32, K, dtype=torch.float32, device=device)
+    P[:D] = -a.T
+    P[DT:DT + D] = b.T
+    P[2 * DT] = -c
+    return P.contiguous()
+
+
+def gmm_stats_into(X: torch.Tensor, P: torch.Tensor, out: torch.Tensor):
+    """Low-level fused E-step: preallocated P [160, K] (see ``_fv_P``) and
+    partials buffer out [nb, K + 2*K*D + 1] -> (Nk, Sx, Sxx, ll) fp64 views
+    of one contiguous sum (no per-call allocations — EM-loop hot path)."""
+    lib = _load()
+    n, D = X.shape
+    K = P.shape[1]
+    nb = out.shape[0]
+    _check(lib.ks_gmm_stats_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _stream()),
+        "ks_gmm_stats_f32")
+    s = out.sum(0, dtype=torch.float64)
+    return (s[:K], s[K:K + K * D].view(K, D),
+            s[K + K * D:K + 2 * K * D].view(K, D), s[K + 2 * K * D:])
+
+
+def gmm_stats_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
+                   w: torch.Tensor, n_blocks: int = 512):
+    """GMM EM E-step sufficient statistics in ONE fused launch
+    (fv_mfma.hip STATS mode): X [n, D] -> (Nk [K], Sx [K, D], Sxx [K, D],
+    ll [1]) in fp64, never materializing gamma [n, K].  Grid-strided over
+    32-row chunks; per-block fp32 partials are summed here in fp64."""
+    lib = _load()
+    X = _prep(X, "gmm_stats.X")
+    n, D = X.shape
+    K = mu.shape[0]
+    a, b, c = _gmm_coeffs(_prep(mu, "mu"), _prep(var, "var"), _prep(w, "w"))
+    P = _fv_P(a, b, c, K, X.device)
+    nb = min(n_blocks, max(1, (n + 31) // 32))
+    width = K + 2 * K * D + 1
+    out = torch.empty(nb, width, dtype=torch.float32, device=X.device)
+    _check(lib.ks_gmm_stats_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _stream()),
+        "ks_gmm_stats_f32")
+    s = out.sum(0, dtype=torch.float64)
+    Nk = s[:K].contiguous()
+    Sx = s[K:K + K * D].view(K, D).contiguous()
+    Sxx = s[K + K * D:K + 2 * K * D].view(K, D).contiguous()
+    ll = s[K + 2 * K * D:].contiguous()
+    return Nk, Sx, Sxx, ll
 
 
 def fisher_vector(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,

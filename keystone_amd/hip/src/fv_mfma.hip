@@ -28,7 +28,11 @@
 #define XSTRIDE (XCP + 1)    // LDS row stride: (161*r + c) % 32 distinct
 #define CHUNK 32
 
-template <int KT>
+// STATS mode: X is ONE flat [n, D] matrix processed grid-strided; block b
+// writes its PARTIAL sufficient statistics to out[b] = [S0 (KT) | S1
+// (KT*D) | S2 (KT*D) | ll (1)] — the GMM EM E-step in one launch (sum the
+// partials host-side).  FV mode (STATS=false): one image per block.
+template <int KT, bool STATS>
 __global__ __launch_bounds__(256, 1)
 void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     const float* __restrict__ P,    // [XCP, KT] padded
@@ -48,10 +52,10 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* Xs0 = (float*)smem;                  // [2][CHUNK][XSTRIDE]
     float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [2][CHUNK][KT + 1]
-    float* S0s = Gs0 + 2 * CHUNK * (KT + 1);    // [KT] broadcast
+    float* S0s = Gs0 + 2 * CHUNK * (KT + 1);    // [KT] broadcast (+ll)
 
     const int img = blockIdx.x;
-    const float* Xi = X + (size_t)img * n * D;
+    const float* Xi = STATS ? X : X + (size_t)img * n * D;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
@@ -69,7 +73,24 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // wave covers the non-MFMA gaps), chunk c+1's global X loads are
     // issued ahead of the span (T14), and only 3 barriers/iteration
     // remain (MFMA span -> LDS writes -> softmax).
-    const int nc = (n + CHUNK - 1) / CHUNK;
+    // chunk index -> starting row (grid-strided in STATS mode)
+    const int nc_all = (n + CHUNK - 1) / CHUNK;
+    const int nc = STATS
+        ? (nc_all - blockIdx.x + (int)gridDim.x - 1) / (int)gridDim.x
+        : nc_all;
+    auto chunk_start = [&](int i) {
+        return STATS ? (blockIdx.x + (size_t)i * gridDim.x) * CHUNK
+                     : (size_t)i * CHUNK;
+    };
+    if (STATS && tid == 0) S0s[KT] = 0.0f;      // ll accumulator
+    if (nc == 0) {
+        if (STATS) {
+            __syncthreads();
+            float* ob = out + (size_t)img * (KT + 2 * KT * D + 1);
+            for (int i = tid; i < KT + 2 * KT * D + 1; i += 256) ob[i] = 0.0f;
+        }
+        return;
+    }
 
     // zero the pad columns (129..XCP-1) once; never touched again
     for (int buf = 0; buf < 2; ++buf)
@@ -101,8 +122,8 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             Xs[tid * XSTRIDE + 2 * DT] = (tid < rows) ? 1.0f : 0.0f;
     };
 
-    stage_load(0, min(CHUNK, n));
-    stage_write(Xs0, min(CHUNK, n));
+    stage_load((int)chunk_start(0), min(CHUNK, n - (int)chunk_start(0)));
+    stage_write(Xs0, min(CHUNK, n - (int)chunk_start(0)));
     __syncthreads();
 
     for (int c = 0; c < nc; ++c) {
@@ -113,7 +134,8 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         float* Gp = Gs0 + (cur ^ 1) * CHUNK * (KT + 1);
         const bool more = c + 1 < nc;
         if (more)
-            stage_load((c + 1) * CHUNK, min(CHUNK, n - (c + 1) * CHUNK));
+            stage_load((int)chunk_start(c + 1),
+                       min(CHUNK, n - (int)chunk_start(c + 1)));
 
         // ---- joint MFMA span: logits(c) + stats(c-1)
         f32x16 lacc[LT];
@@ -163,7 +185,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             }
         }
         if (more)
-            stage_write(Xp, min(CHUNK, n - (c + 1) * CHUNK));
+            stage_write(Xp, min(CHUNK, n - (int)chunk_start(c + 1)));
         __syncthreads();
 
         // ---- softmax(c) on Gs, 8 lanes per row
@@ -188,6 +210,9 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             const float inv = 1.0f / s;
             for (int k = sub; k < KT; k += 8)
                 Gs[row * (KT + 1) + k] *= inv;
+            if (STATS && sub == 0
+                    && row < n - (int)chunk_start(c))
+                atomicAdd(&S0s[KT], m + __logf(s));   // logsumexp sum
         }
         __syncthreads();
     }
@@ -213,6 +238,40 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         }
     }
     __syncthreads();
+
+    if (STATS) {
+        // ---- epilogue (STATS): write raw partials
+        // ob = [S0 (KT) | S1 (KT*D) | S2 (KT*D) | ll]
+        float* ob = out + (size_t)img * (KT + 2 * KT * D + 1);
+#pragma unroll
+        for (int t = 0; t < ST; ++t) {
+            const int ti = wave + 4 * (t / 5);
+            const int tj = t % 5;
+            if (!(FULL || ti < KT32)) continue;
+            if (tj == 4) {
+                if (l31 == 0) {
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) {
+                        const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                        ob[kcomp] = sacc[t][r];
+                    }
+                }
+            } else {
+                const int d = (tj & 1) * 32 + l31;
+                if (d < D) {
+                    const bool is_s2 = tj < 2;   // x^2 columns
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) {
+                        const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                        ob[KT + (is_s2 ? KT * D : 0) + kcomp * D + d] =
+                            sacc[t][r];
+                    }
+                }
+            }
+        }
+        if (tid == 0) ob[KT + 2 * KT * D] = S0s[KT];
+        return;
+    }
 
     // ---- epilogue: S0 broadcast, then FV from (S0, S1, S2)
     // tile (ti, tj=4) col 128 = S0 column (j = l31 + 128 -> l31 == 0)
@@ -269,13 +328,31 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              int B, int n, int D, int K, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(B), block(256);
     switch (K) {
-        case 64:  hipLaunchKernelGGL(fv_mfma_kernel<64>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 128: hipLaunchKernelGGL(fv_mfma_kernel<128>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 256: hipLaunchKernelGGL(fv_mfma_kernel<256>, grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 64:  hipLaunchKernelGGL((fv_mfma_kernel<64, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 128: hipLaunchKernelGGL((fv_mfma_kernel<128, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<256, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+    }
+    return (int)hipGetLastError();
+}
+
+// GMM E-step sufficient statistics over ONE flat [n, D] matrix:
+// out [n_blocks, K + 2*K*D + 1] partials (sum over dim 0 host-side).
+KS_EXPORT int ks_gmm_stats_f32(const float* X, const float* P, float* out,
+                               int n_blocks, int n, int D, int K,
+                               hipStream_t stream) {
+    if (D > DT || (K != 64 && K != 128 && K != 256))
+        return (int)hipErrorInvalidValue;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K + 1) * 4;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
+    dim3 grid(n_blocks), block(256);
+    switch (K) {
+        case 64:  hipLaunchKernelGGL((fv_mfma_kernel<64, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 128: hipLaunchKernelGGL((fv_mfma_kernel<128, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<256, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
     }
     return (int)hipGetLastError();
 }

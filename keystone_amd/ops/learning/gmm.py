@@ -109,12 +109,23 @@ class GaussianMixtureModelEstimator(Estimator):
 
         t = tick("global_stats", t)
         model = GaussianMixtureModel(mu, var, w)
+        # Fused E-step (fv_mfma.hip STATS mode): logits-MFMA -> softmax ->
+        # moment-MFMA grid-strided over the shard, gamma [n, K] never hits
+        # HBM — one launch replaces the 5-kernel eager chain (and the
+        # [n, 2D] Xext hoist below is not needed at all).
+        import os
+        from ... import hip as hip_ops
+        use_fused = (X.is_cuda and X.dtype == torch.float32
+                     and hip_ops.fv_mfma_supported(D, K)
+                     and not os.environ.get("KEYSTONE_GMM_EAGER"))
+
         # Hoist the extended operand [X^2 | X] out of the EM loop: the
         # E-step logits become ONE addmm against the stacked coefficient
         # matrix [-1/2v ; mu/v] (2 HBM-resident copies of X buy ~half the
         # per-iteration kernel launches), and X2 is reused by Sxx.
-        X2 = X * X
-        Xext = torch.cat([X2, X], dim=1)                 # [n, 2D]
+        if not use_fused:
+            X2 = X * X
+            Xext = torch.cat([X2, X], dim=1)             # [n, 2D]
 
         # The per-iteration LOCAL compute (logit addmm -> softmax -> local
         # sufficient statistics) is a fixed launch-bound kernel chain —
@@ -158,7 +169,7 @@ class GaussianMixtureModelEstimator(Estimator):
         # stays eager; revisit when torch's ROCm graph pools can be
         # reliably returned to the main allocator.
         graph = None
-        if X.is_cuda and os.environ.get("KEYSTONE_GMM_GRAPHS"):
+        if X.is_cuda and not use_fused and os.environ.get("KEYSTONE_GMM_GRAPHS"):
             try:
                 fill_coeffs(model)
                 side = torch.cuda.Stream()
@@ -174,16 +185,35 @@ class GaussianMixtureModelEstimator(Estimator):
             except Exception:
                 graph = None
 
+        if use_fused:
+            nb = min(512, max(1, (X.shape[0] + 31) // 32))
+            st_out = torch.empty(nb, K + 2 * K * D + 1,
+                                 dtype=torch.float32, device=dev)
+            st_P = torch.zeros(160, K, dtype=torch.float32, device=dev)
+
+            def fill_P(mdl):
+                inv2v = 0.5 / mdl.variances
+                st_P[:D].copy_(-inv2v.T)
+                st_P[64:64 + D].copy_((mdl.means / mdl.variances).T)
+                st_P[128].copy_(mdl.weights.clamp_min(1e-30).log()
+                                - (mdl.means * mdl.means * inv2v).sum(1)
+                                - 0.5 * mdl.variances.log().sum(1)
+                                - 0.5 * D * math.log(2 * math.pi))
+
         prev_ll = -float("inf")
         for _it in range(self.max_iterations):
-            fill_coeffs(model)
-            if graph is not None:
+            if use_fused:
+                fill_P(model)
+                Nk, Sx, Sxx, ll = hip_ops.gmm_stats_into(X, st_P, st_out)
+            elif graph is not None:
+                fill_coeffs(model)
                 graph.replay()
                 Nk = g_out[0].clone()
                 Sx = g_out[1].clone()
                 Sxx = g_out[2].clone()
                 ll = g_out[3].clone()
             else:
+                fill_coeffs(model)
                 Nk, Sx, Sxx, ll = estep_local()
             rt.all_reduce_many_([Nk, Sx, Sxx, ll])
             Nk_c = Nk.clamp_min(1e-10).unsqueeze(1)

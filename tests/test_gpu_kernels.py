@@ -310,3 +310,51 @@ def test_fv_mfma_kernel_matches_gemm_path(hip):
         err = (out.cpu() - ref).abs().max().item()
         assert out.shape == ref.shape
         assert err < 5e-4, (B, n, D, K, err)
+
+
+def test_gmm_stats_kernel_matches_eager(hip):
+    import math
+    torch.manual_seed(22)
+    for (n, D, K) in [(10000, 64, 256), (4097, 48, 64), (63, 64, 128)]:
+        X = torch.randn(n, D, device="cuda")
+        mu = torch.randn(K, D, device="cuda")
+        var = torch.rand(K, D, device="cuda") + 0.5
+        w = torch.softmax(torch.randn(K, device="cuda"), 0)
+        Nk, Sx, Sxx, ll = hip.gmm_stats_mfma(X, mu, var, w)
+        # eager reference (same formulation as gmm.estep_local)
+        inv2v = 0.5 / var
+        cc = ((mu * mu * inv2v).sum(1) + 0.5 * var.log().sum(1)
+              + 0.5 * D * math.log(2 * math.pi) - w.clamp_min(1e-30).log())
+        lp = -(X * X) @ inv2v.T + X @ (mu / var).T - cc
+        lse = torch.logsumexp(lp, dim=1)
+        gamma = torch.exp(lp - lse.unsqueeze(1))
+        rNk = gamma.sum(0, dtype=torch.float64)
+        rSx = (gamma.T @ X).to(torch.float64)
+        rSxx = (gamma.T @ (X * X)).to(torch.float64)
+        rll = lse.sum(dtype=torch.float64)
+        for got, ref, name, tol in [(Nk, rNk, "Nk", 2e-3),
+                                    (Sx, rSx, "Sx", 5e-3),
+                                    (Sxx, rSxx, "Sxx", 2e-2),
+                                    (ll, rll.reshape(1), "ll", 2e-3)]:
+            err = (got - ref).abs().max().item()
+            scale = ref.abs().max().item() + 1.0
+            assert err / scale < tol, (n, D, K, name, err, scale)
+
+
+def test_gmm_estimator_uses_fused_estep_on_gpu():
+    # fused-stats fit vs forced-eager fit: same model (both fp32 E-steps)
+    import os
+    import keystone_amd as ka
+    from keystone_amd.ops.learning.gmm import GaussianMixtureModelEstimator
+    torch.manual_seed(23)
+    X = torch.randn(20000, 64, device="cuda")
+    ds = ka.from_rows(X)
+    est = GaussianMixtureModelEstimator(64, max_iterations=8, seed=0)
+    m_fused = est._fit(ds)
+    os.environ["KEYSTONE_GMM_EAGER"] = "1"
+    try:
+        m_eager = est._fit(ds)
+    finally:
+        del os.environ["KEYSTONE_GMM_EAGER"]
+    assert (m_fused.means - m_eager.means).abs().max().item() < 1e-2
+    assert (m_fused.weights - m_eager.weights).abs().max().item() < 1e-3
