@@ -99,8 +99,12 @@ class KMeansPlusPlusEstimator(Estimator):
             newC = torch.where(cnt.unsqueeze(1) > 0,
                                sums / cnt.clamp_min(1.0).unsqueeze(1),
                                C.to(torch.float64)).to(dt)
-            shift = float(((newC - C) ** 2).sum())
-            C = newC
-            if shift <= self.tol:
-                break
+            # convergence check every 4 iters: .item() is a device sync
+            if (_it & 3) == 3 or _it == self.max_iterations - 1:
+                shift = float(((newC - C) ** 2).sum())
+                C = newC
+                if shift <= self.tol:
+                    break
+            else:
+                C = newC
         return KMeansModel(C)
