@@ -50,6 +50,19 @@ class KMeansPlusPlusEstimator(Estimator):
         self.tol = tol
 
     def _fit(self, data: TensorDataset) -> KMeansModel:
+        import os
+        import time
+        dbg = os.environ.get("KEYSTONE_GMM_DEBUG")
+
+        def tick(name, t0):
+            if dbg:
+                if data.data.is_cuda:
+                    torch.cuda.synchronize()
+                print(f"[kmeans] {name}: {time.perf_counter() - t0:.4f}s",
+                      flush=True)
+            return time.perf_counter()
+
+        t = time.perf_counter()
         X = data.data
         rt = data.runtime
         K = self.num_clusters
@@ -61,6 +74,7 @@ class KMeansPlusPlusEstimator(Estimator):
         sample = data.sample_local(max(8 * K // max(1, rt.world_size), K),
                                    self.seed)
         S = sample.collect().to(torch.float64).cpu()
+        t = tick("sample_collect", t)
         # chunked greedy (k-means|| flavour): draw `chunk` centers per
         # D^2-weighted multinomial round instead of one — 256 sequential
         # CPU steps at K=256 cost ~0.5 s, 32 chunked rounds ~20 ms, and
@@ -82,6 +96,7 @@ class KMeansPlusPlusEstimator(Estimator):
             d2 = torch.minimum(d2, dists.min(dim=1).values)
             k += take
         C = centers.to(device=dev, dtype=dt)
+        t = tick("plusplus_init", t)
 
         # --- Lloyd iterations.  Per-centroid (sum, count) via a one-hot
         # GEMM (assignment matrix^T @ X) — a large fp64 index_add_ runs at
@@ -107,4 +122,5 @@ class KMeansPlusPlusEstimator(Estimator):
                     break
             else:
                 C = newC
+        tick(f"lloyd({_it + 1} iters)", t)
         return KMeansModel(C)
