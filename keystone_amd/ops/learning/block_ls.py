@@ -319,6 +319,21 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         self.woodbury_threshold = woodbury_threshold
 
     def _fit(self, data: Dataset, labels: TensorDataset) -> BlockLinearMapper:
+        import os
+        import time
+        dbg = os.environ.get("KEYSTONE_SOLVER_DEBUG")
+        _ph: dict = {}
+
+        def tick(name, t0):
+            if dbg:
+                if labels.data.is_cuda:
+                    torch.cuda.synchronize()
+                now = time.perf_counter()
+                _ph[name] = _ph.get(name, 0.0) + (now - t0)
+                return now
+            return t0
+
+        t0 = time.perf_counter()
         getters, rt, n_local = _as_block_getters(data, self.block_size)
         Y = labels.data
         if Y.dim() == 1:
@@ -368,6 +383,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         nb = len(blocks)
         base_w = (1.0 - alpha) / n_tot
         sdt = _solve_dtype(dtype)
+        t0 = tick("setup", t0)
 
         # ---- batched base-factorization pre-pass.  The per-block base
         # matrix B_b = base_w * X_b^T X_b + lam I does NOT depend on the
@@ -389,11 +405,13 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             for b, Xb in enumerate(blocks):
                 torch.matmul(Xb.T.to(sdt), Xb.to(sdt), out=grams_pre[b])
             rt.all_reduce_(grams_pre)
+            t0 = tick("pre_grams", t0)
             eye = torch.eye(db0, dtype=sdt, device=X_full.device)
             Bb = base_w * grams_pre + self.lam * eye
             Lb = torch.linalg.cholesky(Bb)
             Binvs = torch.cholesky_inverse(Lb)
             del Bb, Lb
+            t0 = tick("pre_chol", t0)
 
         Ws: List[Optional[torch.Tensor]] = [None] * nb
         R = Y.clone()
@@ -401,6 +419,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             for b in range(nb):
                 Xb = blocks[b]
                 db = Xb.shape[1]
+                t0 = time.perf_counter()
                 T = R if Ws[b] is None else R + Xb @ Ws[b]
                 xtr_all = Xb.T @ T                     # [db, K]
                 if grams_pre is not None:
@@ -409,6 +428,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                 else:
                     gram_all = Xb.T @ Xb               # shared Gram
                     rt.all_reduce_many_([gram_all, xtr_all])
+                t0 = tick("xtr", t0)
 
                 Wb = torch.zeros(db, K, dtype=dtype, device=Xb.device)
                 if own:
@@ -443,6 +463,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     rhs = base_w * xtr_all[:, own].to(sdt) \
                         + seg[own].T * wc_t
                     RB = Binv @ rhs                          # B^{-1} rhs
+                    t0 = tick("rhs", t0)
                     col = {c: j for j, c in enumerate(own)}
 
                     by_nc: dict = {}
@@ -487,6 +508,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             sol = (RB[:, js].T - corr).to(dtype)  # [G, db]
                             Wb[:, torch.tensor(cls_list,
                                                device=Wb.device)] = sol.T
+                            t0 = tick("woodbury", t0)
                         else:
                             # large n_c: direct per-class factorization
                             B_dir = base_w * gram_all.to(sdt) \
@@ -503,4 +525,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     rt.all_reduce_(Wb)   # disjoint columns -> sum == gather
                 R = T - Xb @ Wb
                 Ws[b] = Wb
+                t0 = tick("update", t0)
+        if dbg:
+            print("[solver] " + " ".join(f"{k}={v:.4f}s"
+                                         for k, v in _ph.items()), flush=True)
         return BlockLinearMapper(Ws)
