@@ -413,6 +413,41 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             del Bb, Lb
             t0 = tick("pre_chol", t0)
 
+        # ---- loop-invariant per-class group tensors.  Everything here
+        # depends only on the (static) class layout, but used to be rebuilt
+        # from Python lists inside the block loop — ~2 ms of host work per
+        # block per epoch at K=1000 (measured, KEYSTONE_SOLVER_DEBUG).
+        groups = own_t = wc_t = None
+        wcs = {}
+        if own:
+            dev_ = X_full.device
+            own_t = torch.tensor(own, device=dev_)
+            wcs = {c: alpha / float(counts_cpu[c]) for c in own}
+            wc_t = torch.tensor([wcs[c] for c in own], dtype=sdt, device=dev_)
+            col = {c: j for j, c in enumerate(own)}
+            by_nc: dict = {}
+            for c in own:
+                s_c, e_c = cls_slice[c]
+                by_nc.setdefault(e_c - s_c, []).append(c)
+            groups = []
+            for n_c, cls_list in by_nc.items():
+                js_l = [col[c] for c in cls_list]
+                js = torch.tensor(js_l, device=dev_)
+                cls_idx = torch.tensor(cls_list, device=dev_)
+                if n_c == 0:
+                    groups.append((0, cls_list, js_l, js, cls_idx,
+                                   None, None, None))
+                    continue
+                s_t = torch.tensor([cls_slice[c][0] for c in cls_list],
+                                   device=dev_)
+                row_idx = (s_t.unsqueeze(1)
+                           + torch.arange(n_c, device=dev_)).reshape(-1)
+                w_row = torch.tensor([wcs[c] for c in cls_list], dtype=sdt,
+                                     device=dev_).sqrt().repeat_interleave(n_c)
+                eye_nc = torch.eye(n_c, dtype=sdt, device=dev_)
+                groups.append((n_c, cls_list, js_l, js, cls_idx, row_idx,
+                               w_row, eye_nc))
+
         Ws: List[Optional[torch.Tensor]] = [None] * nb
         R = Y.clone()
         for _epoch in range(self.num_iter):
@@ -450,7 +485,6 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                         # become GEMMs (hipBLASLt) instead of TRSMs
                         Binv = torch.cholesky_inverse(L)
                     Xb64 = Xb.to(sdt)
-                    wcs = {c: alpha / float(counts_cpu[c]) for c in own}
                     # rhs for ALL classes in two fused passes: the class-c
                     # correction X_c^T T[rows_c, c] is a segmented
                     # reduction over class-sorted rows (one index_add_).
@@ -458,65 +492,54 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     seg = torch.zeros(K, db, dtype=sdt,
                                       device=Xb.device)
                     seg.index_add_(0, cls, Xb64 * v)
-                    wc_t = torch.tensor([wcs[c] for c in own],
-                                        dtype=sdt, device=Xb.device)
-                    rhs = base_w * xtr_all[:, own].to(sdt) \
-                        + seg[own].T * wc_t
+                    rhs = base_w * xtr_all.index_select(1, own_t).to(sdt) \
+                        + seg.index_select(0, own_t).T * wc_t
                     RB = Binv @ rhs                          # B^{-1} rhs
                     t0 = tick("rhs", t0)
-                    col = {c: j for j, c in enumerate(own)}
 
-                    by_nc: dict = {}
-                    for c in own:
-                        s, e = cls_slice[c]
-                        by_nc.setdefault(e - s, []).append(c)
-                    for n_c, cls_list in by_nc.items():
-                        js = [col[c] for c in cls_list]
+                    for (n_c, cls_list, js_l, js, cls_idx, row_idx, w_row,
+                         eye_nc) in groups:
                         if n_c == 0:
-                            Wb[:, torch.tensor(cls_list, device=Wb.device)] \
-                                = RB[:, js].to(dtype)
+                            Wb[:, cls_idx] = RB[:, js].to(dtype)
                             continue
                         if n_c <= self.woodbury_threshold * db:
                             G_n = len(cls_list)
                             # one vectorized gather builds every class's
                             # sqrt(w_c)-scaled rows (a per-class cat loop
                             # costs ~1000 kernel launches per block)
-                            s_t = torch.tensor(
-                                [cls_slice[c][0] for c in cls_list],
-                                device=Xb.device)
-                            row_idx = (s_t.unsqueeze(1)
-                                       + torch.arange(n_c, device=Xb.device)
-                                       ).reshape(-1)            # [G*nc]
-                            w_row = torch.tensor(
-                                [wcs[c] for c in cls_list], dtype=sdt,
-                                device=Xb.device).sqrt().repeat_interleave(n_c)
-                            U = (Xb64[row_idx] * w_row.unsqueeze(1)).T
-                            # [db, G*nc], class-major blocks of nc columns
-                            UBf = Binv @ U                       # [db, G*nc]
-                            Ub = U.T.reshape(G_n, n_c, db)       # [G, nc, db]
-                            UBb = UBf.T.reshape(G_n, n_c, db)
-                            S = torch.eye(n_c, dtype=sdt,
-                                          device=Xb.device) \
-                                + Ub @ UBb.transpose(1, 2)       # [G, nc, nc]
-                            r = RB[:, js].T.unsqueeze(2)         # [G, db, 1]
-                            t = Ub @ r                           # [G, nc, 1]
+                            Urows = Xb64[row_idx] * w_row.unsqueeze(1)
+                            t0 = tick("wb_ubuild", t0)
+                            # Binv is SYMMETRIC, so U^T B^{-1} == (B^{-1}U)^T:
+                            # ONE row-major NN GEMM gives every class's
+                            # B^{-1}-applied rows contiguously (the former
+                            # [db, G*nc] transposed-operand GEMM ran ~0.6x
+                            # the NN rate and forced a reshape copy)
+                            UBr = Urows @ Binv                   # [G*nc, db]
+                            t0 = tick("wb_gemm", t0)
+                            Ub = Urows.reshape(G_n, n_c, db)     # [G, nc, db]
+                            UBb = UBr.reshape(G_n, n_c, db)
+                            rG = RB.T[js].contiguous()           # [G, db]
+                            S = eye_nc + Ub @ UBb.transpose(1, 2)
+                            # t/corr are GEMV-shaped (N=1): fused
+                            # multiply-reduce beats 1000-batch bmm
+                            t = (Ub * rG.unsqueeze(1)).sum(2, keepdim=True)
+                            t0 = tick("wb_smallbmm", t0)
                             if n_c == 1:
                                 z = t / S
                             else:
                                 z = torch.linalg.solve(S, t)
-                            corr = (UBb.transpose(1, 2) @ z).squeeze(2)
-                            sol = (RB[:, js].T - corr).to(dtype)  # [G, db]
-                            Wb[:, torch.tensor(cls_list,
-                                               device=Wb.device)] = sol.T
-                            t0 = tick("woodbury", t0)
+                            t0 = tick("wb_solve", t0)
+                            corr = (UBb * z).sum(1)              # [G, db]
+                            Wb[:, cls_idx] = (rG - corr).to(dtype).T
+                            t0 = tick("wb_corr", t0)
                         else:
                             # large n_c: direct per-class factorization
                             B_dir = base_w * gram_all.to(sdt) \
                                 + self.lam * torch.eye(db, dtype=sdt,
                                                        device=Xb.device)
-                            for c, j in zip(cls_list, js):
-                                s, e = cls_slice[c]
-                                Xc64 = Xb64[s:e]
+                            for c, j in zip(cls_list, js_l):
+                                s_c, e_c = cls_slice[c]
+                                Xc64 = Xb64[s_c:e_c]
                                 G = B_dir + wcs[c] * (Xc64.T @ Xc64)
                                 Lc = torch.linalg.cholesky(G)
                                 Wb[:, c] = torch.cholesky_solve(
