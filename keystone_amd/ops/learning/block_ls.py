@@ -434,8 +434,46 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             # one GEMM per n_c group — ~15 at K=1000 with random labels)
             w_sqrt_rows = (alpha / counts.to(dev_)).sqrt().to(sdt)[cls]
             col = {c: j for j, c in enumerate(own)}
+            # padded single-batch Woodbury: solve EVERY eligible class in
+            # ONE batched update by padding each class's rows to the max
+            # n_c.  Padded rows are zeroed (mask), so S gains identity
+            # rows/cols and the padded z entries vanish — the solve is
+            # exact.  Skipped when padding would inflate flops > 4x
+            # (pathological class skew) — those classes stay in the
+            # per-n_c group loop below.
+            pad = None
+            pad_cls: set = set()
+            if uniform:
+                db0_ = blocks[0].shape[1]
+                wood = [(c, cls_slice[c][1] - cls_slice[c][0])
+                        for c in own
+                        if 0 < (cls_slice[c][1] - cls_slice[c][0])
+                        <= self.woodbury_threshold * db0_]
+                if wood:
+                    ncs = [n_c for _, n_c in wood]
+                    ncmax = max(ncs)
+                    Gw = len(wood)
+                    if Gw * ncmax <= 4 * sum(ncs):
+                        n_rows = max(1, X_full.shape[0])
+                        starts_w = torch.tensor(
+                            [cls_slice[c][0] for c, _ in wood])
+                        ar = torch.arange(ncmax)
+                        ri = (starts_w.unsqueeze(1) + ar).clamp_max(n_rows - 1)
+                        mask = (ar.unsqueeze(0)
+                                < torch.tensor(ncs).unsqueeze(1))
+                        pad = (ri.reshape(-1).to(dev_),
+                               mask.unsqueeze(2).to(device=dev_, dtype=sdt),
+                               torch.tensor([col[c] for c, _ in wood],
+                                            device=dev_),
+                               torch.tensor([c for c, _ in wood],
+                                            device=dev_),
+                               torch.eye(ncmax, dtype=sdt, device=dev_),
+                               Gw, ncmax)
+                        pad_cls = set(c for c, _ in wood)
             by_nc: dict = {}
             for c in own:
+                if c in pad_cls:
+                    continue
                 s_c, e_c = cls_slice[c]
                 by_nc.setdefault(e_c - s_c, []).append(c)
             groups = []
@@ -514,6 +552,21 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     t0 = tick("wb_ubuild", t0)
                     UB_all = U_all @ Binv                        # [n, db]
                     t0 = tick("wb_gemm", t0)
+                    if pad is not None:
+                        (ri_p, mask_p, js_w, cls_idx_w, eye_w,
+                         Gw, ncmax) = pad
+                        Ubp = U_all[ri_p].reshape(Gw, ncmax, db) * mask_p
+                        UBbp = UB_all[ri_p].reshape(Gw, ncmax, db) * mask_p
+                        rGw = RB.T[js_w].contiguous()        # [Gw, db]
+                        S = eye_w + Ubp @ UBbp.transpose(1, 2)
+                        t_p = (Ubp * rGw.unsqueeze(1)).sum(2, keepdim=True)
+                        t0 = tick("wb_smallbmm", t0)
+                        z = t_p / S if ncmax == 1 \
+                            else torch.linalg.solve(S, t_p)
+                        t0 = tick("wb_solve", t0)
+                        corr = (UBbp * z).sum(1)             # [Gw, db]
+                        Wb[:, cls_idx_w] = (rGw - corr).to(dtype).T
+                        t0 = tick("wb_corr", t0)
                     for (n_c, cls_list, js_l, js, cls_idx, row_idx, w_row,
                          eye_nc) in groups:
                         if n_c == 0:
