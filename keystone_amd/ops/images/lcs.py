@@ -69,16 +69,19 @@ class LCSExtractor(Transformer):
         X = imgs.reshape(B * C, H, W)
 
         def pool(z):
-            t = torch.matmul(z.transpose(1, 2), Mh)   # [BC, W, 4*ny]
-            return torch.matmul(t.transpose(1, 2), Mw)  # [BC, 4*ny, 4*nx]
+            # W-direction as ONE flat plain GEMM; H-direction batched;
+            # output stays in the bmm's contiguous [BC, 4*nx, 4*ny] layout
+            # (arr() below absorbs the axis order into its view)
+            t = (z.reshape(B * C * H, W) @ Mw).reshape(B * C, H, -1)
+            return torch.matmul(t.transpose(1, 2), Mh)
 
         m1 = pool(X)
         m2 = pool(X * X)
         std = (m2 - m1 * m1).clamp_min(0).sqrt()
 
-        def arr(m):  # [BC, 4*ny, 4*nx] -> [B, ny, nx, j, i, C] view
-            return m.reshape(B, C, N_SUB, ny, N_SUB, nx) \
-                    .permute(0, 3, 5, 2, 4, 1)
+        def arr(m):  # [BC, 4*nx, 4*ny] -> [B, ny, nx, j, i, C] view
+            return m.reshape(B, C, N_SUB, nx, N_SUB, ny) \
+                    .permute(0, 5, 3, 4, 2, 1)
 
         # one contiguous write builds the final layout
         # [(stat*16 + j*4 + i) * C + c] — identical to the reference's

@@ -172,7 +172,8 @@ class SIFTExtractor(Transformer):
         if nx == 0 or ny == 0:
             return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
                                device=imgs.device, dtype=dt)
-        A1 = omap.reshape(B * N_ORI, H, W) @ Mw        # [B*8, H, 4nx]
+        A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
+            .reshape(B * N_ORI, H, -1)                 # [B*8, H, 4nx]
         A2 = torch.matmul(Mh.T, A1)                    # [B*8, 4ny, 4nx]
         D = A2.reshape(B, N_ORI, N_SPATIAL, ny, N_SPATIAL, nx) \
             .permute(0, 3, 5, 2, 4, 1) \

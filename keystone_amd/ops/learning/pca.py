@@ -41,6 +41,11 @@ class PCATransformer(BatchTransformer):
         if self.mean is not None:
             self.mean = self.mean.to(device=X.device, dtype=X.dtype)
             X = X - self.mean
+        if X.dim() > 2:
+            # flatten leading dims: ONE plain GEMM instead of a strided-
+            # batched one (measured 16.6 ms vs ~5 at [8192, 2900, 128])
+            lead = X.shape[:-1]
+            return (X.reshape(-1, X.shape[-1]) @ self.components)                 .reshape(*lead, -1)
         return X @ self.components
 
     def apply(self, x: torch.Tensor) -> torch.Tensor:
