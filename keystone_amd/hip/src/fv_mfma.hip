@@ -50,9 +50,13 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     constexpr bool FULL = (KT32 % 4 == 0);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
+    // gamma needs NO double buffer: stats(c-1) reads it during the joint
+    // MFMA span, and logits(c) only overwrite it in the write phase after
+    // the intervening barrier.  Halving it puts K=256 at 75 KB LDS ->
+    // TWO co-resident workgroups per CU (2 MFMA waves per SIMD).
     float* Xs0 = (float*)smem;                  // [2][CHUNK][XSTRIDE]
-    float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [2][CHUNK][KT + 1]
-    float* S0s = Gs0 + 2 * CHUNK * (KT + 1);    // [KT] broadcast (+ll)
+    float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [CHUNK][KT + 1]
+    float* S0s = Gs0 + CHUNK * (KT + 1);        // [KT] broadcast (+ll)
 
     const int img = blockIdx.x;
     const float* Xi = STATS ? X : X + (size_t)img * n * D;
@@ -129,9 +133,9 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     for (int c = 0; c < nc; ++c) {
         const int cur = c & 1;
         float* Xs = Xs0 + cur * CHUNK * XSTRIDE;
-        float* Gs = Gs0 + cur * CHUNK * (KT + 1);
+        float* Gs = Gs0;
         float* Xp = Xs0 + (cur ^ 1) * CHUNK * XSTRIDE;
-        float* Gp = Gs0 + (cur ^ 1) * CHUNK * (KT + 1);
+        float* Gp = Gs0;
         const bool more = c + 1 < nc;
         if (more)
             stage_load((int)chunk_start(c + 1),
@@ -221,7 +225,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     {
         const int cur = (nc - 1) & 1;
         float* Xl = Xs0 + cur * CHUNK * XSTRIDE;
-        float* Gl = Gs0 + cur * CHUNK * (KT + 1);
+        float* Gl = Gs0;
 #pragma unroll 1
         for (int kk = 0; kk < CHUNK; kk += 2) {
 #pragma unroll
@@ -328,7 +332,7 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              int B, int n, int D, int K, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K + 1) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 1) + K + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(B), block(256);
     switch (K) {
@@ -346,7 +350,7 @@ KS_EXPORT int ks_gmm_stats_f32(const float* X, const float* P, float* out,
                                hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + 2 * CHUNK * (K + 1) + K + 1) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 1) + K + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(n_blocks), block(256);
     switch (K) {
