@@ -118,6 +118,31 @@ def _sampling_matrix(L: int, bin_size: int, step: int, device, dtype
     return M, npos
 
 
+def _merged_pool_matrix(L: int, bin_size: int, step: int, device, dtype
+                        ) -> Tuple[torch.Tensor, int]:
+    """[L, npos + N_SPATIAL - 1] DISTINCT-cell pooling matrix for the
+    ``step == bin_size`` case: the (j, p) sampling columns of
+    ``_sampling_matrix`` coincide whenever j + p is equal (neighboring
+    descriptors share their spatial pooling cells, the classic dsift
+    structure), so only ``npos + 3`` triangle columns exist.  ~3.7x fewer
+    GEMM flops; descriptor assembly becomes an as_strided gather."""
+    key = ("merged", L, bin_size, step, str(device), dtype)
+    hit = _band_cache.get(key)
+    if hit is not None:
+        return hit
+    span = (N_SPATIAL - 1) * bin_size
+    anchors = list(range(bin_size // 2, L - span - bin_size // 2, step))
+    npos = len(anchors)
+    Q = npos + N_SPATIAL - 1
+    M = torch.zeros(L, Q, device=device, dtype=dtype)
+    idx = torch.arange(L, device=device, dtype=dtype)
+    for q in range(Q):
+        c = anchors[0] + q * step
+        M[:, q] = (1.0 - (idx - c).abs() / bin_size).clamp_min(0)
+    _band_cache[key] = (M, npos)
+    return M, npos
+
+
 class SIFTExtractor(Transformer):
     def __init__(self, step: int = 4, bin_size: int = 4, num_scales: int = 2,
                  scale_step: float = math.sqrt(2.0), clamp: float = 0.2,
@@ -167,17 +192,42 @@ class SIFTExtractor(Transformer):
         # triangular spatial pooling FUSED with grid sampling: two
         # sampling-banded GEMMs (hipBLASLt) instead of dense pooling of
         # the whole map + slicing (~3.5x fewer flops, no gather pass)
-        Mw, nx = _sampling_matrix(W, bin_size, self.step, imgs.device, dt)
-        Mh, ny = _sampling_matrix(H, bin_size, self.step, imgs.device, dt)
-        if nx == 0 or ny == 0:
-            return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
-                               device=imgs.device, dtype=dt)
-        A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
-            .reshape(B * N_ORI, H, -1)                 # [B*8, H, 4nx]
-        A2 = torch.matmul(Mh.T, A1)                    # [B*8, 4ny, 4nx]
-        D = A2.reshape(B, N_ORI, N_SPATIAL, ny, N_SPATIAL, nx) \
-            .permute(0, 3, 5, 2, 4, 1) \
-            .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI).contiguous()
+        if self.step == bin_size:
+            # shared-cell fast path: pool the DISTINCT cells once
+            # ([L, npos+3] matrices), then each descriptor's 4x4 window
+            # is an as_strided view of the pooled map; the reshape below
+            # performs the gather the old path paid for in contiguous()
+            Mw, nx = _merged_pool_matrix(W, bin_size, self.step,
+                                         imgs.device, dt)
+            Mh, ny = _merged_pool_matrix(H, bin_size, self.step,
+                                         imgs.device, dt)
+            if nx == 0 or ny == 0:
+                return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
+                                   device=imgs.device, dtype=dt)
+            Qx, Qy = nx + N_SPATIAL - 1, ny + N_SPATIAL - 1
+            A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
+                .reshape(B * N_ORI, H, Qx)
+            P = torch.matmul(Mh.T, A1) \
+                .reshape(B, N_ORI, Qy, Qx)             # pooled cells
+            s0, s1, s2, s3 = P.stride()
+            D = P.as_strided((B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
+                             (s0, s2, s3, s2, s3, s1)) \
+                .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
+        else:
+            Mw, nx = _sampling_matrix(W, bin_size, self.step,
+                                      imgs.device, dt)
+            Mh, ny = _sampling_matrix(H, bin_size, self.step,
+                                      imgs.device, dt)
+            if nx == 0 or ny == 0:
+                return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
+                                   device=imgs.device, dtype=dt)
+            A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
+                .reshape(B * N_ORI, H, -1)             # [B*8, H, 4nx]
+            A2 = torch.matmul(Mh.T, A1)                # [B*8, 4ny, 4nx]
+            D = A2.reshape(B, N_ORI, N_SPATIAL, ny, N_SPATIAL, nx) \
+                .permute(0, 3, 5, 2, 4, 1) \
+                .reshape(B, ny * nx,
+                         N_SPATIAL * N_SPATIAL * N_ORI).contiguous()
 
         # SIFT normalization: L2 -> clamp 0.2 -> L2 (fused kernel on GPU)
         if D.is_cuda and dt == torch.float32 and D.shape[1] > 0:
