@@ -118,29 +118,33 @@ def _sampling_matrix(L: int, bin_size: int, step: int, device, dtype
     return M, npos
 
 
-def _merged_pool_matrix(L: int, bin_size: int, step: int, device, dtype
-                        ) -> Tuple[torch.Tensor, int]:
-    """[L, npos + N_SPATIAL - 1] DISTINCT-cell pooling matrix for the
-    ``step == bin_size`` case: the (j, p) sampling columns of
-    ``_sampling_matrix`` coincide whenever j + p is equal (neighboring
-    descriptors share their spatial pooling cells, the classic dsift
-    structure), so only ``npos + 3`` triangle columns exist.  ~3.7x fewer
-    GEMM flops; descriptor assembly becomes an as_strided gather."""
-    key = ("merged", L, bin_size, step, str(device), dtype)
+def _merged_pool_matrix(L: int, bin_size: int, step: int, device, dtype):
+    """Distinct-cell pooling matrix on the gcd lattice.  The (j, p)
+    sampling columns of ``_sampling_matrix`` have centers
+    ``bin//2 + p*step + j*bin`` which all lie on the stride-
+    ``g = gcd(step, bin)`` lattice: only Q = (npos-1)*step/g + 3*bin/g + 1
+    distinct columns exist (neighboring descriptors share pooling cells,
+    the classic dsift structure).  Returns (M [L, Q], npos, Q, sy, sj)
+    where column q pools at center ``bin//2 + q*g`` and descriptor (p, j)
+    reads column ``p*sy + j*sj``."""
+    import math as _m
+    key = ("mergedg", L, bin_size, step, str(device), dtype)
     hit = _band_cache.get(key)
     if hit is not None:
         return hit
     span = (N_SPATIAL - 1) * bin_size
     anchors = list(range(bin_size // 2, L - span - bin_size // 2, step))
     npos = len(anchors)
-    Q = npos + N_SPATIAL - 1
-    M = torch.zeros(L, Q, device=device, dtype=dtype)
+    g = _m.gcd(step, bin_size)
+    sy, sj = step // g, bin_size // g
+    Q = (npos - 1) * sy + (N_SPATIAL - 1) * sj + 1 if npos else 0
+    M = torch.zeros(L, max(Q, 1), device=device, dtype=dtype)
     idx = torch.arange(L, device=device, dtype=dtype)
     for q in range(Q):
-        c = anchors[0] + q * step
+        c = bin_size // 2 + q * g
         M[:, q] = (1.0 - (idx - c).abs() / bin_size).clamp_min(0)
-    _band_cache[key] = (M, npos)
-    return M, npos
+    _band_cache[key] = (M, npos, Q, sy, sj)
+    return M, npos, Q, sy, sj
 
 
 class SIFTExtractor(Transformer):
@@ -192,26 +196,27 @@ class SIFTExtractor(Transformer):
         # triangular spatial pooling FUSED with grid sampling: two
         # sampling-banded GEMMs (hipBLASLt) instead of dense pooling of
         # the whole map + slicing (~3.5x fewer flops, no gather pass)
-        if self.step == bin_size:
-            # shared-cell fast path: pool the DISTINCT cells once
-            # ([L, npos+3] matrices), then each descriptor's 4x4 window
-            # is an as_strided view of the pooled map; the reshape below
+        Mwm, nx_m, Qx, sy_x, sj_x = _merged_pool_matrix(
+            W, bin_size, self.step, imgs.device, dt)
+        if nx_m and Qx <= N_SPATIAL * nx_m:
+            # shared-cell fast path: pool the DISTINCT cells once (gcd
+            # lattice), then each descriptor's 4x4 window is an
+            # as_strided view of the pooled map; the reshape below
             # performs the gather the old path paid for in contiguous()
-            Mw, nx = _merged_pool_matrix(W, bin_size, self.step,
-                                         imgs.device, dt)
-            Mh, ny = _merged_pool_matrix(H, bin_size, self.step,
-                                         imgs.device, dt)
+            Mw, nx = Mwm, nx_m
+            Mh, ny, Qy, sy_y, sj_y = _merged_pool_matrix(
+                H, bin_size, self.step, imgs.device, dt)
             if nx == 0 or ny == 0:
                 return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
                                    device=imgs.device, dtype=dt)
-            Qx, Qy = nx + N_SPATIAL - 1, ny + N_SPATIAL - 1
             A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                 .reshape(B * N_ORI, H, Qx)
             P = torch.matmul(Mh.T, A1) \
                 .reshape(B, N_ORI, Qy, Qx)             # pooled cells
             s0, s1, s2, s3 = P.stride()
-            D = P.as_strided((B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
-                             (s0, s2, s3, s2, s3, s1)) \
+            D = P.as_strided(
+                (B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
+                (s0, sy_y * s2, sy_x * s3, sj_y * s2, sj_x * s3, s1)) \
                 .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
         else:
             Mw, nx = _sampling_matrix(W, bin_size, self.step,
