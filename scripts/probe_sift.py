@@ -20,14 +20,14 @@ t("apply_batch total", lambda: e.apply_batch(imgs), 2)
 omap = t("omap", lambda: e._orientation_maps(imgs, 1.2))
 dt = imgs.dtype
 NS, NO = m.N_SPATIAL, m.N_ORI
-Mw, nx = m._merged_pool_matrix(W, 4, 4, imgs.device, dt)
-Mh, ny = m._merged_pool_matrix(H, 4, 4, imgs.device, dt)
-Qx, Qy = nx + 3, ny + 3
+Mw, nx, Qx, syx, sjx = m._merged_pool_matrix(W, 4, 4, imgs.device, dt)
+Mh, ny, Qy, syy, sjy = m._merged_pool_matrix(H, 4, 4, imgs.device, dt)
 A1 = t("A1 flat gemm", lambda: (omap.reshape(B*NO*H, W) @ Mw).reshape(B*NO, H, Qx))
 P = t("P bmm", lambda: torch.matmul(Mh.T, A1).reshape(B, NO, Qy, Qx))
 def gather():
     s0, s1, s2, s3 = P.stride()
-    return P.as_strided((B, ny, nx, NS, NS, NO), (s0, s2, s3, s2, s3, s1)) \
+    return P.as_strided((B, ny, nx, NS, NS, NO),
+                        (s0, syy*s2, syx*s3, sjy*s2, sjx*s3, s1)) \
             .reshape(B, ny*nx, NS*NS*NO)
 D = t("D gather", gather)
 t("norm", lambda: hip_ops.sift_norm_(D.reshape(-1, 128), 0.2))
