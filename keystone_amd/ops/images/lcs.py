@@ -27,24 +27,29 @@ _samp_cache: dict = {}
 
 
 def _lcs_samp_matrix(L: int, s: int, step: int, device, dtype):
-    """[L, N_SUB * ny] matrix: column (j, yi) averages rows
-    ``ys[yi] + j*s .. +s-1`` (the box window of subpatch j at grid row yi).
-    Column order is j-major so the output reshapes to [..., N_SUB, ny]."""
+    """Distinct-cell box-pooling matrix on the gcd lattice (same trick as
+    sift._merged_pool_matrix): subpatch (j, yi) starts at
+    ``yi*step + j*s`` — all multiples of ``g = gcd(step, s)`` — so only
+    Q = (ny-1)*step/g + 3*s/g + 1 distinct box columns exist.  Returns
+    (M [L, Q], ny, Q, sy, sj): subpatch (yi, j) reads column
+    ``yi*sy + j*sj``."""
+    import math as _m
     key = (L, s, step, str(device), dtype)
     hit = _samp_cache.get(key)
     if hit is not None:
         return hit
     Lm = L - s + 1                      # dense-map length (stride-1 pool)
     span = (N_SUB - 1) * s
-    ys = list(range(0, Lm - span, step))
-    ny = len(ys)
-    M = torch.zeros(L, N_SUB * ny, device=device, dtype=dtype)
-    for j in range(N_SUB):
-        for yi, y0 in enumerate(ys):
-            st = y0 + j * s
-            M[st:st + s, j * ny + yi] = 1.0 / s
-    _samp_cache[key] = (M, ny)
-    return M, ny
+    ny = len(range(0, Lm - span, step))
+    g = _m.gcd(step, s)
+    sy, sj = step // g, s // g
+    Q = (ny - 1) * sy + (N_SUB - 1) * sj + 1 if ny else 0
+    M = torch.zeros(L, max(Q, 1), device=device, dtype=dtype)
+    for q in range(Q):
+        st = q * g
+        M[st:st + s, q] = 1.0 / s
+    _samp_cache[key] = (M, ny, Q, sy, sj)
+    return M, ny, Q, sy, sj
 
 
 class LCSExtractor(Transformer):
@@ -61,8 +66,10 @@ class LCSExtractor(Transformer):
             imgs = imgs.unsqueeze(1)
         B, C, H, W = imgs.shape
         s = self.subpatch_size
-        Mh, ny = _lcs_samp_matrix(H, s, self.step, imgs.device, imgs.dtype)
-        Mw, nx = _lcs_samp_matrix(W, s, self.step, imgs.device, imgs.dtype)
+        Mh, ny, Qy, sy_y, sj_y = _lcs_samp_matrix(
+            H, s, self.step, imgs.device, imgs.dtype)
+        Mw, nx, Qx, sy_x, sj_x = _lcs_samp_matrix(
+            W, s, self.step, imgs.device, imgs.dtype)
         if ny == 0 or nx == 0:
             return torch.zeros(B, 0, 2 * C * N_SUB * N_SUB,
                                device=imgs.device, dtype=imgs.dtype)
@@ -70,8 +77,8 @@ class LCSExtractor(Transformer):
 
         def pool(z):
             # W-direction as ONE flat plain GEMM; H-direction batched;
-            # output stays in the bmm's contiguous [BC, 4*nx, 4*ny] layout
-            # (arr() below absorbs the axis order into its view)
+            # output stays in the bmm's contiguous [BC, Qx, Qy] layout
+            # (arr() below absorbs the axis order into its strided view)
             t = (z.reshape(B * C * H, W) @ Mw).reshape(B * C, H, -1)
             return torch.matmul(t.transpose(1, 2), Mh)
 
@@ -79,9 +86,12 @@ class LCSExtractor(Transformer):
         m2 = pool(X * X)
         std = (m2 - m1 * m1).clamp_min(0).sqrt()
 
-        def arr(m):  # [BC, 4*nx, 4*ny] -> [B, ny, nx, j, i, C] view
-            return m.reshape(B, C, N_SUB, nx, N_SUB, ny) \
-                    .permute(0, 5, 3, 4, 2, 1)
+        def arr(m):  # [BC, Qx, Qy] -> [B, ny, nx, j, i, C] lattice view
+            P = m.reshape(B, C, Qx, Qy)
+            p0, p1, p2, p3 = P.stride()
+            return P.as_strided(
+                (B, ny, nx, N_SUB, N_SUB, C),
+                (p0, sy_y * p3, sy_x * p2, sj_y * p3, sj_x * p2, p1))
 
         # one contiguous write builds the final layout
         # [(stat*16 + j*4 + i) * C + c] — identical to the reference's
