@@ -139,13 +139,57 @@ void sift_norm_kernel(float* __restrict__ D, long rows, int dim,
     for (int c = lane; c < dim; c += 64) d[c] *= inv2;
 }
 
+// register-resident variant for dim == 64*NV (the 128-dim SIFT case):
+// the whole row lives in NV registers per lane between the three passes,
+// so HBM traffic is ONE read + ONE write (the generic kernel above walks
+// the row three times; measured 13.3 -> ~6 ms at 29.5M rows).
+template <int NV>
+__global__ __launch_bounds__(256)
+void sift_norm_reg_kernel(float* __restrict__ D, long rows, float clampv,
+                          float eps) {
+    const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= rows) return;
+    const int lane = threadIdx.x & 63;
+    float* d = D + row * (64 * NV);
+    float v[NV];
+    float ss = 0.0f;
+#pragma unroll
+    for (int q = 0; q < NV; ++q) {
+        v[q] = d[lane + 64 * q];
+        ss += v[q] * v[q];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss += __shfl_down(ss, off, 64);
+    ss = __shfl(ss, 0, 64);
+    const float inv = rsqrtf(fmaxf(ss, eps));
+    float ss2 = 0.0f;
+#pragma unroll
+    for (int q = 0; q < NV; ++q) {
+        v[q] = fminf(v[q] * inv, clampv);
+        ss2 += v[q] * v[q];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss2 += __shfl_down(ss2, off, 64);
+    ss2 = __shfl(ss2, 0, 64);
+    const float inv2 = rsqrtf(fmaxf(ss2, eps));
+#pragma unroll
+    for (int q = 0; q < NV; ++q) d[lane + 64 * q] = v[q] * inv2;
+}
+
 KS_EXPORT int ks_sift_norm_f32(float* D, long rows, int dim, float clampv,
                                float eps, hipStream_t stream) {
     if (rows == 0) return 0;
     const int waves_per_block = 4;
     long blocks = (rows + waves_per_block - 1) / waves_per_block;
-    hipLaunchKernelGGL(sift_norm_kernel, dim3((unsigned)blocks),
-                       dim3(64 * waves_per_block), 0, stream, D, rows, dim,
-                       clampv, eps);
+    dim3 g((unsigned)blocks), b(64 * waves_per_block);
+    switch (dim) {
+        case 64:  hipLaunchKernelGGL(sift_norm_reg_kernel<1>, g, b, 0, stream, D, rows, clampv, eps); break;
+        case 128: hipLaunchKernelGGL(sift_norm_reg_kernel<2>, g, b, 0, stream, D, rows, clampv, eps); break;
+        case 192: hipLaunchKernelGGL(sift_norm_reg_kernel<3>, g, b, 0, stream, D, rows, clampv, eps); break;
+        case 256: hipLaunchKernelGGL(sift_norm_reg_kernel<4>, g, b, 0, stream, D, rows, clampv, eps); break;
+        default:
+            hipLaunchKernelGGL(sift_norm_kernel, g, b, 0, stream, D, rows,
+                               dim, clampv, eps);
+    }
     return (int)hipGetLastError();
 }
