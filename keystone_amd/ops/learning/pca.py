@@ -142,11 +142,16 @@ class DistributedPCAEstimator(Estimator):
             # [100k, 128] shard column-by-column in unblocked larfg/ger
             # launches (~18 ms profiled); the Gram GEMM is ~0.1 ms, and
             # in fp64 the condition-squaring is harmless at these sizes.
-            G = X.to(torch.float64).T @ X.to(torch.float64)      # [d, d]
+            X64 = X.to(torch.float64)
+            G = X64.T @ X64                                      # [d, d]
             rt.all_reduce_(G)
-            evals, evecs = torch.linalg.eigh(G.cpu())
+            # eigh stays ON DEVICE: a host LAPACK call mid-pipeline was
+            # measured stalling subsequent GPU stages by ~100+ ms
+            # (OMP/MAGMA thread interference), while rocSOLVER syevd on a
+            # 128x128 fp64 matrix is a few ms
+            evals, evecs = torch.linalg.eigh(G)
             order = torch.argsort(evals, descending=True)[: self.dims]
-            V = _sign_fix(evecs[:, order]).to(X.dtype).to(X.device)
+            V = _sign_fix(evecs[:, order]).to(X.dtype)
             return PCATransformer(
                 V, mean.to(X.dtype) if mean is not None else None)
         R = TSQR.factor(data)
