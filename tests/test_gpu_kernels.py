@@ -358,3 +358,17 @@ def test_gmm_estimator_uses_fused_estep_on_gpu():
         del os.environ["KEYSTONE_GMM_EAGER"]
     assert (m_fused.means - m_eager.means).abs().max().item() < 1e-2
     assert (m_fused.weights - m_eager.weights).abs().max().item() < 1e-3
+
+
+def test_pca_gram_fast_path_matches_tsqr():
+    # GPU small-d fast path (fp64 Gram + eigh) vs the TSQR+SVD route
+    import keystone_amd as ka
+    from keystone_amd.ops.learning.pca import DistributedPCAEstimator
+    torch.manual_seed(24)
+    X = torch.randn(20000, 96)
+    est = DistributedPCAEstimator(32)
+    cpu_rt = Runtime(rank=0, world_size=1, device=torch.device("cpu"))
+    V_cpu = est._fit(ka.TensorDataset(X, cpu_rt)).components
+    V_gpu = est._fit(ka.from_rows(X.cuda())).components.cpu()
+    # directions agree up to sign (fixed) and fp tolerance
+    assert (V_cpu - V_gpu).abs().max().item() < 1e-3

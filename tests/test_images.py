@@ -208,3 +208,61 @@ def test_fisher_vector_batched_matches_single():
     for b in range(4):
         single = fv.apply(X[b])
         assert torch.allclose(batched[b], single, atol=1e-5)
+
+
+def test_lcs_matches_dense_avgpool_reference():
+    # gcd-lattice shared-cell pooling == dense stride-1 avg_pool + sampling
+    import torch.nn.functional as F
+    from keystone_amd.ops.images.lcs import LCSExtractor, N_SUB
+    torch.manual_seed(3)
+    imgs = torch.rand(2, 3, 64, 60)
+    e = LCSExtractor(step=4, subpatch_size=6)
+    D = e.apply_batch(imgs)
+    s, step = 6, 4
+    B, C, H, W = imgs.shape
+    mean = F.avg_pool2d(imgs, s, stride=1)
+    mean2 = F.avg_pool2d(imgs * imgs, s, stride=1)
+    std = (mean2 - mean * mean).clamp_min(0).sqrt()
+    span = (N_SUB - 1) * s
+    ys = torch.arange(0, mean.shape[2] - span, step)
+    xs = torch.arange(0, mean.shape[3] - span, step)
+    parts = []
+    for m in (mean, std):
+        for j in range(N_SUB):
+            for i in range(N_SUB):
+                parts.append(m[:, :, ys + j * s][:, :, :, xs + i * s])
+    Dref = torch.stack(parts, dim=2).permute(0, 3, 4, 2, 1) \
+        .reshape(B, len(ys) * len(xs), -1)
+    assert D.shape == Dref.shape
+    assert (D - Dref).abs().max().item() < 1e-5
+
+
+def test_sift_merged_pooling_matches_sampling_matrix():
+    # the gcd-lattice fast path == the general (j, p)-column formulation
+    import keystone_amd.ops.images.sift as m
+    torch.manual_seed(4)
+    B, H, W = 2, 96, 80
+    imgs = torch.rand(B, H, W)
+    e = m.SIFTExtractor(step=4, bin_size=4, num_scales=1)
+    NS, NO = m.N_SPATIAL, m.N_ORI
+    for bs in (4, 6):  # gcd 4 and gcd 2 cases
+        omap = e._orientation_maps(imgs, 1.2)
+        dt = imgs.dtype
+        Mw, nx = m._sampling_matrix(W, bs, 4, imgs.device, dt)
+        Mh, ny = m._sampling_matrix(H, bs, 4, imgs.device, dt)
+        A1 = (omap.reshape(B * NO * H, W) @ Mw).reshape(B * NO, H, -1)
+        A2 = torch.matmul(Mh.T, A1)
+        Dref = A2.reshape(B, NO, NS, ny, NS, nx).permute(0, 3, 5, 2, 4, 1) \
+            .reshape(B, ny * nx, NS * NS * NO).contiguous()
+        Mwm, nxm, Qx, syx, sjx = m._merged_pool_matrix(W, bs, 4,
+                                                       imgs.device, dt)
+        Mhm, nym, Qy, syy, sjy = m._merged_pool_matrix(H, bs, 4,
+                                                       imgs.device, dt)
+        assert (nxm, nym) == (nx, ny)
+        A1q = (omap.reshape(B * NO * H, W) @ Mwm).reshape(B * NO, H, Qx)
+        P = torch.matmul(Mhm.T, A1q).reshape(B, NO, Qy, Qx)
+        s0, s1, s2, s3 = P.stride()
+        Dfast = P.as_strided((B, ny, nx, NS, NS, NO),
+                             (s0, syy * s2, syx * s3, sjy * s2, sjx * s3, s1)) \
+            .reshape(B, ny * nx, NS * NS * NO)
+        assert (Dref - Dfast).abs().max().item() < 1e-5
