@@ -364,6 +364,7 @@ def test_pca_gram_fast_path_matches_tsqr():
     # GPU small-d fast path (fp64 Gram + eigh) vs the TSQR+SVD route
     import keystone_amd as ka
     from keystone_amd.ops.learning.pca import DistributedPCAEstimator
+    from keystone_amd.parallel.runtime import Runtime
     torch.manual_seed(24)
     X = torch.randn(20000, 96)
     est = DistributedPCAEstimator(32)
