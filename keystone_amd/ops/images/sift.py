@@ -211,15 +211,12 @@ class SIFTExtractor(Transformer):
                                    device=imgs.device, dtype=dt)
             A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                 .reshape(B * N_ORI, H, Qx)
-            # A^T@Mh (3-D left, shared 2-D right) hits Tensile; the
-            # Mh^T@A1 form has a stride-0 LEFT operand and falls back to
-            # rocblas_gemm_batched_general (13.2 vs ~4 ms profiled)
-            P = torch.matmul(A1.transpose(1, 2), Mh) \
-                .reshape(B, N_ORI, Qx, Qy)             # pooled cells
+            P = torch.matmul(Mh.T, A1) \
+                .reshape(B, N_ORI, Qy, Qx)             # pooled cells
             s0, s1, s2, s3 = P.stride()
             D = P.as_strided(
                 (B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
-                (s0, sy_y * s3, sy_x * s2, sj_y * s3, sj_x * s2, s1)) \
+                (s0, sy_y * s2, sy_x * s3, sj_y * s2, sj_x * s3, s1)) \
                 .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
         else:
             Mw, nx = _sampling_matrix(W, bin_size, self.step,
@@ -231,9 +228,9 @@ class SIFTExtractor(Transformer):
                                    device=imgs.device, dtype=dt)
             A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                 .reshape(B * N_ORI, H, -1)             # [B*8, H, 4nx]
-            A2 = torch.matmul(A1.transpose(1, 2), Mh)  # [B*8, 4nx, 4ny]
-            D = A2.reshape(B, N_ORI, N_SPATIAL, nx, N_SPATIAL, ny) \
-                .permute(0, 5, 3, 4, 2, 1) \
+            A2 = torch.matmul(Mh.T, A1)                # [B*8, 4ny, 4nx]
+            D = A2.reshape(B, N_ORI, N_SPATIAL, ny, N_SPATIAL, nx) \
+                .permute(0, 3, 5, 2, 4, 1) \
                 .reshape(B, ny * nx,
                          N_SPATIAL * N_SPATIAL * N_ORI).contiguous()
 
