@@ -54,3 +54,15 @@ for _ in range(3):
 
 torch.cuda.synchronize()
 print("profiled kernel set complete")
+
+# K3 v2: fused MFMA Fisher Vector (the production FV path) + STATS mode
+Xf = torch.randn(256, 2900, 64, device=dev)
+mu = torch.randn(256, 64, device=dev)
+var = torch.rand(256, 64, device=dev) + 0.5
+w = torch.softmax(torch.randn(256, device=dev), 0)
+for _ in range(3):
+    hip.fisher_vector_mfma(Xf, mu, var, w)
+Xs = torch.randn(100_000, 64, device=dev)
+for _ in range(3):
+    hip.gmm_stats_mfma(Xs, mu, var, w)
+torch.cuda.synchronize()
