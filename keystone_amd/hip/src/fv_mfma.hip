@@ -48,6 +48,11 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // guards below become compile-time (a RUNTIME guard around each load
     // makes hipcc branch + wait vmcnt(0) per element — guide §5 trap 4c)
     constexpr bool FULL = (KT32 % 4 == 0);
+    // gamma row stride: (KT+8) % 32 == 8 puts softmax lane (row, sub)
+    // (4 rows x 8 sublanes per 32-lane phase) on bank row*8 + sub -> all
+    // 32 banks distinct.  The previous KT+1 stride gave bank row+sub:
+    // ~3-way conflicts, 2.7e8 counted (profiles/pmc_summary.csv).
+    constexpr int GST = KT + 8;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     // gamma needs NO double buffer: stats(c-1) reads it during the joint
@@ -56,7 +61,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // TWO co-resident workgroups per CU (2 MFMA waves per SIMD).
     float* Xs0 = (float*)smem;                  // [2][CHUNK][XSTRIDE]
     float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [CHUNK][KT + 1]
-    float* S0s = Gs0 + CHUNK * (KT + 1);        // [KT] broadcast (+ll)
+    float* S0s = Gs0 + CHUNK * GST;        // [KT] broadcast (+ll)
 
     const int img = blockIdx.x;
     const float* Xi = STATS ? X : X + (size_t)img * n * D;
@@ -166,7 +171,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     const int ti = wave + 4 * (t / 5);
                     const int tj = t % 5;
                     if (FULL || ti < KT32) {
-                        const float a = Gp[(kk + lk) * (KT + 1) + ti * 32 + l31];
+                        const float a = Gp[(kk + lk) * GST + ti * 32 + l31];
                         const float b = Xp[(kk + lk) * XSTRIDE + tj * 32 + l31];
                         sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
                             a, b, sacc[t], 0, 0, 0);
@@ -184,7 +189,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
                     const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
-                    Gs[row * (KT + 1) + ct * 32 + l31] = lacc[t][r];
+                    Gs[row * GST + ct * 32 + l31] = lacc[t][r];
                 }
             }
         }
@@ -198,14 +203,14 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             const int sub = tid & 7;
             float m = -1e30f;
             for (int k = sub; k < KT; k += 8)
-                m = fmaxf(m, Gs[row * (KT + 1) + k]);
+                m = fmaxf(m, Gs[row * GST + k]);
 #pragma unroll
             for (int off = 4; off; off >>= 1)
                 m = fmaxf(m, __shfl_xor(m, off, 8));
             float s = 0.0f;
             for (int k = sub; k < KT; k += 8) {
-                float e = __expf(Gs[row * (KT + 1) + k] - m);
-                Gs[row * (KT + 1) + k] = e;
+                float e = __expf(Gs[row * GST + k] - m);
+                Gs[row * GST + k] = e;
                 s += e;
             }
 #pragma unroll
@@ -213,7 +218,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 s += __shfl_xor(s, off, 8);
             const float inv = 1.0f / s;
             for (int k = sub; k < KT; k += 8)
-                Gs[row * (KT + 1) + k] *= inv;
+                Gs[row * GST + k] *= inv;
             if (STATS && sub == 0
                     && row < n - (int)chunk_start(c))
                 atomicAdd(&S0s[KT], m + __logf(s));   // logsumexp sum
@@ -233,7 +238,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 const int ti = wave + 4 * (t / 5);
                 const int tj = t % 5;
                 if (FULL || ti < KT32) {
-                    const float a = Gl[(kk + lk) * (KT + 1) + ti * 32 + l31];
+                    const float a = Gl[(kk + lk) * GST + ti * 32 + l31];
                     const float b = Xl[(kk + lk) * XSTRIDE + tj * 32 + l31];
                     sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
                         a, b, sacc[t], 0, 0, 0);
@@ -332,7 +337,7 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              int B, int n, int D, int K, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 1) + K + 1) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(B), block(256);
     switch (K) {
@@ -350,7 +355,7 @@ KS_EXPORT int ks_gmm_stats_f32(const float* X, const float* P, float* out,
                                hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 1) + K + 1) * 4;
+    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(n_blocks), block(256);
     switch (K) {
