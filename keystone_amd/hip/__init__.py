@@ -112,6 +112,9 @@ def _register_extra(lib: ctypes.CDLL) -> None:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
+        lib.ks_sift_omap_pooled_f32.argtypes = [c_p, c_p, c_p] + [c_i] * 7 \
+            + [c_p]
+        lib.ks_sift_omap_pooled_f32.restype = c_i
         lib.ks_sift_norm_f32.argtypes = [c_p, ctypes.c_long, c_i, c_f, c_f,
                                          c_p]
         lib.ks_sift_norm_f32.restype = c_i
@@ -377,6 +380,37 @@ def sift_omap(imgs: torch.Tensor, sigma: float) -> torch.Tensor:
         ctypes.c_void_p(gk.data_ptr()), radius, B, H, W, _stream()),
         "ks_sift_omap_f32")
     return out
+
+
+def sift_omap_pooled(imgs: torch.Tensor, sigma: float, bin_size: int,
+                     g: int, Qx: int) -> torch.Tensor:
+    """Fused blur+gradient+orientation + horizontal triangle pooling on
+    the gcd lattice: [B, H, W] -> [B, 8, H, Qx] (column q pools at center
+    ``bin//2 + q*g``, zero boundary — the merged-matrix A1 operand without
+    the 17 GB omap round-trip or the 36x-waste banded GEMM)."""
+    import math as _m
+    lib = _load()
+    imgs = _prep(imgs, "sift_omap_pooled.imgs")
+    B, H, W = imgs.shape
+    key = (round(sigma, 6), str(imgs.device))
+    gk = _gauss_cache.get(key)
+    if gk is None:
+        radius = min(max(1, int(_m.ceil(3.0 * sigma))), 12)
+        x = torch.arange(-radius, radius + 1, dtype=torch.float32)
+        k = torch.exp(-0.5 * (x / sigma) ** 2)
+        gk = (k / k.sum()).to(imgs.device)
+        _gauss_cache[key] = gk
+    radius = (gk.numel() - 1) // 2
+    out = torch.empty(B, 8, H, Qx, dtype=torch.float32, device=imgs.device)
+    _check(lib.ks_sift_omap_pooled_f32(
+        ctypes.c_void_p(imgs.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_void_p(gk.data_ptr()), radius, bin_size, g, Qx, B, H, W,
+        _stream()), "ks_sift_omap_pooled_f32")
+    return out
+
+
+def sift_omap_pooled_supported(bin_size: int) -> bool:
+    return 2 <= bin_size <= 9
 
 
 def sift_norm_(D: torch.Tensor, clamp: float = 0.2,

@@ -111,6 +111,133 @@ KS_EXPORT int ks_sift_omap_f32(const float* img, float* omap,
     return (int)hipGetLastError();
 }
 
+// Fused orientation-maps + horizontal triangle pooling on the gcd
+// lattice: writes A1 [B, 8, H, Qx] DIRECTLY (the [B, 8, H, W] omap
+// round-trip is 17 GB at bench scale, and the banded pooling GEMM spends
+// 36x the useful flops).  Column q pools at center bin/2 + q*g with
+// weights (1 - |t|/bin), zero boundary — exactly _merged_pool_matrix.
+#define PHMAX 8   // max horizontal pooling halo (bin - 1)
+
+__global__ __launch_bounds__(256)
+void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
+                             float* __restrict__ out,  // [B, 8, H, Qx]
+                             const float* __restrict__ gk,   // [2r+1]
+                             int r, int bin, int g, int Qx,
+                             int H, int W) {
+    const int b = blockIdx.z;
+    const int oy0 = blockIdx.y * OTH;
+    const int ox0 = blockIdx.x * OTW;
+    const int tid = threadIdx.x;
+    const int ph = bin - 1;                  // pooling halo (<= PHMAX)
+
+    const int RH = OTH + 2 + 2 * r;          // raw/hblur rows
+    const int HW2 = OTW + 2 + 2 * ph;        // hblur/vblur cols
+    const int RW2 = HW2 + 2 * r;             // raw cols
+    __shared__ float raw[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * PHMAX + 2 * MAXR];
+    __shared__ float hb[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * PHMAX];
+    __shared__ float vb[OTH + 2][OTW + 2 + 2 * PHMAX];
+    __shared__ float ob8[8][OTH][OTW + 2 * PHMAX];
+
+    const float* im = img + (size_t)b * H * W;
+
+    // ---- load raw tile + halo (replicate boundary)
+    for (int i = tid; i < RH * RW2; i += 256) {
+        int ry = i / RW2, rx = i % RW2;
+        int gy = oy0 - 1 - r + ry;
+        int gx = ox0 - 1 - r - ph + rx;
+        gy = min(max(gy, 0), H - 1);
+        gx = min(max(gx, 0), W - 1);
+        raw[ry][rx] = im[(size_t)gy * W + gx];
+    }
+    __syncthreads();
+
+    // ---- horizontal blur
+    for (int i = tid; i < RH * HW2; i += 256) {
+        int ry = i / HW2, cx = i % HW2;
+        float s = 0.0f;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry][cx + t];
+        hb[ry][cx] = s;
+    }
+    __syncthreads();
+
+    // ---- vertical blur
+    for (int i = tid; i < (OTH + 2) * HW2; i += 256) {
+        int vy = i / HW2, cx = i % HW2;
+        float s = 0.0f;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[vy + t][cx];
+        vb[vy][cx] = s;
+    }
+    __syncthreads();
+
+    // ---- gradient + orientation soft-binning into LDS planes
+    const float inv_step = 8.0f / (2.0f * (float)M_PI);
+    const int OW2 = OTW + 2 * ph;
+    for (int i = tid; i < OTH * OW2; i += 256) {
+        int ty = i / OW2, tx = i % OW2;
+        const int gy = oy0 + ty, gx = ox0 - ph + tx;
+        float vals[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (gy < H && gx >= 0 && gx < W) {
+            float gxv = 0.0f, gyv = 0.0f;
+            if (gx >= 1 && gx + 1 < W)
+                gxv = 0.5f * (vb[ty + 1][tx + 2] - vb[ty + 1][tx]);
+            if (gy >= 1 && gy + 1 < H)
+                gyv = 0.5f * (vb[ty + 2][tx + 1] - vb[ty][tx + 1]);
+            const float mag = sqrtf(gxv * gxv + gyv * gyv);
+            float ang = atan2f(gyv, gxv);
+            if (ang < 0.0f) ang += 2.0f * (float)M_PI;
+            float t = ang * inv_step;
+            int b0 = (int)t;
+            float frac = t - (float)b0;
+            b0 &= 7;
+            const int b1 = (b0 + 1) & 7;
+            vals[b0] = mag * (1.0f - frac);
+            vals[b1] = mag * frac;
+        }
+#pragma unroll
+        for (int o = 0; o < 8; ++o) ob8[o][ty][tx] = vals[o];
+    }
+    __syncthreads();
+
+    // ---- horizontal triangle pooling at the lattice centers in-tile
+    const int binh = bin / 2;
+    // centers c = binh + q*g with c in [ox0, ox0 + OTW)
+    int q_lo = (ox0 - binh + g - 1) / g;
+    if (q_lo < 0) q_lo = 0;
+    int q_hi = (ox0 + OTW - 1 - binh) / g;
+    if (q_hi > Qx - 1) q_hi = Qx - 1;
+    const int nq = q_hi - q_lo + 1;
+    if (nq <= 0) return;
+    const float invb = 1.0f / (float)bin;
+    for (int i = tid; i < 8 * OTH * nq; i += 256) {
+        const int o = i / (OTH * nq);
+        const int rem = i % (OTH * nq);
+        const int ty = rem / nq;
+        const int q = q_lo + rem % nq;
+        const int gy = oy0 + ty;
+        if (gy >= H) continue;
+        const int c = binh + q * g;
+        float s = 0.0f;
+        for (int t = 1 - bin; t <= bin - 1; ++t) {
+            const int x = c + t;
+            if (x < 0 || x >= W) continue;
+            s += (1.0f - fabsf((float)t) * invb) * ob8[o][ty][x - ox0 + ph];
+        }
+        out[(((size_t)b * 8 + o) * H + gy) * Qx + q] = s;
+    }
+}
+
+KS_EXPORT int ks_sift_omap_pooled_f32(const float* img, float* out,
+                                      const float* gauss, int radius,
+                                      int bin, int g, int Qx, int B, int H,
+                                      int W, hipStream_t stream) {
+    if (radius > MAXR || radius < 0 || bin - 1 > PHMAX || bin < 2 || g < 1)
+        return (int)hipErrorInvalidValue;
+    dim3 grid(ks_ceil_div(W, OTW), ks_ceil_div(H, OTH), B);
+    hipLaunchKernelGGL(sift_omap_pooled_kernel, grid, dim3(256), 0, stream,
+                       img, out, gauss, radius, bin, g, Qx, H, W);
+    return (int)hipGetLastError();
+}
+
 // SIFT descriptor normalization: per row (128-dim), L2 -> clamp -> L2.
 // 4 waves per block, one row per wave.
 __global__ __launch_bounds__(256)

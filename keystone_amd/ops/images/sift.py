@@ -191,7 +191,7 @@ class SIFTExtractor(Transformer):
         """[B, H, W] -> [B, n, 128] descriptors at one scale."""
         B, H, W = imgs.shape
         dt = imgs.dtype
-        omap = self._orientation_maps(imgs, sigma)
+        omap = None   # computed lazily — the fused pooled kernel skips it
 
         # triangular spatial pooling FUSED with grid sampling: two
         # sampling-banded GEMMs (hipBLASLt) instead of dense pooling of
@@ -209,8 +209,23 @@ class SIFTExtractor(Transformer):
             if nx == 0 or ny == 0:
                 return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
                                    device=imgs.device, dtype=dt)
-            A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
-                .reshape(B * N_ORI, H, Qx)
+            import os
+            A1 = None
+            g = math.gcd(self.step, bin_size)
+            if (imgs.is_cuda and dt == torch.float32
+                    and os.environ.get("KEYSTONE_SIFT_FUSED", "1") != "0"):
+                from ... import hip as hip_ops
+                if hip_ops.sift_omap_pooled_supported(bin_size):
+                    # fused blur+gradient+orientation+pool kernel: never
+                    # materializes the [B, 8, H, W] omap (17 GB at bench
+                    # scale) and replaces the 36x-waste banded A1 GEMM
+                    A1 = hip_ops.sift_omap_pooled(
+                        imgs.contiguous(), sigma, bin_size, g, Qx) \
+                        .reshape(B * N_ORI, H, Qx)
+            if A1 is None:
+                omap = self._orientation_maps(imgs, sigma)
+                A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
+                    .reshape(B * N_ORI, H, Qx)
             P = torch.matmul(Mh.T, A1) \
                 .reshape(B, N_ORI, Qy, Qx)             # pooled cells
             s0, s1, s2, s3 = P.stride()
@@ -226,6 +241,7 @@ class SIFTExtractor(Transformer):
             if nx == 0 or ny == 0:
                 return torch.zeros(B, 0, N_ORI * N_SPATIAL * N_SPATIAL,
                                    device=imgs.device, dtype=dt)
+            omap = self._orientation_maps(imgs, sigma)
             A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                 .reshape(B * N_ORI, H, -1)             # [B*8, H, 4nx]
             A2 = torch.matmul(Mh.T, A1)                # [B*8, 4ny, 4nx]

@@ -373,3 +373,40 @@ def test_pca_gram_fast_path_matches_tsqr():
     V_gpu = est._fit(ka.from_rows(X.cuda())).components.cpu()
     # directions agree up to sign (fixed) and fp tolerance
     assert (V_cpu - V_gpu).abs().max().item() < 1e-3
+
+
+def test_sift_omap_pooled_matches_gemm_path(hip):
+    # fused blur+grad+orient+pool kernel vs omap -> merged-matrix GEMM
+    import math
+    import keystone_amd.ops.images.sift as m
+    torch.manual_seed(25)
+    for (bs, step, H, W) in [(4, 4, 96, 80), (6, 4, 128, 96), (8, 4, 64, 64)]:
+        imgs = torch.rand(3, H, W, device="cuda")
+        e = m.SIFTExtractor(step=step, bin_size=bs, num_scales=1)
+        sigma = bs / 3.0
+        Mw, nx, Qx, _, _ = m._merged_pool_matrix(W, bs, step, imgs.device,
+                                                 imgs.dtype)
+        if nx == 0:
+            continue
+        omap = hip.sift_omap(imgs.contiguous(), sigma)
+        ref = (omap.reshape(3 * 8 * H, W) @ Mw).reshape(3, 8, H, Qx)
+        g = math.gcd(step, bs)
+        got = hip.sift_omap_pooled(imgs.contiguous(), sigma, bs, g, Qx)
+        err = (got - ref).abs().max().item()
+        assert err < 1e-4, (bs, H, W, err)
+
+
+def test_sift_extractor_fused_matches_unfused_gpu():
+    import os
+    from keystone_amd.ops.images.sift import SIFTExtractor
+    torch.manual_seed(26)
+    imgs = torch.rand(2, 128, 128, device="cuda")
+    e = SIFTExtractor(step=4, bin_size=6, num_scales=2)
+    D_fused = e.apply_batch(imgs)
+    os.environ["KEYSTONE_SIFT_FUSED"] = "0"
+    try:
+        D_gemm = e.apply_batch(imgs)
+    finally:
+        del os.environ["KEYSTONE_SIFT_FUSED"]
+    assert D_fused.shape == D_gemm.shape
+    assert (D_fused - D_gemm).abs().max().item() < 1e-3
