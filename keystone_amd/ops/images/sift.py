@@ -213,12 +213,16 @@ class SIFTExtractor(Transformer):
             A1 = None
             g = math.gcd(self.step, bin_size)
             if (imgs.is_cuda and dt == torch.float32
-                    and os.environ.get("KEYSTONE_SIFT_FUSED", "1") != "0"):
+                    and os.environ.get("KEYSTONE_SIFT_FUSED") == "1"):
                 from ... import hip as hip_ops
                 if hip_ops.sift_omap_pooled_supported(bin_size):
                     # fused blur+gradient+orientation+pool kernel: never
-                    # materializes the [B, 8, H, W] omap (17 GB at bench
-                    # scale) and replaces the 36x-waste banded A1 GEMM
+                    # materializes the [B, 8, H, W] omap and replaces the
+                    # banded A1 GEMM.  OPT-IN: numerics-exact (GPU-tested)
+                    # but measured SLOWER end-to-end than omap-kernel +
+                    # GEMM (78 KB static LDS -> 1 workgroup/CU; extract
+                    # 84 -> 107 ms at bench scale).  Needs dynamic LDS +
+                    # half-precision orientation planes to win — round-2.
                     A1 = hip_ops.sift_omap_pooled(
                         imgs.contiguous(), sigma, bin_size, g, Qx) \
                         .reshape(B * N_ORI, H, Qx)

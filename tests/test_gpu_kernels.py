@@ -402,10 +402,10 @@ def test_sift_extractor_fused_matches_unfused_gpu():
     torch.manual_seed(26)
     imgs = torch.rand(2, 128, 128, device="cuda")
     e = SIFTExtractor(step=4, bin_size=6, num_scales=2)
-    D_fused = e.apply_batch(imgs)
-    os.environ["KEYSTONE_SIFT_FUSED"] = "0"
+    D_gemm = e.apply_batch(imgs)
+    os.environ["KEYSTONE_SIFT_FUSED"] = "1"
     try:
-        D_gemm = e.apply_batch(imgs)
+        D_fused = e.apply_batch(imgs)
     finally:
         del os.environ["KEYSTONE_SIFT_FUSED"]
     assert D_fused.shape == D_gemm.shape
