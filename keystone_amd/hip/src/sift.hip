@@ -133,10 +133,15 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
     const int RH = OTH + 2 + 2 * r;          // raw/hblur rows
     const int HW2 = OTW + 2 + 2 * ph;        // hblur/vblur cols
     const int RW2 = HW2 + 2 * r;             // raw cols
-    __shared__ float raw[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * PHMAX + 2 * MAXR];
-    __shared__ float hb[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * PHMAX];
-    __shared__ float vb[OTH + 2][OTW + 2 + 2 * PHMAX];
-    __shared__ float ob8[8][OTH][OTW + 2 * PHMAX];
+    const int OW2 = OTW + 2 * ph;
+    // dynamic LDS sized to the ACTUAL radius/halo: the static worst-case
+    // (r=12, ph=8) was 78 KB -> 1 workgroup/CU; bench shapes (r<=8) fit
+    // ~68 KB -> 2 co-resident workgroups
+    extern __shared__ float smem2[];
+    float* raw = smem2;                      // [RH][RW2]
+    float* hb = raw + RH * RW2;              // [RH][HW2]
+    float* vb = hb + RH * HW2;               // [OTH+2][HW2]
+    float* ob8 = vb + (OTH + 2) * HW2;       // [8][OTH][OW2]
 
     const float* im = img + (size_t)b * H * W;
 
@@ -147,7 +152,7 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
         int gx = ox0 - 1 - r - ph + rx;
         gy = min(max(gy, 0), H - 1);
         gx = min(max(gx, 0), W - 1);
-        raw[ry][rx] = im[(size_t)gy * W + gx];
+        raw[ry * RW2 + rx] = im[(size_t)gy * W + gx];
     }
     __syncthreads();
 
@@ -155,8 +160,8 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < RH * HW2; i += 256) {
         int ry = i / HW2, cx = i % HW2;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry][cx + t];
-        hb[ry][cx] = s;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry * RW2 + cx + t];
+        hb[ry * HW2 + cx] = s;
     }
     __syncthreads();
 
@@ -164,14 +169,13 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < (OTH + 2) * HW2; i += 256) {
         int vy = i / HW2, cx = i % HW2;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[vy + t][cx];
-        vb[vy][cx] = s;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[(vy + t) * HW2 + cx];
+        vb[vy * HW2 + cx] = s;
     }
     __syncthreads();
 
     // ---- gradient + orientation soft-binning into LDS planes
     const float inv_step = 8.0f / (2.0f * (float)M_PI);
-    const int OW2 = OTW + 2 * ph;
     for (int i = tid; i < OTH * OW2; i += 256) {
         int ty = i / OW2, tx = i % OW2;
         const int gy = oy0 + ty, gx = ox0 - ph + tx;
@@ -179,9 +183,11 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
         if (gy < H && gx >= 0 && gx < W) {
             float gxv = 0.0f, gyv = 0.0f;
             if (gx >= 1 && gx + 1 < W)
-                gxv = 0.5f * (vb[ty + 1][tx + 2] - vb[ty + 1][tx]);
+                gxv = 0.5f * (vb[(ty + 1) * HW2 + tx + 2]
+                              - vb[(ty + 1) * HW2 + tx]);
             if (gy >= 1 && gy + 1 < H)
-                gyv = 0.5f * (vb[ty + 2][tx + 1] - vb[ty][tx + 1]);
+                gyv = 0.5f * (vb[(ty + 2) * HW2 + tx + 1]
+                              - vb[ty * HW2 + tx + 1]);
             const float mag = sqrtf(gxv * gxv + gyv * gyv);
             float ang = atan2f(gyv, gxv);
             if (ang < 0.0f) ang += 2.0f * (float)M_PI;
@@ -194,7 +200,7 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
             vals[b1] = mag * frac;
         }
 #pragma unroll
-        for (int o = 0; o < 8; ++o) ob8[o][ty][tx] = vals[o];
+        for (int o = 0; o < 8; ++o) ob8[(o * OTH + ty) * OW2 + tx] = vals[o];
     }
     __syncthreads();
 
@@ -220,7 +226,8 @@ void sift_omap_pooled_kernel(const float* __restrict__ img,  // [B, H, W]
         for (int t = 1 - bin; t <= bin - 1; ++t) {
             const int x = c + t;
             if (x < 0 || x >= W) continue;
-            s += (1.0f - fabsf((float)t) * invb) * ob8[o][ty][x - ox0 + ph];
+            s += (1.0f - fabsf((float)t) * invb)
+                 * ob8[(o * OTH + ty) * OW2 + (x - ox0 + ph)];
         }
         out[(((size_t)b * 8 + o) * H + gy) * Qx + q] = s;
     }
@@ -232,8 +239,15 @@ KS_EXPORT int ks_sift_omap_pooled_f32(const float* img, float* out,
                                       int W, hipStream_t stream) {
     if (radius > MAXR || radius < 0 || bin - 1 > PHMAX || bin < 2 || g < 1)
         return (int)hipErrorInvalidValue;
+    const int ph = bin - 1;
+    const int RH = OTH + 2 + 2 * radius;
+    const int HW2 = OTW + 2 + 2 * ph;
+    const int RW2 = HW2 + 2 * radius;
+    size_t lds = (size_t)(RH * RW2 + RH * HW2 + (OTH + 2) * HW2
+                          + 8 * OTH * (OTW + 2 * ph)) * 4;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(ks_ceil_div(W, OTW), ks_ceil_div(H, OTH), B);
-    hipLaunchKernelGGL(sift_omap_pooled_kernel, grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL(sift_omap_pooled_kernel, grid, dim3(256), lds, stream,
                        img, out, gauss, radius, bin, g, Qx, H, W);
     return (int)hipGetLastError();
 }
