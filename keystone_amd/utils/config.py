@@ -9,6 +9,19 @@ dataclass config (models/*.py) and the runtime knobs live in ONE
   KEYSTONE_TRACE           JSONL trace output path (utils.tracing)
   KEYSTONE_ALLOW_EAGER     permit eager fallback where a HIP kernel exists
   KEYSTONE_COLL_TIMEOUT_S  RCCL/gloo collective timeout (abort cleanly)
+
+Per-op debug / dispatch switches (read at the call site, not here):
+
+  KEYSTONE_FV_GEMM=1       force the batched-GEMM FV path (skip fv_mfma)
+  KEYSTONE_FV_KERNEL=1     use the v1 fused FV kernel (measured slower)
+  KEYSTONE_GMM_EAGER=1     force the eager GMM E-step (skip STATS kernel)
+  KEYSTONE_GMM_GRAPHS=1    hipGraph-capture the eager E-step (see gmm.py)
+  KEYSTONE_SIFT_FUSED=1    fused SIFT pool kernel (measured slower; opt-in)
+  KEYSTONE_FUSED_GEMM_COS=1  fully-fused MFMA gemm+cos (hybrid is faster)
+  KEYSTONE_GMM_DEBUG=1     per-phase timers in GMM/k-means fits
+  KEYSTONE_SOLVER_DEBUG=1  per-phase timers in the weighted block solver
+  KEYSTONE_BENCH_STAGES=1  per-stage timings from bench.py (stderr JSON)
+  KEYSTONE_NO_GRAPHS=1     disable hipGraph capture everywhere
 """
 
 from __future__ import annotations
