@@ -147,6 +147,15 @@ def _prep(t: torch.Tensor, name: str) -> torch.Tensor:
     return t.contiguous()
 
 
+def _prep_inplace(t: torch.Tensor, name: str) -> torch.Tensor:
+    """For in-place kernels: a silent .contiguous() copy would leave the
+    caller's tensor unmodified — fail loudly on layout regressions."""
+    if not t.is_contiguous():
+        raise RuntimeError(f"{name}: in-place kernel needs a contiguous "
+                           "tensor (got a strided view)")
+    return _prep(t, name)
+
+
 # --------------------------------------------------------------------- ops
 EP_NONE, EP_BIAS, EP_COS, EP_RELU = 0, 1, 2, 3
 
@@ -177,7 +186,7 @@ def gemm_fused(X: torch.Tensor, W: torch.Tensor,
 def bias_cos_(X: torch.Tensor, b: torch.Tensor, scale: float) -> torch.Tensor:
     """In-place ``X = cos(X + b) * scale`` (one HBM pass)."""
     lib = _load()
-    X = _prep(X, "bias_cos.X")
+    X = _prep_inplace(X, "bias_cos.X")
     b = _prep(b, "bias_cos.b")
     rows, cols = X.shape
     _check(lib.ks_bias_cos_f32(
@@ -417,7 +426,7 @@ def sift_norm_(D: torch.Tensor, clamp: float = 0.2,
                eps: float = 1e-20) -> torch.Tensor:
     """In-place SIFT descriptor normalization (L2 -> clamp -> L2) per row."""
     lib = _load()
-    D = _prep(D, "sift_norm.D")
+    D = _prep_inplace(D, "sift_norm.D")
     rows, dim = D.shape
     _check(lib.ks_sift_norm_f32(
         ctypes.c_void_p(D.data_ptr()), ctypes.c_long(rows), dim, clamp, eps,

@@ -138,7 +138,7 @@ class DistributedPCAEstimator(Estimator):
         X = data.data
         if X.is_cuda and X.shape[1] <= 512:
             # small-d fast path: fp64 Gram (ONE GEMM + fused all-reduce)
-            # + replicated CPU eigh.  rocSOLVER's geqrf factors the
+            # + replicated on-device eigh.  rocSOLVER's geqrf factors the
             # [100k, 128] shard column-by-column in unblocked larfg/ger
             # launches (~18 ms profiled); the Gram GEMM is ~0.1 ms, and
             # in fp64 the condition-squaring is harmless at these sizes.

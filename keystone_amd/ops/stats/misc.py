@@ -38,8 +38,9 @@ class RandomSignNode(ElementwiseTransformer):
 
 
 class LinearRectifier(ElementwiseTransformer):
-    """``max(x - alpha, 0.0) + maxVal`` style rectifier; KeystoneML's is
-    ``max(x, alpha)`` (ref: LinearRectifier(maxVal, alpha))."""
+    """``clamp(x - alpha, min=max_val)`` = ``max(max_val, x - alpha)``
+    (ref: LinearRectifier(maxVal, alpha) — same signature/semantics:
+    shift by ``alpha``, floor at ``maxVal``)."""
 
     def __init__(self, max_val: float = 0.0, alpha: float = 0.0):
         self.max_val = max_val

@@ -85,6 +85,19 @@ class Runtime:
                 t.copy_(tc.cpu())
         return t
 
+    def all_reduce_async_(self, t: torch.Tensor):
+        """Start an in-place sum all-reduce and return a handle with
+        ``.wait()`` (or None when single-process).  RCCL runs the reduction
+        on its own stream, so the caller's next GEMM overlaps the wire
+        time — used by the block solvers to hide the per-block Gram
+        reduction behind the next block's Gram GEMM."""
+        if not self.is_distributed:
+            return None
+        if self.backend == "nccl" and not t.is_cuda:
+            self.all_reduce_(t)  # CPU metadata: sync bounce through HBM
+            return None
+        return dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=True)
+
     def all_reduce_many_(self, tensors: List[torch.Tensor], op: str = "sum") -> None:
         """Fused all-reduce of several tensors (one flat buffer => one RCCL
         call; stripes better across the 7 xGMI links than many small calls)."""
@@ -127,7 +140,10 @@ class Runtime:
 
     def broadcast_(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.is_distributed:
-            dist.broadcast(t, src=src)
+            tc = self._coll_dev(t)
+            dist.broadcast(tc, src=src)
+            if tc is not t:
+                t.copy_(tc.cpu())
         return t
 
     def broadcast_obj(self, obj, src: int = 0):
@@ -138,27 +154,45 @@ class Runtime:
         return box[0]
 
     def all_to_all(self, send: List[torch.Tensor]) -> List[torch.Tensor]:
-        """Keyed exchange (Spark shuffle analogue) for GPU-resident rows.
-        RCCL all-to-all on GPU; gloo has no alltoall, so the CPU test path
-        exchanges through all_gather_object."""
+        """Keyed exchange (Spark shuffle analogue) for row tensors, one
+        entry per destination rank; rows may be unequal and zero is fine
+        (classes < world_size produce empty sends).
+
+        RCCL path: ONE ``all_to_all_single`` on a flat [sum_rows, row]
+        buffer with explicit split sizes — a single fused xGMI kernel that
+        handles empty splits, instead of world² point-to-point calls whose
+        zero-size sends are backend-dependent.  gloo has no alltoall, so
+        the CPU test path exchanges through ``all_gather_object``."""
         if not self.is_distributed:
             return send
+        assert len(send) == self.world_size, \
+            f"all_to_all wants {self.world_size} entries, got {len(send)}"
         if self.backend == "gloo":
             gathered: List[Optional[list]] = [None] * self.world_size
             dist.all_gather_object(gathered, [t.cpu() for t in send])
             return [gathered[src][self.rank].to(send[0].device)
                     for src in range(self.world_size)]
-        sdev = self.device if self.backend == "nccl" else torch.device("cpu")
-        sizes = [torch.tensor([s.shape[0]], dtype=torch.int64, device=sdev)
-                 for s in send]
-        recv_sizes = [torch.zeros(1, dtype=torch.int64, device=sdev)
-                      for _ in range(self.world_size)]
-        dist.all_to_all(recv_sizes, sizes)
-        trailing = send[0].shape[1:]
-        recv = [torch.empty((int(n.item()),) + trailing, dtype=send[0].dtype,
-                            device=send[0].device) for n in recv_sizes]
-        dist.all_to_all(recv, [s.contiguous() for s in send])
-        return recv
+        t0 = send[0]
+        trailing = t0.shape[1:]
+        row_elems = 1
+        for s in trailing:
+            row_elems *= int(s)
+        in_splits = [int(s.shape[0]) for s in send]
+        splits_t = torch.tensor(in_splits, dtype=torch.int64, device=self.device)
+        out_splits_t = torch.empty_like(splits_t)
+        dist.all_to_all_single(out_splits_t, splits_t)
+        out_splits = [int(x) for x in out_splits_t.cpu()]
+        flat_in = self._coll_dev(torch.cat(
+            [s.reshape(s.shape[0], row_elems) for s in send]).contiguous())
+        flat_out = torch.empty((sum(out_splits), row_elems),
+                               dtype=flat_in.dtype, device=flat_in.device)
+        dist.all_to_all_single(flat_out, flat_in,
+                               output_split_sizes=out_splits,
+                               input_split_sizes=in_splits)
+        if flat_out.device != t0.device:
+            flat_out = flat_out.to(t0.device)
+        return [p.reshape((p.shape[0],) + trailing)
+                for p in torch.split(flat_out, out_splits, dim=0)]
 
     # ------------------------------------------------------------- helpers
     def shard_bounds(self, n_total: int) -> tuple:

@@ -96,14 +96,18 @@ class GraphExecutor:
                 est = est.optimize(*inputs)
             # fitted-state reuse (ref: SavedStateLoadRule): key by estimator
             # class + hyperparameters when a state cache is enabled
-            from .state_cache import save_fitted, state_cache_dir, try_load
+            from .state_cache import (data_fingerprint, save_fitted,
+                                      state_cache_dir, try_load)
             if state_cache_dir() is not None:
-                cached = try_load(est)
+                # key by (class, hyperparams, fit-data fingerprint): identical
+                # estimators fit on different datasets must not alias
+                tag = data_fingerprint(*inputs[:2])
+                cached = try_load(est, tag)
                 if cached is not None:
                     return cached
             fitted = est.fit(*inputs[:2])
             if state_cache_dir() is not None:
-                save_fitted(est, fitted)
+                save_fitted(est, fitted, tag)
             return fitted
         raise RuntimeError(f"unknown operator {op!r}")
 

@@ -44,6 +44,30 @@ def _estimator_key(est, tag: str = "") -> str:
     return hashlib.sha256(blob.encode()).hexdigest()[:24]
 
 
+def data_fingerprint(*inputs) -> str:
+    """Cheap content fingerprint of the fit data, folded into the cache key
+    so two estimators with identical hyperparameters fit on different
+    datasets never alias (shape + dtype + sum + a strided element sample
+    per tensor input; datum inputs fall back to repr)."""
+    import torch
+    parts = []
+    for x in inputs:
+        t = getattr(x, "data", x)  # TensorDataset -> shard tensor
+        if isinstance(t, torch.Tensor):
+            flat = t.detach().reshape(-1)
+            k = min(64, flat.numel())
+            idx = torch.linspace(0, max(0, flat.numel() - 1), k,
+                                 device=flat.device).long() if k else None
+            sample = flat[idx].to(torch.float64).cpu().numpy().tobytes() \
+                if k else b""
+            parts.append((tuple(t.shape), str(t.dtype),
+                          float(flat.to(torch.float64).sum().cpu()),
+                          hashlib.sha256(sample).hexdigest()[:16]))
+        else:
+            parts.append(repr(t)[:256])
+    return hashlib.sha256(repr(parts).encode()).hexdigest()[:16]
+
+
 def try_load(est, tag: str = ""):
     d = state_cache_dir()
     if d is None:
