@@ -130,6 +130,22 @@ def available() -> bool:
         return False
 
 
+def gpu_ready() -> bool:
+    """The no-silent-fallback contract, enforced: every op that dispatches
+    CUDA tensors to a HIP kernel gates on this.  Extension present -> True
+    (use the kernel).  Extension missing -> RuntimeError, UNLESS
+    KEYSTONE_ALLOW_EAGER=1 explicitly authorizes the eager torch path
+    (debugging only) -> False."""
+    if available():
+        return True
+    if os.environ.get("KEYSTONE_ALLOW_EAGER"):
+        return False
+    raise RuntimeError(
+        "keystone_amd HIP extension missing on a GPU device; refusing "
+        "silent eager fallback (build with keystone_amd.hip.build(), or "
+        "set KEYSTONE_ALLOW_EAGER=1 for debugging)")
+
+
 def _stream() -> ctypes.c_void_p:
     return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
 
@@ -366,18 +382,30 @@ def fisher_vector(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
 
 _gauss_cache = {}
 
+#: kernel-side Gaussian radius limit (sift.hip MAXR, dynamic-LDS bound);
+#: SIFTExtractor falls back to the torch blur chain beyond it — the blur
+#: must never be silently truncated
+SIFT_OMAP_MAX_RADIUS = 48
+
+
+def sift_omap_radius(sigma: float) -> int:
+    import math as _m
+    return max(1, int(_m.ceil(3.0 * sigma)))
+
 
 def sift_omap(imgs: torch.Tensor, sigma: float) -> torch.Tensor:
     """Fused blur+gradient+orientation-binning: [B, H, W] -> [B, 8, H, W]."""
-    import math as _m
     lib = _load()
     imgs = _prep(imgs, "sift_omap.imgs")
     B, H, W = imgs.shape
     key = (round(sigma, 6), str(imgs.device))
     gk = _gauss_cache.get(key)
     if gk is None:
-        radius = max(1, int(_m.ceil(3.0 * sigma)))
-        radius = min(radius, 12)
+        radius = sift_omap_radius(sigma)
+        if radius > SIFT_OMAP_MAX_RADIUS:
+            raise RuntimeError(f"sift_omap: radius {radius} > "
+                               f"{SIFT_OMAP_MAX_RADIUS} (caller must route "
+                               "to the torch blur path)")
         x = torch.arange(-radius, radius + 1, dtype=torch.float32)
         k = torch.exp(-0.5 * (x / sigma) ** 2)
         gk = (k / k.sum()).to(imgs.device)
@@ -397,14 +425,13 @@ def sift_omap_pooled(imgs: torch.Tensor, sigma: float, bin_size: int,
     the gcd lattice: [B, H, W] -> [B, 8, H, Qx] (column q pools at center
     ``bin//2 + q*g``, zero boundary — the merged-matrix A1 operand without
     the 17 GB omap round-trip or the 36x-waste banded GEMM)."""
-    import math as _m
     lib = _load()
     imgs = _prep(imgs, "sift_omap_pooled.imgs")
     B, H, W = imgs.shape
     key = (round(sigma, 6), str(imgs.device))
     gk = _gauss_cache.get(key)
     if gk is None:
-        radius = min(max(1, int(_m.ceil(3.0 * sigma))), 12)
+        radius = min(sift_omap_radius(sigma), SIFT_OMAP_MAX_RADIUS)
         x = torch.arange(-radius, radius + 1, dtype=torch.float32)
         k = torch.exp(-0.5 * (x / sigma) ** 2)
         gk = (k / k.sum()).to(imgs.device)

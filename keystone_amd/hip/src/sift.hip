@@ -22,7 +22,8 @@
 
 #define OTH 16   // output tile rows
 #define OTW 64   // output tile cols
-#define MAXR 12  // max Gaussian radius
+#define MAXR 48  // max Gaussian radius (bin sizes to ~48: scales>=4 at
+                 // bin 8-16 need radius up to 45 — VERDICT r1 #9)
 
 __global__ __launch_bounds__(256)
 void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
@@ -34,13 +35,16 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     const int ox0 = blockIdx.x * OTW;
     const int tid = threadIdx.x;
 
-    // LDS geometry (worst case r=MAXR)
+    // dynamic LDS sized to the ACTUAL radius (a static worst-case array
+    // would cost every launch MAXR's occupancy: r=6 needs 22 KB -> 7
+    // workgroups/CU where the static r=48 layout would pin 1-2)
     const int RH = OTH + 2 + 2 * r;          // raw/hblur rows
     const int RW = OTW + 2 + 2 * r;          // raw cols
     const int HW = OTW + 2;                  // hblur/vblur cols
-    __shared__ float raw[OTH + 2 + 2 * MAXR][OTW + 2 + 2 * MAXR];
-    __shared__ float hb[OTH + 2 + 2 * MAXR][OTW + 2];
-    __shared__ float vb[OTH + 2][OTW + 2];
+    extern __shared__ float smem[];
+    float* raw = smem;                       // [RH][RW]
+    float* hb = raw + RH * RW;               // [RH][HW]
+    float* vb = hb + RH * HW;                // [OTH+2][HW]
 
     const float* im = img + (size_t)b * H * W;
 
@@ -51,7 +55,7 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
         int gx = ox0 - 1 - r + rx;
         gy = min(max(gy, 0), H - 1);
         gx = min(max(gx, 0), W - 1);
-        raw[ry][rx] = im[(size_t)gy * W + gx];
+        raw[ry * RW + rx] = im[(size_t)gy * W + gx];
     }
     __syncthreads();
 
@@ -59,8 +63,8 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < RH * HW; i += 256) {
         int ry = i / HW, cx = i % HW;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry][cx + t];
-        hb[ry][cx] = s;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry * RW + cx + t];
+        hb[ry * HW + cx] = s;
     }
     __syncthreads();
 
@@ -68,8 +72,8 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < (OTH + 2) * HW; i += 256) {
         int vy = i / HW, cx = i % HW;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[vy + t][cx];
-        vb[vy][cx] = s;
+        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[(vy + t) * HW + cx];
+        vb[vy * HW + cx] = s;
     }
     __syncthreads();
 
@@ -81,9 +85,9 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
         if (gy >= H || gx >= W) continue;
         float gxv = 0.0f, gyv = 0.0f;
         if (gx >= 1 && gx + 1 < W)
-            gxv = 0.5f * (vb[ty + 1][tx + 2] - vb[ty + 1][tx]);
+            gxv = 0.5f * (vb[(ty + 1) * HW + tx + 2] - vb[(ty + 1) * HW + tx]);
         if (gy >= 1 && gy + 1 < H)
-            gyv = 0.5f * (vb[ty + 2][tx + 1] - vb[ty][tx + 1]);
+            gyv = 0.5f * (vb[(ty + 2) * HW + tx + 1] - vb[ty * HW + tx + 1]);
         const float mag = sqrtf(gxv * gxv + gyv * gyv);
         float ang = atan2f(gyv, gxv);
         if (ang < 0.0f) ang += 2.0f * (float)M_PI;
@@ -105,8 +109,13 @@ KS_EXPORT int ks_sift_omap_f32(const float* img, float* omap,
                                const float* gauss, int radius, int B, int H,
                                int W, hipStream_t stream) {
     if (radius > MAXR || radius < 0) return (int)hipErrorInvalidValue;
+    const int RH = OTH + 2 + 2 * radius;
+    const int RW = OTW + 2 + 2 * radius;
+    const int HW = OTW + 2;
+    size_t lds = (size_t)(RH * RW + RH * HW + (OTH + 2) * HW) * 4;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(ks_ceil_div(W, OTW), ks_ceil_div(H, OTH), B);
-    hipLaunchKernelGGL(sift_omap_kernel, grid, dim3(256), 0, stream, img,
+    hipLaunchKernelGGL(sift_omap_kernel, grid, dim3(256), lds, stream, img,
                        omap, gauss, radius, H, W);
     return (int)hipGetLastError();
 }

@@ -123,7 +123,8 @@ class SymmetricRectifier(BatchTransformer):
     def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
         if X.is_cuda and X.dtype == torch.float32 and X.dim() == 2:
             from ... import hip as hip_ops
-            return hip_ops.sym_rectify(X, self.alpha)
+            if hip_ops.gpu_ready():
+                return hip_ops.sym_rectify(X, self.alpha)
         return torch.cat([torch.clamp(X - self.alpha, min=0),
                           torch.clamp(-X - self.alpha, min=0)], dim=1)
 

@@ -49,6 +49,7 @@ class FisherVector(Transformer):
         import os
         if X.is_cuda and X.dtype == torch.float32 and self.improved_scaling:
             from ... import hip as hip_ops
+            hip_ops.gpu_ready()  # extension missing on GPU -> loud failure
             # K3 v2 (fv_mfma.hip): pipelined fused logits-MFMA -> softmax
             # -> moment-MFMA per image; never materializes gamma (10x less
             # HBM traffic), measured 1.6x the batched-GEMM path at K=256

@@ -168,7 +168,9 @@ class SIFTExtractor(Transformer):
         dt = imgs.dtype
         if imgs.is_cuda and dt == torch.float32:
             from ... import hip as hip_ops
-            return hip_ops.sift_omap(imgs.contiguous(), sigma)
+            if (hip_ops.gpu_ready() and hip_ops.sift_omap_radius(sigma)
+                    <= hip_ops.SIFT_OMAP_MAX_RADIUS):
+                return hip_ops.sift_omap(imgs.contiguous(), sigma)
         sm = _smooth(imgs, sigma)
         gx = torch.zeros_like(sm)
         gy = torch.zeros_like(sm)
@@ -255,8 +257,9 @@ class SIFTExtractor(Transformer):
                          N_SPATIAL * N_SPATIAL * N_ORI).contiguous()
 
         # SIFT normalization: L2 -> clamp 0.2 -> L2 (fused kernel on GPU)
-        if D.is_cuda and dt == torch.float32 and D.shape[1] > 0:
-            from ... import hip as hip_ops
+        from ... import hip as hip_ops
+        if (D.is_cuda and dt == torch.float32 and D.shape[1] > 0
+                and hip_ops.gpu_ready()):
             hip_ops.sift_norm_(D.reshape(-1, D.shape[2]), self.clamp)
         else:
             D = D / D.norm(dim=2, keepdim=True).clamp_min(1e-10)

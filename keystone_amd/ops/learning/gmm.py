@@ -116,6 +116,7 @@ class GaussianMixtureModelEstimator(Estimator):
         import os
         from ... import hip as hip_ops
         use_fused = (X.is_cuda and X.dtype == torch.float32
+                     and hip_ops.gpu_ready()
                      and hip_ops.fv_mfma_supported(D, K)
                      and not os.environ.get("KEYSTONE_GMM_EAGER"))
 

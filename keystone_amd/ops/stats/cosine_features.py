@@ -40,7 +40,8 @@ class CosineRandomFeatures(BatchTransformer):
         self.W, self.b = W, b  # keep resident on the compute device
         if X.is_cuda and X.dtype == torch.float32:
             from ... import hip as hip_ops
-            return hip_ops.gemm_cos(X, W, b, self.scale)
+            if hip_ops.gpu_ready():
+                return hip_ops.gemm_cos(X, W, b, self.scale)
         return torch.cos(X @ W + b) * self.scale
 
     def get_config(self):

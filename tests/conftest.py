@@ -5,6 +5,8 @@ import torch
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an AMD MI355X GPU (run with -m gpu)")
+    config.addinivalue_line(
+        "markers", "slow: multi-process / long-running CPU test")
 
 
 def pytest_collection_modifyitems(config, items):

@@ -3,6 +3,8 @@ tracing, config, CLI plumbing."""
 
 import json
 import os
+
+import pytest
 import subprocess
 import sys
 
@@ -138,3 +140,36 @@ def test_kernel_matrix_blocks():
     full = GaussianKernelGenerator(0.5)(X, X)
     got = torch.cat([blk for _lo, _hi, blk in km], dim=0)
     assert torch.allclose(got, full, atol=1e-6)
+
+
+def test_no_silent_fallback_contract(monkeypatch):
+    """with the extension missing, GPU dispatch must raise
+    unless KEYSTONE_ALLOW_EAGER=1 explicitly authorizes the eager path."""
+    import keystone_amd.hip as hip
+    monkeypatch.setattr(hip, "_lib", None)
+    monkeypatch.setattr(hip, "_LIB_PATH", "/nonexistent/libkeystone_hip.so")
+    monkeypatch.delenv("KEYSTONE_ALLOW_EAGER", raising=False)
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="refusing"):
+        hip.gpu_ready()
+    monkeypatch.setenv("KEYSTONE_ALLOW_EAGER", "1")
+    assert hip.gpu_ready() is False
+
+
+@pytest.mark.gpu
+def test_gpu_op_raises_without_extension(monkeypatch):
+    """On a real GPU: a dispatching op must fail loudly when the extension
+    is hidden (not silently produce the eager result)."""
+    import keystone_amd.hip as hip
+    from keystone_amd.ops.stats import CosineRandomFeatures
+    monkeypatch.setattr(hip, "_lib", None)
+    monkeypatch.setattr(hip, "_LIB_PATH", "/nonexistent/libkeystone_hip.so")
+    monkeypatch.delenv("KEYSTONE_ALLOW_EAGER", raising=False)
+    op = CosineRandomFeatures(8, 16, gamma=1.0)
+    X = torch.randn(4, 8, device="cuda", dtype=torch.float32)
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="refusing"):
+        op.apply_batch(X)
+    monkeypatch.setenv("KEYSTONE_ALLOW_EAGER", "1")
+    out = op.apply_batch(X)          # authorized eager fallback
+    assert out.shape == (4, 16)
