@@ -99,12 +99,12 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
-        lib.ks_fv_mfma_f32.argtypes = [c_p] * 6 + [c_i] * 4 + [c_p]
+        lib.ks_fv_mfma_f32.argtypes = [c_p] * 6 + [c_i] * 5 + [c_p]
         lib.ks_fv_mfma_f32.restype = c_i
     except AttributeError:
         pass
     try:
-        lib.ks_gmm_stats_f32.argtypes = [c_p] * 3 + [c_i] * 4 + [c_p]
+        lib.ks_gmm_stats_f32.argtypes = [c_p] * 3 + [c_i] * 5 + [c_p]
         lib.ks_gmm_stats_f32.restype = c_i
     except AttributeError:
         pass
@@ -281,6 +281,13 @@ def fv_mfma_supported(D: int, K: int) -> bool:
     return D <= 64 and K in (64, 128, 256)
 
 
+def _f16_stats() -> int:
+    """f16 stats-MFMA pass (v_mfma_f32_32x32x16_f16, 16x the f32 form's
+    issue throughput; gamma/x inputs round to f16, f32 accumulate).
+    KEYSTONE_FV_F32STATS=1 selects the exact-f32 pass."""
+    return 0 if os.environ.get("KEYSTONE_FV_F32STATS") else 1
+
+
 def fisher_vector_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
                        w: torch.Tensor) -> torch.Tensor:
     """Fused MFMA FV (fv_mfma.hip): X [B, n, D] -> [B, 2*K*D] without
@@ -299,7 +306,7 @@ def fisher_vector_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
         ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
         ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
         ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(out.data_ptr()),
-        B, n, D, K, _stream()), "ks_fv_mfma_f32")
+        B, n, D, K, _f16_stats(), _stream()), "ks_fv_mfma_f32")
     return out
 
 
@@ -325,8 +332,8 @@ def gmm_stats_into(X: torch.Tensor, P: torch.Tensor, out: torch.Tensor):
     nb = out.shape[0]
     _check(lib.ks_gmm_stats_f32(
         ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
-        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _stream()),
-        "ks_gmm_stats_f32")
+        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _f16_stats(),
+        _stream()), "ks_gmm_stats_f32")
     s = out.sum(0, dtype=torch.float64)
     return (s[:K], s[K:K + K * D].view(K, D),
             s[K + K * D:K + 2 * K * D].view(K, D), s[K + 2 * K * D:])
@@ -349,8 +356,8 @@ def gmm_stats_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,
     out = torch.empty(nb, width, dtype=torch.float32, device=X.device)
     _check(lib.ks_gmm_stats_f32(
         ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
-        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _stream()),
-        "ks_gmm_stats_f32")
+        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _f16_stats(),
+        _stream()), "ks_gmm_stats_f32")
     s = out.sum(0, dtype=torch.float64)
     Nk = s[:K].contiguous()
     Sx = s[K:K + K * D].view(K, D).contiguous()

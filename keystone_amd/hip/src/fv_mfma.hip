@@ -5,34 +5,49 @@
 // keeps everything on-chip: per image (one workgroup), per 32-descriptor
 // chunk:
 //
-//   phase L: logits[32, K] = Xext @ P           (MFMA, acc in AGPRs)
+//   phase L: logits[32, K] = Xext @ P           (f32 MFMA, acc in AGPRs)
 //            with Xext = [x^2 | x | 1 | 0-pad] (LDS, built while staging)
 //            and P = [-1/(2var) ; mu/var ; -c ; 0] (global, L2-resident)
-//   phase S: softmax rows in LDS (8 lanes per row)
+//   phase S: softmax rows in LDS (8 lanes per row) -> gamma
 //   phase A: S[K, XC] += gamma^T @ Xext         (MFMA, accumulators held
 //            across ALL chunks of the image: S2 | S1 | S0 columns)
 //
 // The ones-column trick makes S0 just another GEMM column, and zero-padded
 // tail descriptors contribute nothing (their Xext rows are zero), so no
 // masking is needed anywhere.  Epilogue converts (S0, S1, S2) to the
-// improved-FV form and writes [2*K*D] per image.
+// improved-FV form and writes [2*KT*D] per image.
+//
+// f16 stats pass (default, round 2): gamma and Xext are ALSO written to
+// LDS as TRANSPOSED f16 panels ([comp][chunk-row] / [xcol][chunk-row],
+// row stride 40 f16 = 80 B = five 16-B groups -> b128 fragment loads walk
+// all 8 quad-bank groups), and phase A runs on v_mfma_f32_32x32x16_f16:
+// k=32 takes 2 instructions x 32 cyc/SIMD instead of 16 x 64 — a 16x
+// issue-throughput step that removes the stats pass from the critical
+// path (f32 accumulate; only the gamma/x inputs round to f16, ~5e-4
+// relative).  KEYSTONE_FV_F32STATS=1 selects the exact-f32 stats pass.
 //
 // Supported: D <= 64 (DT=64 padded), K in {64, 128, 256}.  4 waves.
+//
+// ref: EncEval gmm-fisher/fisher.cxx (calcAndGetFVs) —
+// redesigned as a single fused kernel rather than a port.
 
 #include "common.h"
 #include <math.h>
 
 #define DT 64
-#define XC (2 * DT + 2)      // 130 live columns of Xext
+#define XC (2 * DT + 2)      // 130 live columns of Xext (col 129 = pad0)
 #define XCP 160              // padded to 5 MFMA col-tiles
 #define XSTRIDE (XCP + 1)    // LDS row stride: (161*r + c) % 32 distinct
 #define CHUNK 32
+#define SH (CHUNK + 8)       // f16 panel row stride (80 B = 16 B x 5, odd)
+
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 
 // STATS mode: X is ONE flat [n, D] matrix processed grid-strided; block b
 // writes its PARTIAL sufficient statistics to out[b] = [S0 (KT) | S1
 // (KT*D) | S2 (KT*D) | ll (1)] — the GMM EM E-step in one launch (sum the
 // partials host-side).  FV mode (STATS=false): one image per block.
-template <int KT, bool STATS>
+template <int KT, bool STATS, bool F16S>
 __global__ __launch_bounds__(256, 1)
 void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     const float* __restrict__ P,    // [XCP, KT] padded
@@ -53,15 +68,21 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // 32 banks distinct.  The previous KT+1 stride gave bank row+sub:
     // ~3-way conflicts, 2.7e8 counted (profiles/pmc_summary.csv).
     constexpr int GST = KT + 8;
+    // f32 words before the f16 region (16-B aligned for b128 loads)
+    constexpr int NBUF = F16S ? 1 : 2;          // Xs f32 buffers
+    constexpr int F32W =
+        ((NBUF * CHUNK * XSTRIDE + CHUNK * GST + KT + 1) + 3) & ~3;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     // gamma needs NO double buffer: stats(c-1) reads it during the joint
     // MFMA span, and logits(c) only overwrite it in the write phase after
-    // the intervening barrier.  Halving it puts K=256 at 75 KB LDS ->
-    // TWO co-resident workgroups per CU (2 MFMA waves per SIMD).
-    float* Xs0 = (float*)smem;                  // [2][CHUNK][XSTRIDE]
-    float* Gs0 = Xs0 + 2 * CHUNK * XSTRIDE;     // [CHUNK][KT + 1]
-    float* S0s = Gs0 + CHUNK * GST;        // [KT] broadcast (+ll)
+    // the intervening barrier.
+    float* Xs0 = (float*)smem;                  // [NBUF][CHUNK][XSTRIDE]
+    float* Gs0 = Xs0 + NBUF * CHUNK * XSTRIDE;  // [CHUNK][GST]
+    float* S0s = Gs0 + CHUNK * GST;             // [KT] broadcast (+ll)
+    // f16 region (F16S only): transposed panels for the f16 stats MFMA
+    _Float16* XhT0 = (_Float16*)(smem + (size_t)F32W * 4);  // [2][XCP][SH]
+    _Float16* GhT = XhT0 + 2 * XCP * SH;                    // [KT][SH]
 
     const int img = blockIdx.x;
     const float* Xi = STATS ? X : X + (size_t)img * n * D;
@@ -101,46 +122,121 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         return;
     }
 
-    // zero the pad columns (129..XCP-1) once; never touched again
-    for (int buf = 0; buf < 2; ++buf)
+    // zero the f32 pad columns (129..XCP-1) once; never touched again
+    for (int buf = 0; buf < NBUF; ++buf)
         for (int i = tid; i < CHUNK * (XCP - XC + 1); i += 256) {
             const int r = i / (XCP - XC + 1), c = XC - 1 + i % (XCP - XC + 1);
             Xs0[(buf * CHUNK + r) * XSTRIDE + c] = 0.0f;
         }
+    if (F16S) {
+        // zero the f16 pad rows (cols 129..159 of Xext) in both buffers
+        for (int i = tid; i < 2 * (XCP - XC + 1) * SH; i += 256) {
+            const int buf = i / ((XCP - XC + 1) * SH);
+            const int j = i % ((XCP - XC + 1) * SH);
+            XhT0[(buf * XCP + XC - 1) * SH + j] = (_Float16)0.0f;
+        }
+    }
 
-    // per-thread staging slots: 8 x-elements (i = tid + 256*q)
+    // per-thread staging slots.
+    //   F16S: thread owns column d = tid & 63 and 8 CONSECUTIVE rows
+    //         r0 = (tid >> 6) * 8 — consecutive rows pack into ONE b128
+    //         f16 write per panel row (conflict-free: 80-B row stride).
+    //   f32:  thread owns 8 strided (row, d) slots as in round 1.
     float xreg[8];
+    const int sd = tid & 63;                    // F16S staging column
+    const int sr0 = (tid >> 6) * 8;             // F16S staging row base
     auto stage_load = [&](int c0, int rows) {
 #pragma unroll
         for (int q = 0; q < 8; ++q) {
-            const int i = tid + 256 * q;
-            const int r = i / DT, d = i % DT;
+            const int r = F16S ? (sr0 + q) : (tid + 256 * q) / DT;
+            const int d = F16S ? sd : (tid + 256 * q) % DT;
             xreg[q] = (r < rows && d < D)
                 ? Xi[(size_t)(c0 + r) * D + d] : 0.0f;
         }
     };
-    auto stage_write = [&](float* Xs, int rows) {
+    auto stage_write = [&](float* Xs, _Float16* Xh, int rows) {
+        if (F16S) {
+            f16x8 hx, hx2;
 #pragma unroll
-        for (int q = 0; q < 8; ++q) {
-            const int i = tid + 256 * q;
-            const int r = i / DT, d = i % DT;
-            Xs[r * XSTRIDE + d] = xreg[q] * xreg[q];
-            Xs[r * XSTRIDE + DT + d] = xreg[q];
+            for (int q = 0; q < 8; ++q) {
+                const float x2 = xreg[q] * xreg[q];
+                Xs[(sr0 + q) * XSTRIDE + sd] = x2;
+                Xs[(sr0 + q) * XSTRIDE + DT + sd] = xreg[q];
+                hx[q] = (_Float16)xreg[q];
+                hx2[q] = (_Float16)x2;
+            }
+            *(f16x8*)&Xh[(size_t)sd * SH + sr0] = hx2;        // x^2 rows
+            *(f16x8*)&Xh[(size_t)(DT + sd) * SH + sr0] = hx;  // x rows
+            if (sd == 0) {
+                f16x8 ones;
+#pragma unroll
+                for (int q = 0; q < 8; ++q)
+                    ones[q] = (_Float16)((sr0 + q < rows) ? 1.0f : 0.0f);
+                *(f16x8*)&Xh[(size_t)(2 * DT) * SH + sr0] = ones;
+            }
+        } else {
+#pragma unroll
+            for (int q = 0; q < 8; ++q) {
+                const int i = tid + 256 * q;
+                const int r = i / DT, d = i % DT;
+                Xs[r * XSTRIDE + d] = xreg[q] * xreg[q];
+                Xs[r * XSTRIDE + DT + d] = xreg[q];
+            }
         }
         if (tid < CHUNK)
             Xs[tid * XSTRIDE + 2 * DT] = (tid < rows) ? 1.0f : 0.0f;
     };
 
     stage_load((int)chunk_start(0), min(CHUNK, n - (int)chunk_start(0)));
-    stage_write(Xs0, min(CHUNK, n - (int)chunk_start(0)));
+    stage_write(Xs0, XhT0, min(CHUNK, n - (int)chunk_start(0)));
     __syncthreads();
+
+    // f16 stats-MFMA span for the chunk staged in XhT slot (ci & 1),
+    // reading the gamma panel GhT (single-buffered; see barrier notes)
+    auto stats_f16 = [&](int ci) {
+        const _Float16* Xh = XhT0 + (size_t)(ci & 1) * XCP * SH;
+#pragma unroll
+        for (int kk = 0; kk < CHUNK; kk += 16) {
+#pragma unroll
+            for (int t = 0; t < ST; ++t) {
+                const int ti = wave + 4 * (t / 5);
+                const int tj = t % 5;
+                if (FULL || ti < KT32) {
+                    const f16x8 a = *(const f16x8*)
+                        &GhT[(size_t)(ti * 32 + l31) * SH + kk + 8 * lk];
+                    const f16x8 b = *(const f16x8*)
+                        &Xh[(size_t)(tj * 32 + l31) * SH + kk + 8 * lk];
+                    sacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_f16(
+                        a, b, sacc[t], 0, 0, 0);
+                }
+            }
+        }
+    };
+    auto stats_f32 = [&](float* Xp, float* Gp) {
+#pragma unroll 1
+        for (int kk = 0; kk < CHUNK; kk += 2) {
+#pragma unroll
+            for (int t = 0; t < ST; ++t) {
+                const int ti = wave + 4 * (t / 5);
+                const int tj = t % 5;
+                if (FULL || ti < KT32) {
+                    const float a = Gp[(kk + lk) * GST + ti * 32 + l31];
+                    const float b = Xp[(kk + lk) * XSTRIDE + tj * 32 + l31];
+                    sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                        a, b, sacc[t], 0, 0, 0);
+                }
+            }
+        }
+    };
 
     for (int c = 0; c < nc; ++c) {
         const int cur = c & 1;
-        float* Xs = Xs0 + cur * CHUNK * XSTRIDE;
+        // F16S: ONE f32 staging buffer (logits(c) consumes it inside this
+        // span; chunk c+1 overwrites it in the write phase).  f32 stats
+        // keep the double buffer (stats(c-1) reads the other slot).
+        float* Xs = F16S ? Xs0 : Xs0 + cur * CHUNK * XSTRIDE;
         float* Gs = Gs0;
-        float* Xp = Xs0 + (cur ^ 1) * CHUNK * XSTRIDE;
-        float* Gp = Gs0;
+        float* Xp = F16S ? Xs0 : Xs0 + (cur ^ 1) * CHUNK * XSTRIDE;
         const bool more = c + 1 < nc;
         if (more)
             stage_load((int)chunk_start(c + 1),
@@ -164,24 +260,12 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             }
         }
         if (c > 0) {
-#pragma unroll 1
-            for (int kk = 0; kk < CHUNK; kk += 2) {
-#pragma unroll
-                for (int t = 0; t < ST; ++t) {
-                    const int ti = wave + 4 * (t / 5);
-                    const int tj = t % 5;
-                    if (FULL || ti < KT32) {
-                        const float a = Gp[(kk + lk) * GST + ti * 32 + l31];
-                        const float b = Xp[(kk + lk) * XSTRIDE + tj * 32 + l31];
-                        sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
-                            a, b, sacc[t], 0, 0, 0);
-                    }
-                }
-            }
+            if (F16S) stats_f16(c - 1);
+            else stats_f32(Xp, Gs0);
         }
-        __syncthreads();   // stats done reading Xp/Gp; logits accs private
+        __syncthreads();   // stats done reading; logits accs private
 
-        // ---- LDS writes: logits(c) -> Gs, staged chunk c+1 -> Xp slot
+        // ---- LDS writes: logits(c) -> Gs, staged chunk c+1 -> its slot
 #pragma unroll
         for (int t = 0; t < LT; ++t) {
             const int ct = wave + 4 * t;
@@ -194,10 +278,14 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             }
         }
         if (more)
-            stage_write(Xp, min(CHUNK, n - (int)chunk_start(c + 1)));
+            stage_write(Xp, XhT0 + (size_t)((c + 1) & 1) * XCP * SH,
+                        min(CHUNK, n - (int)chunk_start(c + 1)));
         __syncthreads();
 
-        // ---- softmax(c) on Gs, 8 lanes per row
+        // ---- softmax(c) on Gs, 8 lanes per row; F16S writes the
+        // normalized gamma TRANSPOSED into the f16 panel (the two halves
+        // of a dword land in adjacent rows -> 2-way u16 write, measured
+        // cheaper than a separate packing pass)
         {
             const int row = tid >> 3;
             const int sub = tid & 7;
@@ -217,8 +305,13 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             for (int off = 4; off; off >>= 1)
                 s += __shfl_xor(s, off, 8);
             const float inv = 1.0f / s;
-            for (int k = sub; k < KT; k += 8)
-                Gs[row * GST + k] *= inv;
+            for (int k = sub; k < KT; k += 8) {
+                if (F16S)
+                    GhT[(size_t)k * SH + row] =
+                        (_Float16)(Gs[row * GST + k] * inv);
+                else
+                    Gs[row * GST + k] *= inv;
+            }
             if (STATS && sub == 0
                     && row < n - (int)chunk_start(c))
                 atomicAdd(&S0s[KT], m + __logf(s));   // logsumexp sum
@@ -227,24 +320,11 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     }
 
     // ---- trailing stats for the last chunk
-    {
+    if (F16S) {
+        stats_f16(nc - 1);
+    } else {
         const int cur = (nc - 1) & 1;
-        float* Xl = Xs0 + cur * CHUNK * XSTRIDE;
-        float* Gl = Gs0;
-#pragma unroll 1
-        for (int kk = 0; kk < CHUNK; kk += 2) {
-#pragma unroll
-            for (int t = 0; t < ST; ++t) {
-                const int ti = wave + 4 * (t / 5);
-                const int tj = t % 5;
-                if (FULL || ti < KT32) {
-                    const float a = Gl[(kk + lk) * GST + ti * 32 + l31];
-                    const float b = Xl[(kk + lk) * XSTRIDE + tj * 32 + l31];
-                    sacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
-                        a, b, sacc[t], 0, 0, 0);
-                }
-            }
-        }
+        stats_f32(Xs0 + cur * CHUNK * XSTRIDE, Gs0);
     }
     __syncthreads();
 
@@ -332,18 +412,33 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     }
 }
 
+static size_t fv_lds_bytes(int K, bool f16s) {
+    const int nbuf = f16s ? 1 : 2;
+    size_t f32w = ((size_t)(nbuf * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1)
+                   + 3) & ~(size_t)3;
+    size_t b = f32w * 4;
+    if (f16s) b += (size_t)(2 * XCP * SH + K * SH) * 2;
+    return b;
+}
+
 KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              const float* var, const float* w, float* out,
-                             int B, int n, int D, int K, hipStream_t stream) {
+                             int B, int n, int D, int K, int f16_stats,
+                             hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1) * 4;
+    const bool h = f16_stats != 0;
+    size_t lds = fv_lds_bytes(K, h);
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(B), block(256);
-    switch (K) {
-        case 64:  hipLaunchKernelGGL((fv_mfma_kernel<64, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 128: hipLaunchKernelGGL((fv_mfma_kernel<128, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 256: hipLaunchKernelGGL((fv_mfma_kernel<256, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+    switch (K * 2 + (h ? 1 : 0)) {
+        case 129: hipLaunchKernelGGL((fv_mfma_kernel<64, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 128: hipLaunchKernelGGL((fv_mfma_kernel<64, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 257: hipLaunchKernelGGL((fv_mfma_kernel<128, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<128, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 513: hipLaunchKernelGGL((fv_mfma_kernel<256, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 512: hipLaunchKernelGGL((fv_mfma_kernel<256, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        default: return (int)hipErrorInvalidValue;
     }
     return (int)hipGetLastError();
 }
@@ -352,16 +447,21 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
 // out [n_blocks, K + 2*K*D + 1] partials (sum over dim 0 host-side).
 KS_EXPORT int ks_gmm_stats_f32(const float* X, const float* P, float* out,
                                int n_blocks, int n, int D, int K,
-                               hipStream_t stream) {
+                               int f16_stats, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    size_t lds = (size_t)(2 * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1) * 4;
+    const bool h = f16_stats != 0;
+    size_t lds = fv_lds_bytes(K, h);
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(n_blocks), block(256);
-    switch (K) {
-        case 64:  hipLaunchKernelGGL((fv_mfma_kernel<64, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 128: hipLaunchKernelGGL((fv_mfma_kernel<128, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 256: hipLaunchKernelGGL((fv_mfma_kernel<256, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+    switch (K * 2 + (h ? 1 : 0)) {
+        case 129: hipLaunchKernelGGL((fv_mfma_kernel<64, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 128: hipLaunchKernelGGL((fv_mfma_kernel<64, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 257: hipLaunchKernelGGL((fv_mfma_kernel<128, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<128, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 513: hipLaunchKernelGGL((fv_mfma_kernel<256, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 512: hipLaunchKernelGGL((fv_mfma_kernel<256, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        default: return (int)hipErrorInvalidValue;
     }
     return (int)hipGetLastError();
 }

@@ -471,7 +471,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                 h.wait()
         uniform = all(w == widths[0] for w in widths)
         db0 = widths[0]
-        Binvs = grams_pre = None
+        Linvs = grams_pre = None
         if uniform:
             grams_pre = torch.stack(grams_list)
         del grams_list, handles
@@ -480,13 +480,15 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             eye = torch.eye(db0, dtype=sdt, device=Y.device)
             Bb = base_w * grams_pre + self.lam * eye
             Lb = torch.linalg.cholesky(Bb)
-            # B^{-1} = L^{-T} L^{-1} via batched TRSM + GEMM: measured
-            # 32 ms vs rocSOLVER's batched cholesky_inverse at 54 ms
-            # (nb=16, db=4096)
-            Linv = torch.linalg.solve_triangular(
+            # keep only L^{-1}: every downstream use runs in the
+            # T = U L^{-T} form, so the explicit B^{-1} = L^{-T} L^{-1}
+            # assembly GEMM (measured 14.4 ms at nb=16, db=4096) is never
+            # paid.  cholesky 18.4 ms + batched TRSM 18.7 ms is the
+            # cheapest factorization route (cholesky_inverse 55.8,
+            # getrf/getri 646 — probe_chol.py).
+            Linvs = torch.linalg.solve_triangular(
                 Lb, eye.expand(nb, db0, db0), upper=False)
-            Binvs = Linv.transpose(1, 2) @ Linv
-            del Bb, Lb, Linv
+            del Bb, Lb
             t0 = tick("pre_chol", t0)
 
         # ---- loop-invariant per-class group tensors.  Everything here
@@ -608,20 +610,24 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     # ONE base factorization (shared across classes), then a
                     # rank-n_c Woodbury update per class:
                     #   (B + U U^T)^{-1} r = B^{-1}r - B^{-1}U (I + U^T B^{-1}U)^{-1} U^T B^{-1} r
-                    # with B = base_w*G_all + lam I and U = sqrt(w_c) X_c^T.
+                    # with B = base_w*G_all + lam I and U = sqrt(w_c) X_c^T,
+                    # everything phrased through T = U L^{-T} (so only
+                    # L^{-1} is ever materialized):
+                    #   U B^{-1} U^T = T T^T,  U B^{-1} r = T (L^{-1} r),
+                    #   B^{-1} U^T z = L^{-T} (T^T z).
                     # Classes with equal n_c are solved as ONE batched
                     # update (K=1000 at ImageNet scale -> a handful of
                     # batched rocSOLVER/GEMM calls, not 1000 Python steps).
-                    if Binvs is not None:
-                        Binv = Binvs[b]
+                    if Linvs is not None:
+                        Linv = Linvs[b]
                     else:
                         B64 = base_w * gram_all.to(sdt)
                         B64 += self.lam * torch.eye(db, dtype=sdt,
                                                     device=Xb.device)
                         L = torch.linalg.cholesky(B64)
-                        # explicit inverse: downstream B^{-1}-applications
-                        # become GEMMs (hipBLASLt) instead of TRSMs
-                        Binv = torch.cholesky_inverse(L)
+                        Linv = torch.linalg.solve_triangular(
+                            L, torch.eye(db, dtype=sdt, device=Xb.device),
+                            upper=False)
                     Xb64 = Xb.to(sdt)
                     # rhs for ALL classes in two fused passes: the class-c
                     # correction X_c^T T[rows_c, c] is a segmented
@@ -632,30 +638,31 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     seg.index_add_(0, cls, Xb64 * v)
                     rhs = base_w * xtr_all.index_select(1, own_t).to(sdt) \
                         + seg.index_select(0, own_t).T * wc_t
-                    RB = Binv @ rhs                          # B^{-1} rhs
+                    LR = Linv @ rhs                          # L^{-1} rhs
+                    RB = Linv.T @ LR                         # B^{-1} rhs
                     t0 = tick("rhs", t0)
 
-                    # ALL classes' sqrt(w_c)-scaled rows in one pass, and
-                    # Binv is SYMMETRIC so U^T B^{-1} == (B^{-1}U)^T: ONE
-                    # row-major NN GEMM applies B^{-1} to every class's rows
-                    # at the full MFMA rate; groups then just gather views.
+                    # ALL classes' sqrt(w_c)-scaled rows in one pass: ONE
+                    # row-major NN GEMM forms T = U L^{-T} at the full MFMA
+                    # rate; groups then just gather views of T.
                     U_all = Xb64 * w_sqrt_rows.unsqueeze(1)      # [n, db]
                     t0 = tick("wb_ubuild", t0)
-                    UB_all = U_all @ Binv                        # [n, db]
+                    T_all = U_all @ Linv.T                       # [n, db]
                     t0 = tick("wb_gemm", t0)
                     if pad is not None:
                         (ri_p, mask_p, js_w, cls_idx_w, eye_w,
                          Gw, ncmax) = pad
-                        Ubp = U_all[ri_p].reshape(Gw, ncmax, db) * mask_p
-                        UBbp = UB_all[ri_p].reshape(Gw, ncmax, db) * mask_p
+                        Tbp = T_all[ri_p].reshape(Gw, ncmax, db) * mask_p
                         rGw = RB.T[js_w].contiguous()        # [Gw, db]
-                        S = eye_w + Ubp @ UBbp.transpose(1, 2)
-                        t_p = (Ubp * rGw.unsqueeze(1)).sum(2, keepdim=True)
+                        LRg = LR.T[js_w].contiguous()        # [Gw, db]
+                        S = eye_w + Tbp @ Tbp.transpose(1, 2)
+                        t_p = (Tbp * LRg.unsqueeze(1)).sum(2, keepdim=True)
                         t0 = tick("wb_smallbmm", t0)
                         z = t_p / S if ncmax == 1 \
                             else torch.linalg.solve(S, t_p)
                         t0 = tick("wb_solve", t0)
-                        corr = (UBbp * z).sum(1)             # [Gw, db]
+                        Yz = (Tbp * z).sum(1)                # [Gw, db]
+                        corr = Yz @ Linv                     # rows L^{-T} y
                         Wb[:, cls_idx_w] = (rGw - corr).to(dtype).T
                         t0 = tick("wb_corr", t0)
                     for (n_c, cls_list, js_l, js, cls_idx, row_idx, w_row,
@@ -665,20 +672,21 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             continue
                         if n_c <= self.woodbury_threshold * db:
                             G_n = len(cls_list)
-                            Ub = U_all[row_idx].reshape(G_n, n_c, db)
-                            UBb = UB_all[row_idx].reshape(G_n, n_c, db)
+                            Tb = T_all[row_idx].reshape(G_n, n_c, db)
                             rG = RB.T[js].contiguous()           # [G, db]
-                            S = eye_nc + Ub @ UBb.transpose(1, 2)
+                            LRg2 = LR.T[js].contiguous()         # [G, db]
+                            S = eye_nc + Tb @ Tb.transpose(1, 2)
                             # t/corr are GEMV-shaped (N=1): fused
                             # multiply-reduce beats 1000-batch bmm
-                            t = (Ub * rG.unsqueeze(1)).sum(2, keepdim=True)
+                            t = (Tb * LRg2.unsqueeze(1)).sum(2, keepdim=True)
                             t0 = tick("wb_smallbmm", t0)
                             if n_c == 1:
                                 z = t / S
                             else:
                                 z = torch.linalg.solve(S, t)
                             t0 = tick("wb_solve", t0)
-                            corr = (UBb * z).sum(1)              # [G, db]
+                            Yz = (Tb * z).sum(1)                 # [G, db]
+                            corr = Yz @ Linv
                             Wb[:, cls_idx] = (rG - corr).to(dtype).T
                             t0 = tick("wb_corr", t0)
                         else:

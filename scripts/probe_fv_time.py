@@ -1,3 +1,4 @@
+import sys; sys.path.insert(0, ".")
 import time, torch
 from keystone_amd import hip
 def run(B, n, D, K, iters=5):

@@ -295,7 +295,7 @@ def test_solvers_gpu_match_cpu_fp32():
     assert (Wc - Wg).abs().max().item() < 1e-2
 
 
-def test_fv_mfma_kernel_matches_gemm_path(hip):
+def test_fv_mfma_kernel_matches_gemm_path(hip, monkeypatch):
     from keystone_amd.ops.images.fisher_vector import FisherVector
     from keystone_amd.ops.learning.gmm import GaussianMixtureModel
     torch.manual_seed(21)
@@ -304,12 +304,23 @@ def test_fv_mfma_kernel_matches_gemm_path(hip):
         mu = torch.randn(K, D, device="cuda")
         var = torch.rand(K, D, device="cuda") + 0.5
         w = torch.softmax(torch.randn(K, device="cuda"), 0)
-        out = hip.fisher_vector_mfma(X, mu, var, w)
         fv = FisherVector(GaussianMixtureModel(mu.cpu(), var.cpu(), w.cpu()))
         ref = fv.apply_batch(X.cpu())
-        err = (out.cpu() - ref).abs().max().item()
+        # default f16 stats pass: gamma/x inputs round to f16 (f32
+        # accumulate) — measured max abs ~1.2e-3, mean ~1e-5 vs the fp32
+        # oracle on N(0,1) descriptors (outputs are O(1))
+        monkeypatch.delenv("KEYSTONE_FV_F32STATS", raising=False)
+        out = hip.fisher_vector_mfma(X, mu, var, w)
         assert out.shape == ref.shape
-        assert err < 5e-4, (B, n, D, K, err)
+        d = (out.cpu() - ref).abs()
+        assert d.max().item() < 2.5e-3, (B, n, D, K, d.max().item())
+        assert d.mean().item() < 5e-5, (B, n, D, K, d.mean().item())
+        # exact-f32 stats pass keeps the strict bound
+        monkeypatch.setenv("KEYSTONE_FV_F32STATS", "1")
+        out32 = hip.fisher_vector_mfma(X, mu, var, w)
+        err32 = (out32.cpu() - ref).abs().max().item()
+        assert err32 < 5e-4, (B, n, D, K, err32)
+        monkeypatch.delenv("KEYSTONE_FV_F32STATS", raising=False)
 
 
 def test_gmm_stats_kernel_matches_eager(hip):
