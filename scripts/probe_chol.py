@@ -1,3 +1,4 @@
+import sys; sys.path.insert(0, ".")
 """Micro-probe: fastest way to get B^{-1} (or an applicable factor) for
 nb=16 SPD matrices of size 4096 (the weighted solver pre-pass, 51.5 ms
 measured).  Candidates:

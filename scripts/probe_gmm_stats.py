@@ -1,3 +1,4 @@
+import sys; sys.path.insert(0, ".")
 """A/B the fused STATS E-step vs eager at bench scale (GPU probe)."""
 import os, time, torch
 from keystone_amd import hip
