@@ -232,20 +232,19 @@ class SIFTExtractor(Transformer):
                 omap = self._orientation_maps(imgs, sigma)
                 A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                     .reshape(B * N_ORI, H, Qx)
-            # H-pooling as ONE plain GEMM: the broadcast bmm
-            # Mh.T @ [B*8, H, Qx] fell to rocBLAS's batched-general
-            # fallback (13.2 ms at ~4.5 TF, profiled); permuting A1 to
-            # [H, B*8*Qx] (2.8 GB copy, ~0.7 ms) puts the contraction on
-            # hipBLASLt's plain path.  P layout [Qy, B*8, Qx] feeds the
-            # descriptor gather below via as_strided directly.
-            Pf = torch.matmul(
-                Mh.T, A1.permute(1, 0, 2).reshape(H, B * N_ORI * Qx))
-            P = Pf.view(Qy, B * N_ORI, Qx)
+            # H-pooling as ONE tall-skinny plain GEMM.  Both the broadcast
+            # bmm Mh.T @ [B*8, H, Qx] AND the [Qy, H] @ [H, B*8*Qx] flat
+            # form land on rocBLAS's batched-general / small-M fallback
+            # (13.1 ms at 4.5 TF, profiled); transposing A1 (2.8 GB copy,
+            # sub-ms) gives [B*8*Qx, H] @ [H, Qy] — the same shape class
+            # as the A1 GEMM, which hipBLASLt runs at ~100 TF.
+            A1t = A1.transpose(1, 2).reshape(B * N_ORI * Qx, H)
+            P = torch.matmul(A1t, Mh).view(B * N_ORI, Qx, Qy)
             s0, s1, s2 = P.stride()
             D = P.as_strided(
                 (B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
-                (N_ORI * s1, sy_y * s0, sy_x * s2,
-                 sj_y * s0, sj_x * s2, s1)) \
+                (N_ORI * s0, sy_y * s2, sy_x * s1,
+                 sj_y * s2, sj_x * s1, s0)) \
                 .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
         else:
             Mw, nx = _sampling_matrix(W, bin_size, self.step,
