@@ -76,13 +76,11 @@ class LCSExtractor(Transformer):
         X = imgs.reshape(B * C, H, W)
 
         def pool(z):
-            # both directions as flat plain GEMMs (the strided-batched
-            # t^T @ Mh form falls to rocBLAS's slow batched-general
-            # kernel for these skinny shapes — same fix as SIFT's
-            # H-pooling); the transpose copy is sub-ms
+            # W-direction as ONE flat plain GEMM; H-direction batched;
+            # output stays in the bmm's contiguous [BC, Qx, Qy] layout
+            # (arr() below absorbs the axis order into its strided view)
             t = (z.reshape(B * C * H, W) @ Mw).reshape(B * C, H, -1)
-            t2 = t.transpose(1, 2).reshape(-1, H)
-            return (t2 @ Mh).view(B * C, -1, Mh.shape[1])
+            return torch.matmul(t.transpose(1, 2), Mh)
 
         m1 = pool(X)
         m2 = pool(X * X)

@@ -191,6 +191,20 @@ class SIFTExtractor(Transformer):
     def _one_scale(self, imgs: torch.Tensor, bin_size: int, sigma: float
                    ) -> torch.Tensor:
         """[B, H, W] -> [B, n, 128] descriptors at one scale."""
+        import os
+        import time
+        dbg = os.environ.get("KEYSTONE_SIFT_DEBUG") and imgs.is_cuda
+        _ph = {}
+
+        def tick(name, t0):
+            if dbg:
+                torch.cuda.synchronize()
+                now = time.perf_counter()
+                _ph[name] = _ph.get(name, 0.0) + (now - t0)
+                return now
+            return t0
+
+        t0 = time.perf_counter()
         B, H, W = imgs.shape
         dt = imgs.dtype
         omap = None   # computed lazily — the fused pooled kernel skips it
@@ -230,22 +244,24 @@ class SIFTExtractor(Transformer):
                         .reshape(B * N_ORI, H, Qx)
             if A1 is None:
                 omap = self._orientation_maps(imgs, sigma)
+                t0 = tick("omap", t0)
                 A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
                     .reshape(B * N_ORI, H, Qx)
-            # H-pooling as ONE tall-skinny plain GEMM.  Both the broadcast
-            # bmm Mh.T @ [B*8, H, Qx] AND the [Qy, H] @ [H, B*8*Qx] flat
-            # form land on rocBLAS's batched-general / small-M fallback
-            # (13.1 ms at 4.5 TF, profiled); transposing A1 (2.8 GB copy,
-            # sub-ms) gives [B*8*Qx, H] @ [H, Qy] — the same shape class
-            # as the A1 GEMM, which hipBLASLt runs at ~100 TF.
-            A1t = A1.transpose(1, 2).reshape(B * N_ORI * Qx, H)
-            P = torch.matmul(A1t, Mh).view(B * N_ORI, Qx, Qy)
+                t0 = tick("a1_gemm", t0)
+            # H-pooling as a batched-A @ broadcast-B bmm ([G, m, K] @
+            # [K, n] — hipBLASLt strided-batched, the fast form; both the
+            # broadcast-A form Mh.T @ [B*8, H, Qx] and a transpose+flat
+            # rewrite measured 13 ms on rocBLAS fallbacks / transpose
+            # copies).  transpose(1, 2) is stride flags only, no copy.
+            P = torch.matmul(A1.transpose(1, 2), Mh)   # [B*8, Qx, Qy]
+            t0 = tick("p_gemm", t0)
             s0, s1, s2 = P.stride()
             D = P.as_strided(
                 (B, ny, nx, N_SPATIAL, N_SPATIAL, N_ORI),
                 (N_ORI * s0, sy_y * s2, sy_x * s1,
                  sj_y * s2, sj_x * s1, s0)) \
                 .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
+            t0 = tick("gather", t0)
         else:
             Mw, nx = _sampling_matrix(W, bin_size, self.step,
                                       imgs.device, dt)
@@ -273,6 +289,11 @@ class SIFTExtractor(Transformer):
             D = D.clamp_max(self.clamp)
             D = D / D.norm(dim=2, keepdim=True).clamp_min(1e-10)
 
+        t0 = tick("norm", t0)
+        if dbg:
+            print(f"[sift bin={bin_size}] "
+                  + " ".join(f"{k}={v * 1000:.1f}ms" for k, v in _ph.items()),
+                  flush=True)
         if self.append_xy:
             span = (N_SPATIAL - 1) * bin_size
             ys = torch.arange(bin_size // 2, H - span - bin_size // 2,
