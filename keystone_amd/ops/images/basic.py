@@ -34,7 +34,12 @@ class GrayScaler(BatchTransformer):
         self._w = self._w.to(device=X.device, dtype=X.dtype)
         if X.shape[1] == 1:
             return X[:, 0]
-        return torch.einsum("bchw,c->bhw", X, self._w)
+        # explicit weighted sum: the einsum form lowers to a degenerate
+        # [B*H*W, 3] @ [3, 1] GEMM that hipBLASLt rejects
+        # (HIPBLAS_STATUS_NOT_SUPPORTED) and a slow fallback runs instead
+        # — 23.2 ms vs ~2 ms for these elementwise passes (8192x3x256x256)
+        w = self._w
+        return X[:, 0] * w[0] + X[:, 1] * w[1] + X[:, 2] * w[2]
 
     def apply(self, x):
         if x.dim() == 2:
