@@ -136,12 +136,78 @@ def _read_pgm(path: str) -> torch.Tensor:
     return img.reshape(h, w).float() / maxv
 
 
-def ImageNetLoader(path: str, label_map: Optional[dict] = None, runtime=None
-                   ) -> Tuple[ItemDataset, TensorDataset]:
-    """Directory of images (.npy/.pt/.pgm), label = parent dir name via
-    label_map or sorted-dir index (ref: ImageNetLoader — the tar-of-JPEG
-    variant needs a JPEG decoder this image does not ship)."""
+_PIL_EXT = (".jpg", ".jpeg", ".png", ".ppm", ".bmp")
+
+
+def _decode_file(fp: str, color: bool, device) -> Optional[torch.Tensor]:
+    if fp.endswith(".pgm"):
+        return _read_pgm(fp).to(device)
+    if fp.endswith(".npy"):
+        import numpy as np
+        return torch.from_numpy(np.load(fp)).float().to(device)
+    if fp.endswith(".pt"):
+        return torch.load(fp).float().to(device)
+    if fp.lower().endswith(_PIL_EXT):
+        from .input_pipeline import decode_image_bytes
+        with open(fp, "rb") as fh:
+            return decode_image_bytes(fh.read(), color=color).to(device)
+    return None
+
+
+def ImageNetLoader(path: str, label_map: Optional[dict] = None, runtime=None,
+                   color: bool = True) -> Tuple[ItemDataset, TensorDataset]:
+    """Real-format image ingestion (ref: ImageNetLoader + ImageLoaderUtils).
+
+    Two layouts:
+      * directory-per-class of image files — JPEG/PNG/PPM/BMP (PIL
+        decode), plus .pgm/.npy/.pt;
+      * a directory of ``*.tar`` shards whose members are named
+        ``<class>/<image>.JPEG`` — the reference's tar-of-JPEG sharding.
+        Tar files are sharded across ranks when there are at least
+        world_size of them (each rank streams whole shards); otherwise
+        members are sharded individually.
+    Labels come from the class directory/prefix via ``label_map`` or the
+    sorted-name index."""
     rt = runtime or get_runtime()
+    tars = sorted(f for f in os.listdir(path) if f.endswith(".tar")) \
+        if os.path.isdir(path) else []
+    imgs, labels = [], []
+    if tars:
+        import tarfile
+        from .input_pipeline import decode_image_bytes
+        # discover classes across all shards for a stable label index
+        if label_map is None:
+            names = set()
+            for t in tars:
+                with tarfile.open(os.path.join(path, t)) as tf:
+                    for m in tf.getmembers():
+                        if "/" in m.name:
+                            names.add(m.name.split("/", 1)[0])
+            label_map = {c: i for i, c in enumerate(sorted(names))}
+        if len(tars) >= rt.world_size:
+            my_tars, my_members = _shard(tars, rt), None
+        else:
+            my_tars, my_members = tars, True
+        idx = 0
+        for t in my_tars:
+            with tarfile.open(os.path.join(path, t)) as tf:
+                for m in sorted(tf.getmembers(), key=lambda m: m.name):
+                    if not m.isfile() or "/" not in m.name:
+                        continue
+                    cls = m.name.split("/", 1)[0]
+                    if cls not in label_map:
+                        continue
+                    if my_members is not None \
+                            and idx % rt.world_size != rt.rank:
+                        idx += 1
+                        continue
+                    idx += 1
+                    data = tf.extractfile(m).read()
+                    imgs.append(decode_image_bytes(data, color).to(rt.device))
+                    labels.append(label_map[cls])
+        return (from_items(imgs, rt),
+                from_rows(torch.tensor(labels, dtype=torch.long), rt))
+
     entries = []
     classes = sorted(d for d in os.listdir(path)
                      if os.path.isdir(os.path.join(path, d)))
@@ -150,17 +216,11 @@ def ImageNetLoader(path: str, label_map: Optional[dict] = None, runtime=None
         for f in sorted(os.listdir(os.path.join(path, c))):
             entries.append((os.path.join(path, c, f), cmap[c]))
     entries = _shard(entries, rt)
-    imgs, labels = [], []
     for fp, ci in entries:
-        if fp.endswith(".pgm"):
-            imgs.append(_read_pgm(fp).to(rt.device))
-        elif fp.endswith(".npy"):
-            import numpy as np
-            imgs.append(torch.from_numpy(np.load(fp)).float().to(rt.device))
-        elif fp.endswith(".pt"):
-            imgs.append(torch.load(fp).float().to(rt.device))
-        else:
+        t = _decode_file(fp, color, rt.device)
+        if t is None:
             continue
+        imgs.append(t)
         labels.append(ci)
     return (from_items(imgs, rt),
             from_rows(torch.tensor(labels, dtype=torch.long), rt))

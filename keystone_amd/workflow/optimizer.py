@@ -28,6 +28,7 @@ Rewritten for the MI355X runtime:
 
 from __future__ import annotations
 
+import os
 from typing import Callable, Dict, List, Optional
 
 import torch
@@ -289,7 +290,14 @@ def default_optimizer() -> Optimizer:
         _DEFAULT = Optimizer([
             ("cleanup", 3, [EquivalentNodeMergeRule(), UnusedBranchRemovalRule()]),
             ("fusion", 3, [ElementwiseFusionRule()]),
-            ("caching", 1, [AutoCacheRule()]),
+            # profile=True: the ICDE'17 sampled-cost greedy planner is the
+            # DEFAULT — cache insertion ordered by measured saved-seconds
+            # per byte.  Zero cost for graphs with no multi-consumer
+            # transformer nodes (the profiler only runs when candidates
+            # exist); KEYSTONE_NO_PROFILE_CACHE=1 restores the
+            # consumer-count heuristic.
+            ("caching", 1, [AutoCacheRule(
+                profile=not os.environ.get("KEYSTONE_NO_PROFILE_CACHE"))]),
         ])
     return _DEFAULT
 
