@@ -39,6 +39,7 @@ def parse_args():
     p.add_argument("--gmm-k", type=int, default=256)
     p.add_argument("--pca-dims", type=int, default=64)
     p.add_argument("--sift-step", type=int, default=6)
+    p.add_argument("--sift-scales", type=int, default=2)
     p.add_argument("--solver-epochs", type=int, default=1)
     p.add_argument("--n-sample", type=int, default=100000,
                    help="descriptors sampled per branch for PCA/GMM fits")
@@ -73,7 +74,8 @@ def main():
 
     conf = ImageNetFVConfig(
         num_classes=args.num_classes, sift_step=args.sift_step,
-        sift_bin=6, sift_scales=2, lcs_step=args.sift_step, lcs_subpatch=6,
+        sift_bin=6, sift_scales=args.sift_scales,
+        lcs_step=args.sift_step, lcs_subpatch=6,
         pca_dims=args.pca_dims, gmm_k=args.gmm_k,
         n_sample=min(args.n_sample, 50 * args.images_per_gpu * 36),
         use_lcs=not args.no_lcs, block_size=args.block_size,

@@ -109,6 +109,12 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_gmm_mstep_f32.argtypes = [c_p, c_p, c_i, c_i, c_i, c_f, c_f,
+                                         c_p, c_p, c_p, c_p, c_p, c_p]
+        lib.ks_gmm_mstep_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
@@ -337,6 +343,41 @@ def gmm_stats_into(X: torch.Tensor, P: torch.Tensor, out: torch.Tensor):
     s = out.sum(0, dtype=torch.float64)
     return (s[:K], s[K:K + K * D].view(K, D),
             s[K + K * D:K + 2 * K * D].view(K, D), s[K + 2 * K * D:])
+
+
+def gmm_stats_raw(X: torch.Tensor, P: torch.Tensor,
+                  out: torch.Tensor) -> None:
+    """E-step partials kernel ONLY (no host-side sum): pairs with
+    gmm_mstep_ for the two-launch-per-iteration EM loop."""
+    lib = _load()
+    n, D = X.shape
+    K = P.shape[1]
+    nb = out.shape[0]
+    _check(lib.ks_gmm_stats_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(P.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), nb, n, D, K, _f16_stats(),
+        _stream()), "ks_gmm_stats_f32")
+
+
+def gmm_mstep_(partials: torch.Tensor, pre,
+               n_tot: float, min_var: float, mu: torch.Tensor,
+               var: torch.Tensor, w: torch.Tensor, P: torch.Tensor,
+               ll: torch.Tensor) -> None:
+    """Fused M-step + next-iteration operand build (gmm_mstep.hip): sums
+    the E-step partials (or consumes the pre-reduced fp64 stats vector
+    ``pre`` on multi-GPU), updates (mu, var, w) in place, rebuilds the
+    fv_mfma P operand, and writes the log-likelihood scalar — ONE launch
+    replacing ~18 torch small-ops per EM iteration."""
+    lib = _load()
+    nb = partials.shape[0]
+    K, D = mu.shape
+    _check(lib.ks_gmm_mstep_f32(
+        ctypes.c_void_p(partials.data_ptr()),
+        ctypes.c_void_p(pre.data_ptr()) if pre is not None else None,
+        nb, K, D, n_tot, min_var,
+        ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
+        ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(P.data_ptr()),
+        ctypes.c_void_p(ll.data_ptr()), _stream()), "ks_gmm_mstep_f32")
 
 
 def gmm_stats_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,

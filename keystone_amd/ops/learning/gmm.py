@@ -187,10 +187,21 @@ class GaussianMixtureModelEstimator(Estimator):
                 graph = None
 
         if use_fused:
+            # two launches per EM iteration: the E-step partials kernel
+            # (fv_mfma STATS mode) + the fused M-step/operand-build kernel
+            # (gmm_mstep.hip) — the ~18 torch small-ops that rebuilt
+            # (mu, var, w, P) each iteration were ~0.7 ms of launch
+            # overhead, comparable to the E-step kernel itself.  This is
+            # the answer to the shelved hipGraph capture:
+            # same launch-count win, no graph memory-pool interaction.
             nb = min(512, max(1, (X.shape[0] + 31) // 32))
-            st_out = torch.empty(nb, K + 2 * K * D + 1,
-                                 dtype=torch.float32, device=dev)
+            width = K + 2 * K * D + 1
+            st_out = torch.empty(nb, width, dtype=torch.float32, device=dev)
             st_P = torch.zeros(160, K, dtype=torch.float32, device=dev)
+            mu_b = model.means.to(torch.float32).contiguous()
+            var_b = model.variances.to(torch.float32).contiguous()
+            w_b = model.weights.to(torch.float32).contiguous()
+            ll_b = torch.zeros(1, dtype=torch.float32, device=dev)
 
             def fill_P(mdl):
                 inv2v = 0.5 / mdl.variances
@@ -201,12 +212,30 @@ class GaussianMixtureModelEstimator(Estimator):
                                 - 0.5 * mdl.variances.log().sum(1)
                                 - 0.5 * D * math.log(2 * math.pi))
 
+            fill_P(model)      # operand for iteration 0 (init model)
+            prev_ll = -float("inf")
+            for _it in range(self.max_iterations):
+                hip_ops.gmm_stats_raw(X, st_P, st_out)
+                pre = None
+                if rt.is_distributed:
+                    pre = st_out.sum(0, dtype=torch.float64)
+                    rt.all_reduce_(pre)
+                hip_ops.gmm_mstep_(st_out, pre, float(n_tot),
+                                   self.min_variance, mu_b, var_b, w_b,
+                                   st_P, ll_b)
+                if (_it & 3) == 3:
+                    ll_now = ll_b.item() / n_tot
+                    if abs(ll_now - prev_ll) < self.tol * max(1.0, abs(prev_ll)):
+                        break
+                    prev_ll = ll_now
+            model = GaussianMixtureModel(mu_b.to(dt), var_b.to(dt),
+                                         w_b.to(dt))
+            tick(f"em_loop({_it + 1} iters)", t)
+            return model
+
         prev_ll = -float("inf")
         for _it in range(self.max_iterations):
-            if use_fused:
-                fill_P(model)
-                Nk, Sx, Sxx, ll = hip_ops.gmm_stats_into(X, st_P, st_out)
-            elif graph is not None:
+            if graph is not None:
                 fill_coeffs(model)
                 graph.replay()
                 Nk = g_out[0].clone()

@@ -33,7 +33,7 @@ def build_and_run(n, d, feats, classes):
     Y = ka.from_rows(
         ClassLabelIndicatorsFromIntLabels(classes).apply_batch(y.data).float())
     featurize = StandardScaler().with_data(X).and_then(
-        CosineRandomFeatures(d, feats, gamma=0.05, seed=1))
+        CosineRandomFeatures(d, feats, gamma=1.0 / d, seed=1))
     pipe = featurize.and_then(
         BlockLeastSquaresEstimator(block_size=1024, num_iter=3, lam=1e-3),
         X, Y).and_then(MaxClassifier())
