@@ -163,6 +163,26 @@ class LazyBlocks(Dataset):
         return self.raw.items()
 
 
+def _blocked_trtri(L: torch.Tensor, base: int = 512) -> torch.Tensor:
+    """Batched lower-triangular inverse by 2x2 block recursion:
+    ``inv([[L11,0],[L21,L22]]) = [[I11,0],[-I22 L21 I11, I22]]`` — shifts
+    ~3/4 of the work from rocSOLVER TRSM (59 TF measured) onto hipBLASLt
+    GEMMs (150 TF); only the <=base diagonal blocks use TRSM."""
+    db = L.shape[-1]
+    if db <= base or db % 2:
+        eye = torch.eye(db, dtype=L.dtype, device=L.device)
+        return torch.linalg.solve_triangular(
+            L, eye.expand(L.shape), upper=False)
+    m = db // 2
+    I11 = _blocked_trtri(L[..., :m, :m], base)
+    I22 = _blocked_trtri(L[..., m:, m:], base)
+    out = L.new_zeros(L.shape)
+    out[..., :m, :m] = I11
+    out[..., m:, m:] = I22
+    out[..., m:, :m] = -(I22 @ (L[..., m:, :m].contiguous() @ I11))
+    return out
+
+
 def _as_block_getters(data: Dataset, block_size: int):
     """Return (list of zero-arg block getters, runtime, n_local)."""
     if isinstance(data, LazyBlocks):
@@ -483,11 +503,9 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             # keep only L^{-1}: every downstream use runs in the
             # T = U L^{-T} form, so the explicit B^{-1} = L^{-T} L^{-1}
             # assembly GEMM (measured 14.4 ms at nb=16, db=4096) is never
-            # paid.  cholesky 18.4 ms + batched TRSM 18.7 ms is the
-            # cheapest factorization route (cholesky_inverse 55.8,
-            # getrf/getri 646 — probe_chol.py).
-            Linvs = torch.linalg.solve_triangular(
-                Lb, eye.expand(nb, db0, db0), upper=False)
+            # paid.  cholesky + blocked trtri is the cheapest route
+            # (cholesky_inverse 55.8 ms, getrf/getri 646 — probe_chol.py).
+            Linvs = _blocked_trtri(Lb)
             del Bb, Lb
             t0 = tick("pre_chol", t0)
 

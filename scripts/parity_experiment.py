@@ -73,8 +73,14 @@ def heldout_saturation(rt):
 
 
 def bcd_vs_oracle(rt):
-    """Benchmark-width featurization, then BCD vs the exact dual ridge."""
-    train, ytr = synthetic_images(2048, h=256, w=256, num_classes=16,
+    """Benchmark-width featurization, then BCD vs the exact dual ridge.
+
+    n = 8192 > block_size = 4096 so blocks are thin: with n < d_b every
+    single block is overcomplete and can near-interpolate the residual,
+    which drives BCD into its pathologically slow regime (measured: a
+    2.6e-3 prediction gap static to 8 digits across epochs at n = 2048).
+    The reference always ran n >> d_b (1.28M images)."""
+    train, ytr = synthetic_images(8192, h=256, w=256, num_classes=16,
                                   seed=7, runtime=rt, device=rt.device,
                                   channels=3, class_structure=True)
     conf = ImageNetFVConfig(num_classes=16, sift_step=6, sift_bin=6,
@@ -99,8 +105,11 @@ def bcd_vs_oracle(rt):
     pred_star = G @ alpha
 
     out = {"n": n, "d": d}
-    # fp32 (production dtype): plateaus at the fp32 GEMM/Cholesky noise
-    # floor; fp64 shows the true BCD convergence behaviour
+    # Wide regime (d = 65,536 >> n): predictions agree with the oracle to
+    # ~1% after ONE epoch; the remaining gap is BCD's known glacial tail
+    # in the overparameterized regime (CPU study in the repo history:
+    # identical in fp64, so not a precision artifact).  The reference's
+    # production regime was n = 1.28M >> d.
     for tag, Xs, Ys in (("fp32", X, Y), ("fp64", X64, Y.to(torch.float64))):
         for epochs in (1, 3, 5):
             m = BlockLeastSquaresEstimator(block_size=4096, num_iter=epochs,
@@ -111,6 +120,24 @@ def bcd_vs_oracle(rt):
             out[f"{tag}_epochs_{epochs}"] = rel
             print(f"[bcd-vs-oracle] {tag} epochs={epochs} d={d} "
                   f"rel_pred_err={rel:.2e}", flush=True)
+
+    # Classical regime (n > d — the reference's production shape): a
+    # 4096-wide slice of the SAME pipeline features; BCD must converge
+    # monotonically to the exact primal ridge solution.
+    dn = 4096
+    Xn = X64[:, :dn].contiguous()
+    Gp = Xn.T @ Xn + lam * torch.eye(dn, dtype=torch.float64,
+                                     device=X.device)
+    W_star = torch.linalg.solve(Gp, Xn.T @ Y.to(torch.float64))
+    for epochs in (1, 3, 5, 10):
+        m = BlockLeastSquaresEstimator(block_size=1024, num_iter=epochs,
+                                       lam=lam, fit_intercept=False).fit(
+            ka.from_rows(Xn), ka.from_rows(Y.to(torch.float64)))
+        W = torch.cat(m.blocks, dim=0)
+        rel = ((W - W_star).norm() / W_star.norm()).item()
+        out[f"narrow_epochs_{epochs}"] = rel
+        print(f"[bcd-vs-oracle] narrow n>d epochs={epochs} d={dn} "
+              f"rel_W_err={rel:.2e}", flush=True)
     return out
 
 
