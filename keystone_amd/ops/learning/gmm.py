@@ -216,9 +216,10 @@ class GaussianMixtureModelEstimator(Estimator):
             prev_ll = -float("inf")
             for _it in range(self.max_iterations):
                 hip_ops.gmm_stats_raw(X, st_P, st_out)
-                pre = None
+                # pre-reduce with torch's tree reduction: the kernel's
+                # per-wave nb-loop measured 0.33 ms vs sum+mstep ~0.05
+                pre = st_out.sum(0, dtype=torch.float64)
                 if rt.is_distributed:
-                    pre = st_out.sum(0, dtype=torch.float64)
                     rt.all_reduce_(pre)
                 hip_ops.gmm_mstep_(st_out, pre, float(n_tot),
                                    self.min_variance, mu_b, var_b, w_b,
