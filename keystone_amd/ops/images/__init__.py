@@ -1,5 +1,5 @@
 from .basic import (GrayScaler, ImageVectorizer, PixelScaler, Pooler,
-                    SymmetricRectifier, Windower)
+                    SymmetricRectifier, FastWindower, Windower)
 from .convolver import Convolver
 from .patches import CenterCornerPatcher, RandomImageTransformer, RandomPatcher
 from .sift import SIFTExtractor
@@ -9,7 +9,7 @@ from .fisher_vector import FisherVector
 
 __all__ = [
     "GrayScaler", "PixelScaler", "ImageVectorizer", "Windower", "Pooler",
-    "SymmetricRectifier", "Convolver", "CenterCornerPatcher",
+    "SymmetricRectifier", "FastWindower", "Convolver", "CenterCornerPatcher",
     "RandomPatcher", "RandomImageTransformer", "SIFTExtractor",
     "LCSExtractor", "DaisyExtractor", "FisherVector",
 ]

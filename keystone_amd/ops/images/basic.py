@@ -85,6 +85,34 @@ class Windower(Transformer):
         return F.unfold(X, self.size, stride=self.stride).transpose(1, 2).contiguous()
 
 
+class FastWindower(Transformer):
+    """Same patch stream as Windower via a zero-copy as_strided view
+    (ref: FastWindower): no im2col kernel — the gather happens once at
+    the consumer's reshape instead of materializing inside the op."""
+
+    def __init__(self, stride: int, size: int):
+        self.stride = stride
+        self.size = size
+
+    def apply_batch(self, X: torch.Tensor) -> torch.Tensor:
+        """[B, C?, H, W] -> [B, n_patches, C*size*size]"""
+        if X.dim() == 3:
+            X = X.unsqueeze(1)
+        B, C, H, W = X.shape
+        k, st = self.size, self.stride
+        ny = (H - k) // st + 1
+        nx = (W - k) // st + 1
+        s0, s1, s2, s3 = X.stride()
+        v = X.as_strided((B, ny, nx, C, k, k),
+                         (s0, st * s2, st * s3, s1, s2, s3))
+        return v.reshape(B, ny * nx, C * k * k)
+
+    def apply(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() == 2:
+            x = x.unsqueeze(0)
+        return self.apply_batch(x.unsqueeze(0)).squeeze(0)
+
+
 class Pooler(BatchTransformer):
     """Spatial pooling with stride/size and a pixel function
     (ref: Pooler(stride, poolSize, pixelFunc, poolFunc) — the CIFAR

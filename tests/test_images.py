@@ -266,3 +266,16 @@ def test_sift_merged_pooling_matches_sampling_matrix():
                              (s0, syy * s2, syx * s3, sjy * s2, sjx * s3, s1)) \
             .reshape(B, ny * nx, NS * NS * NO)
         assert (Dref - Dfast).abs().max().item() < 1e-5
+
+
+def test_fast_windower_matches_windower():
+    from keystone_amd.ops.images.basic import FastWindower, Windower
+    torch.manual_seed(4)
+    for shape in [(2, 3, 17, 23), (3, 16, 16)]:
+        X = torch.randn(*shape)
+        for stride, size in [(2, 5), (3, 4)]:
+            slow = Windower(stride, size).apply_batch(X)
+            fast = FastWindower(stride, size).apply_batch(X)
+            assert torch.equal(slow, fast), (shape, stride, size)
+    x = torch.randn(3, 12, 14)
+    assert torch.equal(Windower(2, 4).apply(x), FastWindower(2, 4).apply(x))
