@@ -163,6 +163,34 @@ class LazyBlocks(Dataset):
         return self.raw.items()
 
 
+def _chol_linv(A: torch.Tensor, base: int = 512) -> torch.Tensor:
+    """Batched L^{-1} of the Cholesky factor of SPD ``A`` WITHOUT ever
+    forming L: Schur recursion
+        chol([[A11,.],[A21,A22]]) = [[L11,0],[L21,L22]],
+        L21 = A21 L11^{-T},  S = A22 - L21 L21^T,
+        inv  = [[I11,0],[-I22 (L21 I11), I22]]
+    — every step except the <=base diagonal factorizations is a hipBLASLt
+    GEMM.  rocSOLVER's batched potrf at 4096 runs ~40 TF and the follow-up
+    TRSM ~59 (probe_chol.py); this route keeps ~3/4 of the flops on the
+    150 TF GEMM path."""
+    db = A.shape[-1]
+    if db <= base or db % 2:
+        L = torch.linalg.cholesky(A)
+        eye = torch.eye(db, dtype=A.dtype, device=A.device)
+        return torch.linalg.solve_triangular(L, eye.expand(L.shape),
+                                             upper=False)
+    m = db // 2
+    I11 = _chol_linv(A[..., :m, :m], base)
+    L21 = A[..., m:, :m].contiguous() @ I11.transpose(-1, -2)
+    S = A[..., m:, m:] - L21 @ L21.transpose(-1, -2)
+    I22 = _chol_linv(S, base)
+    out = A.new_zeros(A.shape)
+    out[..., :m, :m] = I11
+    out[..., m:, m:] = I22
+    out[..., m:, :m] = -(I22 @ (L21 @ I11))
+    return out
+
+
 def _blocked_trtri(L: torch.Tensor, base: int = 512) -> torch.Tensor:
     """Batched lower-triangular inverse by 2x2 block recursion:
     ``inv([[L11,0],[L21,L22]]) = [[I11,0],[-I22 L21 I11, I22]]`` — shifts
@@ -499,14 +527,13 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         if uniform and own:
             eye = torch.eye(db0, dtype=sdt, device=Y.device)
             Bb = base_w * grams_pre + self.lam * eye
-            Lb = torch.linalg.cholesky(Bb)
             # keep only L^{-1}: every downstream use runs in the
-            # T = U L^{-T} form, so the explicit B^{-1} = L^{-T} L^{-1}
-            # assembly GEMM (measured 14.4 ms at nb=16, db=4096) is never
-            # paid.  cholesky + blocked trtri is the cheapest route
-            # (cholesky_inverse 55.8 ms, getrf/getri 646 — probe_chol.py).
-            Linvs = _blocked_trtri(Lb)
-            del Bb, Lb
+            # T = U L^{-T} form, so neither L nor B^{-1} is materialized;
+            # the Schur-recursive factorization keeps the work on GEMMs
+            # (cholesky 18.4 + TRSM 18.7 ms at nb=16, db=4096 before;
+            # cholesky_inverse 55.8, getrf/getri 646 — probe_chol.py).
+            Linvs = _chol_linv(Bb)
+            del Bb
             t0 = tick("pre_chol", t0)
 
         # ---- loop-invariant per-class group tensors.  Everything here

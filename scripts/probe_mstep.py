@@ -50,3 +50,11 @@ t("blocked_trtri", lambda: _blocked_trtri(L), 3)
 I1 = torch.linalg.solve_triangular(L, eye.expand(nbq, db, db), upper=False)
 I2 = _blocked_trtri(L)
 print("trtri err:", (I1 - I2).abs().max().item())
+
+from keystone_amd.ops.learning.block_ls import _chol_linv
+t("chol+trtri current", lambda: _blocked_trtri(torch.linalg.cholesky(B)), 3)
+t("schur _chol_linv", lambda: _chol_linv(B), 3)
+I3 = _chol_linv(B)
+Lref = torch.linalg.cholesky(B)
+Iref = torch.linalg.solve_triangular(Lref, eye.expand(nbq, db, db), upper=False)
+print("chol_linv err:", ((I3 - Iref).abs().max() / Iref.abs().max()).item())
