@@ -142,12 +142,16 @@ def main():
     if not args.cpu_smoke:
         from keystone_amd.loaders.input_pipeline import (ImageBatchPipeline,
                                                          encode_jpeg_batch)
-        n_inp = min(2048, args.images_per_gpu)
-        blobs = encode_jpeg_batch(imgs.data[:n_inp])
+        # per-rank corpus shrinks with world size: decode threads are a
+        # HOST resource shared by all ranks on the node
+        n_inp = min(2048 // rt.world_size, args.images_per_gpu)
+        blobs = encode_jpeg_batch(imgs.data[:n_inp],
+                                  workers=max(2, 16 // rt.world_size))
         labs = list(range(n_inp))
         from keystone_amd.parallel.dataset import TensorDataset
         loader = ImageBatchPipeline(blobs, labs, batch_size=256,
-                                    color=not args.gray, device=dev)
+                                    color=not args.gray, device=dev,
+                                    workers=max(2, 16 // rt.world_size))
         for batch, _l in loader:        # warm (decode caches, pin alloc)
             pipe.apply(TensorDataset(batch, rt)).get()
             break
