@@ -288,10 +288,15 @@ def fv_mfma_supported(D: int, K: int) -> bool:
 
 
 def _f16_stats() -> int:
-    """f16 stats-MFMA pass (v_mfma_f32_32x32x16_f16, 16x the f32 form's
-    issue throughput; gamma/x inputs round to f16, f32 accumulate).
-    KEYSTONE_FV_F32STATS=1 selects the exact-f32 pass."""
-    return 0 if os.environ.get("KEYSTONE_FV_F32STATS") else 1
+    """Stats-pass mode: 1 = f16 MFMA (v_mfma_f32_32x32x16_f16, 16x the
+    f32 form's issue throughput; gamma/x round to f16, f32 accumulate),
+    2 = f16 + K-split (two co-resident workgroups per image at K=256),
+    0 = exact f32 (KEYSTONE_FV_F32STATS=1)."""
+    if os.environ.get("KEYSTONE_FV_F32STATS"):
+        return 0
+    if os.environ.get("KEYSTONE_FV_SPLIT"):
+        return 2
+    return 1
 
 
 def fisher_vector_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,

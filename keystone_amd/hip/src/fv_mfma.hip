@@ -47,8 +47,17 @@ typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 // writes its PARTIAL sufficient statistics to out[b] = [S0 (KT) | S1
 // (KT*D) | S2 (KT*D) | ll (1)] — the GMM EM E-step in one launch (sum the
 // partials host-side).  FV mode (STATS=false): one image per block.
-template <int KT, bool STATS, bool F16S>
-__global__ __launch_bounds__(256, 1)
+// SPLIT (f16-only, OPT-IN via KEYSTONE_FV_SPLIT=1): TWO workgroups per
+// image (gridDim.y == 2) — each computes the full logits+softmax but
+// accumulates stats for only HALF the components.  Halved sacc (80
+// AGPRs) + trimmed LDS (78.5 KB at K=256) let 2 workgroups co-reside
+// per CU, aiming to fill the 69% non-MFMA time seen at occupancy 1.
+// MEASURED: 48.6 ms vs 33.9 for the joint-span schedule at K=256/B=8192
+// — the duplicated logits MFMA (+93% issue work) and the 5-barrier
+// schedule cost more than the co-residency buys.  Kept opt-in as the
+// documented A/B; the joint-span f16 kernel is the production path.
+template <int KT, bool STATS, bool F16S, bool SPLIT>
+__global__ __launch_bounds__(256, SPLIT ? 2 : 1)
 void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     const float* __restrict__ P,    // [XCP, KT] padded
                     const float* __restrict__ mu,   // [KT, D]
@@ -58,11 +67,14 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     int n, int D) {
     constexpr int KT32 = KT / 32;               // logits col-tiles
     constexpr int LT = (KT32 + 3) / 4;          // logits tiles per wave (max)
-    constexpr int ST = 5 * ((KT32 + 3) / 4);    // stats tiles per wave (max)
+    constexpr int KH = SPLIT ? KT / 2 : KT;     // stats component span
+    constexpr int KH32 = KH / 32;
+    constexpr int ST = 5 * ((KH32 + 3) / 4);    // stats tiles per wave (max)
     // when KT32 is a multiple of 4 every wave's tiles are valid: the tile
     // guards below become compile-time (a RUNTIME guard around each load
     // makes hipcc branch + wait vmcnt(0) per element — guide §5 trap 4c)
     constexpr bool FULL = (KT32 % 4 == 0);
+    constexpr bool FULLS = (KH32 % 4 == 0);
     // gamma row stride: (KT+8) % 32 == 8 puts softmax lane (row, sub)
     // (4 rows x 8 sublanes per 32-lane phase) on bank row*8 + sub -> all
     // 32 banks distinct.  The previous KT+1 stride gave bank row+sub:
@@ -81,10 +93,12 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     float* Gs0 = Xs0 + NBUF * CHUNK * XSTRIDE;  // [CHUNK][GST]
     float* S0s = Gs0 + CHUNK * GST;             // [KT] broadcast (+ll)
     // f16 region (F16S only): transposed panels for the f16 stats MFMA
-    _Float16* XhT0 = (_Float16*)(smem + (size_t)F32W * 4);  // [2][XCP][SH]
-    _Float16* GhT = XhT0 + 2 * XCP * SH;                    // [KT][SH]
+    constexpr int NXH = SPLIT ? 1 : 2;          // XhT buffers
+    _Float16* XhT0 = (_Float16*)(smem + (size_t)F32W * 4);  // [NXH][XCP][SH]
+    _Float16* GhT = XhT0 + NXH * XCP * SH;                  // [KH][SH]
 
     const int img = blockIdx.x;
+    const int kh0 = SPLIT ? (int)blockIdx.y * KH : 0;   // component base
     const float* Xi = STATS ? X : X + (size_t)img * n * D;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -129,8 +143,8 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             Xs0[(buf * CHUNK + r) * XSTRIDE + c] = 0.0f;
         }
     if (F16S) {
-        // zero the f16 pad rows (cols 129..159 of Xext) in both buffers
-        for (int i = tid; i < 2 * (XCP - XC + 1) * SH; i += 256) {
+        // zero the f16 pad rows (cols 129..159 of Xext) in every buffer
+        for (int i = tid; i < NXH * (XCP - XC + 1) * SH; i += 256) {
             const int buf = i / ((XCP - XC + 1) * SH);
             const int j = i % ((XCP - XC + 1) * SH);
             XhT0[(buf * XCP + XC - 1) * SH + j] = (_Float16)0.0f;
@@ -194,14 +208,14 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // f16 stats-MFMA span for the chunk staged in XhT slot (ci & 1),
     // reading the gamma panel GhT (single-buffered; see barrier notes)
     auto stats_f16 = [&](int ci) {
-        const _Float16* Xh = XhT0 + (size_t)(ci & 1) * XCP * SH;
+        const _Float16* Xh = XhT0 + (SPLIT ? 0 : (size_t)(ci & 1) * XCP * SH);
 #pragma unroll
         for (int kk = 0; kk < CHUNK; kk += 16) {
 #pragma unroll
             for (int t = 0; t < ST; ++t) {
                 const int ti = wave + 4 * (t / 5);
                 const int tj = t % 5;
-                if (FULL || ti < KT32) {
+                if (FULLS || ti < KH32) {
                     const f16x8 a = *(const f16x8*)
                         &GhT[(size_t)(ti * 32 + l31) * SH + kk + 8 * lk];
                     const f16x8 b = *(const f16x8*)
@@ -229,6 +243,119 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         }
     };
 
+    // logits(c) MFMA span over Xs (shared by both schedules)
+    auto logits_span = [&](f32x16* lacc) {
+#pragma unroll
+        for (int t = 0; t < LT; ++t) lacc[t] = (f32x16)(0.0f);
+#pragma unroll 5
+        for (int kk = 0; kk < XC; kk += 2) {
+            const float a = Xs0[l31 * XSTRIDE + kk + lk];
+#pragma unroll
+            for (int t = 0; t < LT; ++t) {
+                const int ct = wave + 4 * t;
+                if (FULL || ct < KT32) {
+                    const float b = P[(size_t)(kk + lk) * KT + ct * 32 + l31];
+                    lacc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                        a, b, lacc[t], 0, 0, 0);
+                }
+            }
+        }
+    };
+    auto gs_write = [&](const f32x16* lacc) {
+#pragma unroll
+        for (int t = 0; t < LT; ++t) {
+            const int ct = wave + 4 * t;
+            if (FULL || ct < KT32) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                    Gs0[row * GST + ct * 32 + l31] = lacc[t][r];
+                }
+            }
+        }
+    };
+    // softmax over full K; F16S writes the own-half gamma panel
+    auto softmax_chunk = [&](int c) {
+        const int row = tid >> 3;
+        const int sub = tid & 7;
+        float m = -1e30f;
+        for (int k = sub; k < KT; k += 8)
+            m = fmaxf(m, Gs0[row * GST + k]);
+#pragma unroll
+        for (int off = 4; off; off >>= 1)
+            m = fmaxf(m, __shfl_xor(m, off, 8));
+        float s = 0.0f;
+        for (int k = sub; k < KT; k += 8) {
+            float e = __expf(Gs0[row * GST + k] - m);
+            Gs0[row * GST + k] = e;
+            s += e;
+        }
+#pragma unroll
+        for (int off = 4; off; off >>= 1)
+            s += __shfl_xor(s, off, 8);
+        const float inv = 1.0f / s;
+        for (int k = sub; k < KT; k += 8) {
+            if (F16S) {
+                if (!SPLIT || ((unsigned)(k - kh0) < (unsigned)KH))
+                    GhT[(size_t)(k - kh0) * SH + row] =
+                        (_Float16)(Gs0[row * GST + k] * inv);
+            } else {
+                Gs0[row * GST + k] *= inv;
+            }
+        }
+        if (STATS && sub == 0 && (!SPLIT || kh0 == 0)
+                && row < n - (int)chunk_start(c))
+            atomicAdd(&S0s[KT], m + __logf(s));   // logsumexp sum
+    };
+
+    if (SPLIT) {
+        // 5-barrier schedule: [logits] [Gs + Xs(c+1)] [softmax -> GhT]
+        // [stats(c)] [XhT(c+1)].  Spans are no longer joint — the
+        // co-resident second workgroup fills the gaps instead.
+        for (int c = 0; c < nc; ++c) {
+            const bool more = c + 1 < nc;
+            const int rows1 = more
+                ? min(CHUNK, n - (int)chunk_start(c + 1)) : 0;
+            if (more) stage_load((int)chunk_start(c + 1), rows1);
+            f32x16 lacc[LT];
+            logits_span(lacc);
+            __syncthreads();
+            gs_write(lacc);
+            if (more) {
+                // f32 staging only (XhT(c) still feeds stats(c))
+#pragma unroll
+                for (int q = 0; q < 8; ++q) {
+                    Xs0[(sr0 + q) * XSTRIDE + sd] = xreg[q] * xreg[q];
+                    Xs0[(sr0 + q) * XSTRIDE + DT + sd] = xreg[q];
+                }
+                if (tid < CHUNK)
+                    Xs0[tid * XSTRIDE + 2 * DT] = (tid < rows1) ? 1.0f : 0.0f;
+            }
+            __syncthreads();
+            softmax_chunk(c);
+            __syncthreads();
+            stats_f16(c);
+            __syncthreads();
+            if (more) {
+                f16x8 hx, hx2;
+#pragma unroll
+                for (int q = 0; q < 8; ++q) {
+                    hx[q] = (_Float16)xreg[q];
+                    hx2[q] = (_Float16)(xreg[q] * xreg[q]);
+                }
+                *(f16x8*)&XhT0[(size_t)sd * SH + sr0] = hx2;
+                *(f16x8*)&XhT0[(size_t)(DT + sd) * SH + sr0] = hx;
+                if (sd == 0) {
+                    f16x8 ones;
+#pragma unroll
+                    for (int q = 0; q < 8; ++q)
+                        ones[q] = (_Float16)((sr0 + q < rows1) ? 1.0f : 0.0f);
+                    *(f16x8*)&XhT0[(size_t)(2 * DT) * SH + sr0] = ones;
+                }
+                __syncthreads();
+            }
+        }
+    } else
     for (int c = 0; c < nc; ++c) {
         const int cur = c & 1;
         // F16S: ONE f32 staging buffer (logits(c) consumes it inside this
@@ -319,8 +446,9 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         __syncthreads();
     }
 
-    // ---- trailing stats for the last chunk
-    if (F16S) {
+    // ---- trailing stats for the last chunk (SPLIT runs stats in-loop)
+    if (SPLIT) {
+    } else if (F16S) {
         stats_f16(nc - 1);
     } else {
         const int cur = (nc - 1) & 1;
@@ -336,12 +464,12 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         for (int t = 0; t < ST; ++t) {
             const int ti = wave + 4 * (t / 5);
             const int tj = t % 5;
-            if (!(FULL || ti < KT32)) continue;
+            if (!(FULLS || ti < KH32)) continue;
             if (tj == 4) {
                 if (l31 == 0) {
 #pragma unroll
                     for (int r = 0; r < 16; ++r) {
-                        const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                        const int kcomp = kh0 + ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
                         ob[kcomp] = sacc[t][r];
                     }
                 }
@@ -351,14 +479,14 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                     const bool is_s2 = tj < 2;   // x^2 columns
 #pragma unroll
                     for (int r = 0; r < 16; ++r) {
-                        const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                        const int kcomp = kh0 + ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
                         ob[KT + (is_s2 ? KT * D : 0) + kcomp * D + d] =
                             sacc[t][r];
                     }
                 }
             }
         }
-        if (tid == 0) ob[KT + 2 * KT * D] = S0s[KT];
+        if (tid == 0 && kh0 == 0) ob[KT + 2 * KT * D] = S0s[KT];
         return;
     }
 
@@ -368,10 +496,10 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     for (int t = 0; t < ST; ++t) {
         const int ti = wave + 4 * (t / 5);
         const int tj = t % 5;
-        if (ti < KT32 && tj == 4 && l31 == 0) {
+        if (ti < KH32 && tj == 4 && l31 == 0) {
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+                const int kcomp = kh0 + ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
                 S0s[kcomp] = sacc[t][r];
             }
         }
@@ -386,7 +514,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     for (int t = 0; t < ST; ++t) {
         const int ti = wave + 4 * (t / 5);
         const int tj = t % 5;
-        if (ti >= KT32 || tj >= 2) continue;      // S2 tiles drive the pair
+        if (ti >= KH32 || tj >= 2) continue;      // S2 tiles drive the pair
         const int d = tj * 32 + l31;
         if (d >= D) continue;
         // find the matching S1 tile (tj + 2) in this wave's tile list:
@@ -396,7 +524,7 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
         const f32x16 s1 = sacc[t + 2];
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-            const int kcomp = ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
+            const int kcomp = kh0 + ti * 32 + (r & 3) + 8 * (r >> 2) + 4 * lk;
             const float s0 = S0s[kcomp];
             const float m = mu[kcomp * D + d];
             const float v = var[kcomp * D + d];
@@ -412,12 +540,15 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     }
 }
 
-static size_t fv_lds_bytes(int K, bool f16s) {
-    const int nbuf = f16s ? 1 : 2;
+// mode: 0 = f32 stats, 1 = f16 stats, 2 = f16 stats + K-split (2 WG/img)
+static size_t fv_lds_bytes(int K, int mode) {
+    const int nbuf = mode ? 1 : 2;
     size_t f32w = ((size_t)(nbuf * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1)
                    + 3) & ~(size_t)3;
     size_t b = f32w * 4;
-    if (f16s) b += (size_t)(2 * XCP * SH + K * SH) * 2;
+    const int nxh = (mode == 2) ? 1 : 2;
+    const int kh = (mode == 2) ? K / 2 : K;
+    if (mode) b += (size_t)(nxh * XCP * SH + kh * SH) * 2;
     return b;
 }
 
@@ -427,17 +558,19 @@ KS_EXPORT int ks_fv_mfma_f32(const float* X, const float* P, const float* mu,
                              hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    const bool h = f16_stats != 0;
-    size_t lds = fv_lds_bytes(K, h);
+    int mode = f16_stats;
+    if (mode == 2 && K != 256) mode = 1;   // split only tuned for K=256
+    size_t lds = fv_lds_bytes(K, mode);
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
-    dim3 grid(B), block(256);
-    switch (K * 2 + (h ? 1 : 0)) {
-        case 129: hipLaunchKernelGGL((fv_mfma_kernel<64, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 128: hipLaunchKernelGGL((fv_mfma_kernel<64, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 257: hipLaunchKernelGGL((fv_mfma_kernel<128, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 256: hipLaunchKernelGGL((fv_mfma_kernel<128, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 513: hipLaunchKernelGGL((fv_mfma_kernel<256, false, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
-        case 512: hipLaunchKernelGGL((fv_mfma_kernel<256, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+    dim3 grid(B, mode == 2 ? 2 : 1), block(256);
+    switch (K * 4 + mode) {
+        case 257: hipLaunchKernelGGL((fv_mfma_kernel<64, false, true, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<64, false, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 513: hipLaunchKernelGGL((fv_mfma_kernel<128, false, true, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 512: hipLaunchKernelGGL((fv_mfma_kernel<128, false, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 1025: hipLaunchKernelGGL((fv_mfma_kernel<256, false, true, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 1024: hipLaunchKernelGGL((fv_mfma_kernel<256, false, false, false>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
+        case 1026: hipLaunchKernelGGL((fv_mfma_kernel<256, false, true, true>), grid, block, lds, stream, X, P, mu, var, w, out, n, D); break;
         default: return (int)hipErrorInvalidValue;
     }
     return (int)hipGetLastError();
@@ -450,17 +583,19 @@ KS_EXPORT int ks_gmm_stats_f32(const float* X, const float* P, float* out,
                                int f16_stats, hipStream_t stream) {
     if (D > DT || (K != 64 && K != 128 && K != 256))
         return (int)hipErrorInvalidValue;
-    const bool h = f16_stats != 0;
-    size_t lds = fv_lds_bytes(K, h);
+    int mode = f16_stats;
+    if (mode == 2 && K != 256) mode = 1;
+    size_t lds = fv_lds_bytes(K, mode);
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
-    dim3 grid(n_blocks), block(256);
-    switch (K * 2 + (h ? 1 : 0)) {
-        case 129: hipLaunchKernelGGL((fv_mfma_kernel<64, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 128: hipLaunchKernelGGL((fv_mfma_kernel<64, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 257: hipLaunchKernelGGL((fv_mfma_kernel<128, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 256: hipLaunchKernelGGL((fv_mfma_kernel<128, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 513: hipLaunchKernelGGL((fv_mfma_kernel<256, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
-        case 512: hipLaunchKernelGGL((fv_mfma_kernel<256, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+    dim3 grid(n_blocks, mode == 2 ? 2 : 1), block(256);
+    switch (K * 4 + mode) {
+        case 257: hipLaunchKernelGGL((fv_mfma_kernel<64, true, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 256: hipLaunchKernelGGL((fv_mfma_kernel<64, true, false, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 513: hipLaunchKernelGGL((fv_mfma_kernel<128, true, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 512: hipLaunchKernelGGL((fv_mfma_kernel<128, true, false, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 1025: hipLaunchKernelGGL((fv_mfma_kernel<256, true, true, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 1024: hipLaunchKernelGGL((fv_mfma_kernel<256, true, false, false>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
+        case 1026: hipLaunchKernelGGL((fv_mfma_kernel<256, true, true, true>), grid, block, lds, stream, X, P, nullptr, nullptr, nullptr, out, n, D); break;
         default: return (int)hipErrorInvalidValue;
     }
     return (int)hipGetLastError();
