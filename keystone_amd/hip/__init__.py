@@ -115,6 +115,11 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_woodbury_f32.argtypes = [c_p] * 6 + [c_i, c_i, c_p]
+        lib.ks_woodbury_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
@@ -383,6 +388,27 @@ def gmm_mstep_(partials: torch.Tensor, pre,
         ctypes.c_void_p(mu.data_ptr()), ctypes.c_void_p(var.data_ptr()),
         ctypes.c_void_p(w.data_ptr()), ctypes.c_void_p(P.data_ptr()),
         ctypes.c_void_p(ll.data_ptr()), _stream()), "ks_gmm_mstep_f32")
+
+
+WOODBURY_NCMAX = 32
+
+
+def woodbury_yz(T_all: torch.Tensor, LRt: torch.Tensor,
+                starts: torch.Tensor, ncs: torch.Tensor):
+    """Per-class Woodbury inner solve (woodbury.hip): one workgroup per
+    class computes z = (I + T_c T_c^T)^{-1} (T_c LR_c) and returns
+    Yz [G, db] = T_c^T z (zero rows for empty classes) plus a skipped
+    mask for classes with nc > 32 (torch fallback)."""
+    lib = _load()
+    G, db = LRt.shape
+    Yz = torch.empty(G, db, dtype=torch.float32, device=T_all.device)
+    skipped = torch.zeros(G, dtype=torch.int32, device=T_all.device)
+    _check(lib.ks_woodbury_f32(
+        ctypes.c_void_p(T_all.data_ptr()), ctypes.c_void_p(LRt.data_ptr()),
+        ctypes.c_void_p(starts.data_ptr()), ctypes.c_void_p(ncs.data_ptr()),
+        ctypes.c_void_p(Yz.data_ptr()), ctypes.c_void_p(skipped.data_ptr()),
+        G, db, _stream()), "ks_woodbury_f32")
+    return Yz, skipped
 
 
 def gmm_stats_mfma(X: torch.Tensor, mu: torch.Tensor, var: torch.Tensor,

@@ -1,0 +1,153 @@
+// Per-class Woodbury inner solve for the weighted block solver (gfx950).
+//
+// ref: BlockWeightedLeastSquaresEstimator — per score
+// column c the update needs
+//     z_c  = (I + T_c T_c^T)^{-1} (T_c L^{-1} r_c),
+//     Yz_c = T_c^T z_c                                  [db]
+// with T_c = the class's rows of T = U L^{-T} (nc rows, nc ~ 8 at
+// ImageNet scale).  The torch path materialized padded [G, ncmax, db]
+// gathers, a mask multiply, a batched bmm for S = I + T T^T, a batched
+// solve and a weighted reduce (~0.45 ms per 4096-block at K=1000); this
+// kernel does all of it in ONE launch, one workgroup per class:
+//
+//   chunked over db: stage T_c tile + LR chunk in LDS, accumulate the
+//   nc(nc+1)/2 pair dots (8 lanes per pair, shuffle-reduced) and t;
+//   wave 0 solves the SPD nc x nc system by Gauss elimination in LDS;
+//   second pass over the class rows forms Yz = T_c^T z.
+//
+// Classes with nc == 0 write Yz = 0 (the caller's RB term stands alone);
+// nc > NCMAX (pathological skew) also write Yz = 0 and set skipped[g]=1
+// so the caller solves them on the torch path.
+
+#include "common.h"
+
+#define NCMAX 32
+#define CW 256    // db chunk staged per iteration
+
+__global__ __launch_bounds__(256)
+void woodbury_kernel(const float* __restrict__ T_all,  // [n, db]
+                     const float* __restrict__ LRt,    // [G, db]
+                     const int* __restrict__ starts,   // [G]
+                     const int* __restrict__ ncs,      // [G]
+                     float* __restrict__ Yz,           // [G, db]
+                     int* __restrict__ skipped,        // [G]
+                     int db) {
+    const int g = blockIdx.x;
+    const int tid = threadIdx.x;
+    const int nc = ncs[g];
+    float* yz = Yz + (size_t)g * db;
+    if (nc == 0 || nc > NCMAX) {
+        for (int d = tid; d < db; d += 256) yz[d] = 0.0f;
+        if (tid == 0) skipped[g] = (nc > NCMAX) ? 1 : 0;
+        return;
+    }
+    if (tid == 0) skipped[g] = 0;
+    const int r0 = starts[g];
+    const float* Tg = T_all + (size_t)r0 * db;
+    const float* lr = LRt + (size_t)g * db;
+
+    __shared__ float Ts[NCMAX][CW + 4];
+    __shared__ float Ls[CW];
+    __shared__ float S[NCMAX][NCMAX + 1];
+    __shared__ float tv[NCMAX];
+    __shared__ float zv[NCMAX];
+
+    const int npair = nc * (nc + 1) / 2;
+    // pair p -> (i, j), i <= j (row-major upper triangle walk)
+    // thread layout for the dot phase: 32 groups of 8 lanes
+    const int grp = tid >> 3;
+    const int sub = tid & 7;
+
+    // zero accumulators
+    for (int p = tid; p < NCMAX * (NCMAX + 1); p += 256)
+        ((float*)S)[p] = 0.0f;
+    if (tid < NCMAX) tv[tid] = 0.0f;
+    __syncthreads();
+
+    for (int c0 = 0; c0 < db; c0 += CW) {
+        const int cw = min(CW, db - c0);
+        // stage the class tile + LR chunk (coalesced per row)
+        for (int i = tid; i < nc * CW; i += 256) {
+            const int r = i / CW, c = i % CW;
+            Ts[r][c] = (c < cw) ? Tg[(size_t)r * db + c0 + c] : 0.0f;
+        }
+        for (int c = tid; c < CW; c += 256)
+            Ls[c] = (c < cw) ? lr[c0 + c] : 0.0f;
+        __syncthreads();
+
+        // pair dots: group handles pairs grp, grp+32, ...; 8 lanes split
+        // the chunk; plus nc "t" dots as pair ids npair..npair+nc-1
+        for (int p = grp; p < npair + nc; p += 32) {
+            float acc = 0.0f;
+            if (p < npair) {
+                // unrank upper-triangle pair
+                int i = 0, rem = p;
+                while (rem >= nc - i) { rem -= nc - i; ++i; }
+                const int j = i + rem;
+                for (int c = sub; c < CW; c += 8)
+                    acc += Ts[i][c] * Ts[j][c];
+#pragma unroll
+                for (int off = 4; off; off >>= 1)
+                    acc += __shfl_down(acc, off, 8);
+                if (sub == 0) S[i][j] += acc;   // one owning group per pair
+            } else {
+                const int i = p - npair;
+                for (int c = sub; c < CW; c += 8)
+                    acc += Ts[i][c] * Ls[c];
+#pragma unroll
+                for (int off = 4; off; off >>= 1)
+                    acc += __shfl_down(acc, off, 8);
+                if (sub == 0) tv[i] += acc;
+            }
+        }
+        __syncthreads();
+    }
+
+    // mirror + identity; then Gauss elimination (SPD, no pivoting) by
+    // wave 0: serial over k, lane-parallel over columns
+    if (tid < 64) {
+        for (int i = tid; i < nc; i += 64) {
+            S[i][i] += 1.0f;
+            for (int j = 0; j < i; ++j) S[i][j] = S[j][i];
+        }
+    }
+    __syncthreads();
+    if (tid < 64) {
+        const int lane = tid;
+        for (int k = 0; k < nc; ++k) {
+            const float inv = 1.0f / S[k][k];
+            for (int j = k + 1; j < nc; ++j) {
+                const float f = S[j][k] * inv;
+                for (int c = k + lane; c < nc; c += 64)
+                    S[j][c] -= f * S[k][c];
+                if (lane == 0) tv[j] -= f * tv[k];
+            }
+        }
+        // back substitution (lane 0 serial — nc^2/2 ops, negligible)
+        if (lane == 0) {
+            for (int i = nc - 1; i >= 0; --i) {
+                float s = tv[i];
+                for (int j = i + 1; j < nc; ++j) s -= S[i][j] * zv[j];
+                zv[i] = s / S[i][i];
+            }
+        }
+    }
+    __syncthreads();
+
+    // Yz = T_c^T z  (coalesced second pass over the class rows)
+    for (int d = tid; d < db; d += 256) {
+        float acc = 0.0f;
+        for (int i = 0; i < nc; ++i)
+            acc += zv[i] * Tg[(size_t)i * db + d];
+        yz[d] = acc;
+    }
+}
+
+KS_EXPORT int ks_woodbury_f32(const float* T_all, const float* LRt,
+                              const int* starts, const int* ncs, float* Yz,
+                              int* skipped, int G, int db,
+                              hipStream_t stream) {
+    hipLaunchKernelGGL(woodbury_kernel, dim3(G), dim3(256), 0, stream,
+                       T_all, LRt, starts, ncs, Yz, skipped, db);
+    return (int)hipGetLastError();
+}

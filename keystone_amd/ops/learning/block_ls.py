@@ -542,6 +542,16 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         # block per epoch at K=1000 (measured, KEYSTONE_SOLVER_DEBUG).
         groups = own_t = wc_t = w_sqrt_rows = None
         wcs = {}
+        # GPU fp32: the per-class Woodbury inner solve (gather + S bmm +
+        # batched solve + reduce, ~0.45 ms/block at K=1000 in torch) runs
+        # as ONE hand kernel per block (hip/src/woodbury.hip); classes
+        # with nc > 32 fall back to the torch group machinery below.
+        use_wbk = False
+        wbk_starts = wbk_ncs = None
+        if own and Y.is_cuda and sdt == torch.float32 \
+                and not os.environ.get("KEYSTONE_NO_WBK"):
+            from ... import hip as hip_ops
+            use_wbk = hip_ops.gpu_ready()
         if own:
             dev_ = Y.device
             own_t = torch.tensor(own, device=dev_)
@@ -559,9 +569,17 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
             # exact.  Skipped when padding would inflate flops > 4x
             # (pathological class skew) — those classes stay in the
             # per-n_c group loop below.
+            if use_wbk:
+                from ... import hip as hip_ops
+                wbk_starts = torch.tensor(
+                    [cls_slice[c][0] for c in own], dtype=torch.int32,
+                    device=dev_)
+                wbk_ncs = torch.tensor(
+                    [cls_slice[c][1] - cls_slice[c][0] for c in own],
+                    dtype=torch.int32, device=dev_)
             pad = None
             pad_cls: set = set()
-            if uniform:
+            if uniform and not use_wbk:
                 db0_ = db0
                 wood = [(c, cls_slice[c][1] - cls_slice[c][0])
                         for c in own
@@ -593,6 +611,8 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                 if c in pad_cls:
                     continue
                 s_c, e_c = cls_slice[c]
+                if use_wbk and 0 <= (e_c - s_c) <= 32:
+                    continue  # handled by the woodbury kernel
                 by_nc.setdefault(e_c - s_c, []).append(c)
             groups = []
             for n_c, cls_list in by_nc.items():
@@ -694,6 +714,18 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                     t0 = tick("wb_ubuild", t0)
                     T_all = U_all @ Linv.T                       # [n, db]
                     t0 = tick("wb_gemm", t0)
+                    if use_wbk:
+                        from ... import hip as hip_ops
+                        LRt = LR.T.contiguous()                  # [G, db]
+                        Yz_all, _sk = hip_ops.woodbury_yz(
+                            T_all, LRt, wbk_starts, wbk_ncs)
+                        t0 = tick("wb_kernel", t0)
+                        corr_all = Yz_all @ Linv                 # [G, db]
+                        # nc == 0 / skipped rows have Yz == 0 -> Wb = RB;
+                        # nc > 32 columns are overwritten by the torch
+                        # group machinery below
+                        Wb[:, own_t] = (RB - corr_all.T).to(dtype)
+                        t0 = tick("wb_corr", t0)
                     if pad is not None:
                         (ri_p, mask_p, js_w, cls_idx_w, eye_w,
                          Gw, ncmax) = pad

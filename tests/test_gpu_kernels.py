@@ -442,3 +442,28 @@ def test_sift_large_radius_parity(hip):
     Dc = s4.apply_batch(imgs.cpu())
     assert Dg.shape == Dc.shape
     assert float((Dg.cpu() - Dc).abs().mean()) < 2e-3
+
+
+def test_woodbury_kernel_matches_torch_path(hip, monkeypatch):
+    """Weighted solver: the fused per-class Woodbury kernel must produce
+    the same model as the torch pad/bmm path (and both must satisfy the
+    per-column weighted normal equations)."""
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    torch.manual_seed(31)
+    n, d, K = 500, 256, 17            # ragged nc, incl. some > 32
+    X = torch.randn(n, d, device="cuda")
+    y = torch.cat([torch.zeros(80), torch.arange(n - 80) % (K - 1) + 1]).long()
+    Y = torch.full((n, K), -1.0, device="cuda")
+    Y[torch.arange(n), y.cuda()] = 1.0
+    kw = dict(block_size=64, num_iter=5, lam=1e-3, mixture_weight=0.5)
+    monkeypatch.delenv("KEYSTONE_NO_WBK", raising=False)
+    mk = BlockWeightedLeastSquaresEstimator(**kw).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    monkeypatch.setenv("KEYSTONE_NO_WBK", "1")
+    mt = BlockWeightedLeastSquaresEstimator(**kw).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    Wk = torch.cat(mk.blocks, dim=0)
+    Wt = torch.cat(mt.blocks, dim=0)
+    err = (Wk - Wt).abs().max().item()
+    assert err < 1e-4, err
