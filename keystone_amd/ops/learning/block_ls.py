@@ -163,7 +163,7 @@ class LazyBlocks(Dataset):
         return self.raw.items()
 
 
-def _chol_linv(A: torch.Tensor, base: int = 512) -> torch.Tensor:
+def _chol_linv(A: torch.Tensor, base: int = 1024) -> torch.Tensor:
     """Batched L^{-1} of the Cholesky factor of SPD ``A`` WITHOUT ever
     forming L: Schur recursion
         chol([[A11,.],[A21,A22]]) = [[L11,0],[L21,L22]],
@@ -191,7 +191,7 @@ def _chol_linv(A: torch.Tensor, base: int = 512) -> torch.Tensor:
     return out
 
 
-def _blocked_trtri(L: torch.Tensor, base: int = 512) -> torch.Tensor:
+def _blocked_trtri(L: torch.Tensor, base: int = 1024) -> torch.Tensor:
     """Batched lower-triangular inverse by 2x2 block recursion:
     ``inv([[L11,0],[L21,L22]]) = [[I11,0],[-I22 L21 I11, I22]]`` — shifts
     ~3/4 of the work from rocSOLVER TRSM (59 TF measured) onto hipBLASLt
