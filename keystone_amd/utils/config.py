@@ -18,7 +18,18 @@ Per-op debug / dispatch switches (read at the call site, not here):
   KEYSTONE_GMM_GRAPHS=1    hipGraph-capture the eager E-step (see gmm.py)
   KEYSTONE_SIFT_FUSED=1    fused SIFT pool kernel (measured slower; opt-in)
   KEYSTONE_FUSED_GEMM_COS=1  fully-fused MFMA gemm+cos (hybrid is faster)
+  KEYSTONE_FV_F32STATS=1   exact-f32 stats pass in fv_mfma (default: f16
+                           inputs / f32 accumulate on mfma_32x32x16_f16)
+  KEYSTONE_FV_SPLIT=1      K-split 2-workgroups-per-image fv_mfma variant
+                           (measured slower at K=256; kept as the A/B)
+  KEYSTONE_NO_WBK=1        disable the fused per-class Woodbury kernel
+                           (torch padded-bmm path instead)
+  KEYSTONE_NO_PROFILE_CACHE=1  AutoCacheRule falls back to the
+                           consumer-count heuristic (no sampled profiling)
+  KEYSTONE_STATE_CACHE     fitted-estimator reuse directory (keys include
+                           a fit-data fingerprint)
   KEYSTONE_GMM_DEBUG=1     per-phase timers in GMM/k-means fits
+  KEYSTONE_SIFT_DEBUG=1    per-phase timers in SIFT extraction
   KEYSTONE_SOLVER_DEBUG=1  per-phase timers in the weighted block solver
   KEYSTONE_BENCH_STAGES=1  per-stage timings from bench.py (stderr JSON)
   KEYSTONE_NO_GRAPHS=1     disable hipGraph capture everywhere
