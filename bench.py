@@ -199,6 +199,12 @@ def main():
                 "gmm_pca_sample": conf.n_sample,
                 "solver_block_size": args.block_size,
                 "feature_dim": info.get("feature_dim"),
+                # FV stats-pass inner detail, disclosed: gamma/x operands
+                # round to f16, accumulation stays f32
+                # (KEYSTONE_FV_F32STATS=1 for the exact-f32 pass; held-out
+                # parity measured unchanged — scripts/parity_experiment.py)
+                "fv_stats": "f16in_f32acc"
+                if not os.environ.get("KEYSTONE_FV_F32STATS") else "f32",
                 "fit_time_s_per_step": ms_per_step / 1000.0,
                 "serve_images_per_sec": serve_images_per_sec,
                 "serve_with_input_images_per_sec": serve_input_ips,

@@ -45,8 +45,14 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     float* raw = smem;                       // [RH][RW]
     float* hb = raw + RH * RW;               // [RH][HW]
     float* vb = hb + RH * HW;                // [OTH+2][HW]
+    float* gks = vb + (OTH + 2) * HW;        // [2r+1] staged taps
 
     const float* im = img + (size_t)b * H * W;
+
+    // stage the Gaussian taps ONCE: gk[t] inside the blur loops was a
+    // per-lane GLOBAL load per tap per output element (13+ loads each) —
+    // latency the 1-wave-deep loops cannot hide
+    for (int t = tid; t <= 2 * r; t += 256) gks[t] = gk[t];
 
     // ---- load raw tile + halo (replicate boundary)
     for (int i = tid; i < RH * RW; i += 256) {
@@ -63,7 +69,7 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < RH * HW; i += 256) {
         int ry = i / HW, cx = i % HW;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * raw[ry * RW + cx + t];
+        for (int t = 0; t <= 2 * r; ++t) s += gks[t] * raw[ry * RW + cx + t];
         hb[ry * HW + cx] = s;
     }
     __syncthreads();
@@ -72,7 +78,7 @@ void sift_omap_kernel(const float* __restrict__ img,  // [B, H, W]
     for (int i = tid; i < (OTH + 2) * HW; i += 256) {
         int vy = i / HW, cx = i % HW;
         float s = 0.0f;
-        for (int t = 0; t <= 2 * r; ++t) s += gk[t] * hb[(vy + t) * HW + cx];
+        for (int t = 0; t <= 2 * r; ++t) s += gks[t] * hb[(vy + t) * HW + cx];
         vb[vy * HW + cx] = s;
     }
     __syncthreads();
@@ -112,7 +118,8 @@ KS_EXPORT int ks_sift_omap_f32(const float* img, float* omap,
     const int RH = OTH + 2 + 2 * radius;
     const int RW = OTW + 2 + 2 * radius;
     const int HW = OTW + 2;
-    size_t lds = (size_t)(RH * RW + RH * HW + (OTH + 2) * HW) * 4;
+    size_t lds = (size_t)(RH * RW + RH * HW + (OTH + 2) * HW
+                          + 2 * radius + 1) * 4;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     dim3 grid(ks_ceil_div(W, OTW), ks_ceil_div(H, OTH), B);
     hipLaunchKernelGGL(sift_omap_kernel, grid, dim3(256), lds, stream, img,
