@@ -142,7 +142,8 @@ def test_sift_oracle_single_descriptor():
                         continue
                     raw[j, i, b0] += m * (1 - frac) * wy * wx
                     raw[j, i, (b0 + 1) % N_ORI] += m * frac * wy * wx
-    v = raw.reshape(-1)
+    # the extractor's element order is (o, jx, jy) — see sift.py
+    v = raw.permute(2, 1, 0).reshape(-1)
     v = v / v.norm().clamp_min(1e-10)
     v = v.clamp_max(0.2)
     v = v / v.norm().clamp_min(1e-10)
@@ -252,7 +253,7 @@ def test_sift_merged_pooling_matches_sampling_matrix():
         Mh, ny = m._sampling_matrix(H, bs, 4, imgs.device, dt)
         A1 = (omap.reshape(B * NO * H, W) @ Mw).reshape(B * NO, H, -1)
         A2 = torch.matmul(Mh.T, A1)
-        Dref = A2.reshape(B, NO, NS, ny, NS, nx).permute(0, 3, 5, 2, 4, 1) \
+        Dref = A2.reshape(B, NO, NS, ny, NS, nx).permute(0, 3, 5, 1, 4, 2) \
             .reshape(B, ny * nx, NS * NS * NO).contiguous()
         Mwm, nxm, Qx, syx, sjx = m._merged_pool_matrix(W, bs, 4,
                                                        imgs.device, dt)
@@ -262,8 +263,9 @@ def test_sift_merged_pooling_matches_sampling_matrix():
         A1q = (omap.reshape(B * NO * H, W) @ Mwm).reshape(B * NO, H, Qx)
         P = torch.matmul(Mhm.T, A1q).reshape(B, NO, Qy, Qx)
         s0, s1, s2, s3 = P.stride()
-        Dfast = P.as_strided((B, ny, nx, NS, NS, NO),
-                             (s0, syy * s2, syx * s3, sjy * s2, sjx * s3, s1)) \
+        Dfast = P.as_strided((B, ny, nx, NO, NS, NS),
+                             (s0, syy * s2, syx * s3, s1,
+                              sjx * s3, sjy * s2)) \
             .reshape(B, ny * nx, NS * NS * NO)
         assert (Dref - Dfast).abs().max().item() < 1e-5
 
