@@ -33,6 +33,17 @@ def main():
     tm.add_argument("--num-features", type=int, default=8192)
     tm.add_argument("--epochs", type=int, default=3)
 
+    cf = sub.add_parser("cifar")
+    cf.add_argument("--n", type=int, default=2000)
+    cf.add_argument("--num-filters", type=int, default=256)
+    cf.add_argument("--data-bin", help="CIFAR-10 binary file")
+
+    vc = sub.add_parser("voc")
+    vc.add_argument("--images", type=int, default=128)
+    vc.add_argument("--image-size", type=int, default=128)
+    vc.add_argument("--num-classes", type=int, default=20)
+    vc.add_argument("--data-dir", help="VOC-style dir with index.json")
+
     im = sub.add_parser("imagenet-fv")
     im.add_argument("--images", type=int, default=128)
     im.add_argument("--image-size", type=int, default=256)
@@ -100,6 +111,56 @@ def main():
         m = MulticlassClassifierEvaluator(147).evaluate(pipe.apply(train), labels)
         if rt.rank == 0:
             print(f"train {m.summary()}  ({time.time()-t0:.1f}s)")
+
+    elif args.app == "cifar":
+        import torch
+        from keystone_amd.evaluation import MulticlassClassifierEvaluator
+        from keystone_amd.loaders import CifarLoader
+        from keystone_amd.models import CifarConfig, fit_cifar_patches
+        import keystone_amd as ka
+        if args.data_bin:
+            ld = CifarLoader(args.data_bin)
+            imgs, labels = ld.data, ld.labels
+        else:
+            g = torch.Generator().manual_seed(1)
+            X = torch.rand(args.n, 3, 32, 32, generator=g)
+            y = torch.randint(0, 10, (args.n,), generator=g)
+            # class-dependent mean shift makes labels learnable
+            X = (X + 0.25 * y.view(-1, 1, 1, 1) / 10).clamp(0, 1)
+            imgs, labels = ka.from_rows(X.to(rt.device)), \
+                ka.from_rows(y.to(rt.device))
+        pipe = fit_cifar_patches(imgs, labels, CifarConfig(
+            num_filters=args.num_filters))
+        m = MulticlassClassifierEvaluator(10).evaluate(pipe.apply(imgs),
+                                                       labels)
+        if rt.rank == 0:
+            print(f"train {m.summary()}  ({time.time()-t0:.1f}s)")
+
+    elif args.app == "voc":
+        import torch
+        from keystone_amd.evaluation import MeanAveragePrecisionEvaluator
+        from keystone_amd.loaders import VOCLoader, synthetic_images
+        from keystone_amd.models import VOCConfig, fit_voc
+        import keystone_amd as ka
+        if args.data_dir:
+            imgs, ml = VOCLoader(args.data_dir, args.num_classes)
+        else:
+            imgs, y = synthetic_images(args.images, h=args.image_size,
+                                       w=args.image_size,
+                                       num_classes=args.num_classes,
+                                       class_structure=True)
+            # single-positive multi-label matrix from the int labels
+            import torch as _t
+            M = _t.zeros(y.data.shape[0], args.num_classes,
+                         device=y.data.device)
+            M[_t.arange(y.data.shape[0]), y.data] = 1.0
+            ml = ka.from_rows(M)
+        pipe = fit_voc(imgs, ml, VOCConfig(num_classes=args.num_classes))
+        scores = pipe.apply(imgs)
+        ev = MeanAveragePrecisionEvaluator(args.num_classes).evaluate(
+            scores, ml)
+        if rt.rank == 0:
+            print(f"train mAP {ev['mAP']:.4f}  ({time.time()-t0:.1f}s)")
 
     elif args.app == "imagenet-fv":
         from keystone_amd.evaluation import MulticlassClassifierEvaluator

@@ -1,6 +1,7 @@
 """End-to-end pipeline tests for the five driver configs (small CPU sizes).
 ref: SURVEY.md §4 — the driver configs are the integration benchmarks."""
 
+import os
 import pytest
 import torch
 
@@ -148,3 +149,23 @@ def test_imagenet_fv_generalizes_on_structured_classes():
     # variant); >= 0.45 at chance 0.25 over 48 samples is p << 1e-3 —
     # the FV stack genuinely generalizes (measured ~0.58)
     assert acc > 0.45, f"test accuracy {acc} vs chance {1/K}"
+
+
+def test_cli_cifar_and_voc(tmp_path):
+    """The two remaining reference example apps run end-to-end from the
+    CLI (ref L6: RandomPatchCifar, VOCSIFTFisher)."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "bin/run_pipeline.py", "cifar", "--n", "300",
+         "--num-filters", "32"], cwd=root, capture_output=True, text=True,
+        timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "train accuracy" in out.stdout
+    out = subprocess.run(
+        [sys.executable, "bin/run_pipeline.py", "voc", "--images", "16",
+         "--image-size", "64"], cwd=root, capture_output=True, text=True,
+        timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "train mAP" in out.stdout
