@@ -113,3 +113,23 @@ def test_string_label_indicators_roundtrip(tmp_path):
     pipe.save(path)
     loaded = ka.Pipeline.load(path)
     assert torch.equal(pipe.apply("b").get(), loaded.apply("b").get())
+
+
+def test_weighted_block_mapper_with_intercept_roundtrips(tmp_path):
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    torch.manual_seed(8)
+    n, d, K = 60, 10, 3
+    X = torch.randn(n, d, dtype=torch.float64) + 0.3
+    y = torch.arange(n) % K
+    Y = torch.full((n, K), -1.0, dtype=torch.float64)
+    Y[torch.arange(n), y] = 1.0
+    m = BlockWeightedLeastSquaresEstimator(
+        block_size=5, num_iter=20, lam=1e-2, mixture_weight=0.5,
+        fit_intercept=True).fit(ka.from_rows(X), ka.from_rows(Y))
+    pipe = ka.Identity().and_then(m)
+    path = str(tmp_path / "bwls")
+    pipe.save(path)
+    loaded = ka.Pipeline.load(path)
+    a = pipe.apply(X).get()
+    b = loaded.apply(X).get()
+    assert torch.allclose(a, b, atol=1e-12)
