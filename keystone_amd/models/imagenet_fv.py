@@ -112,10 +112,17 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
 
     import os
     tm = info if os.environ.get("KEYSTONE_BENCH_STAGES") else None
+
+    def _sync():
+        # stage walls are only honest if the queue drains at each boundary
+        if tm is not None and train_imgs.data.is_cuda:
+            torch.cuda.synchronize()
+
     sift = GrayScaler().and_then(
         SIFTExtractor(conf.sift_step, conf.sift_bin, conf.sift_scales))
     sift_chain, sift_feats = _fit_branch(sift, train_imgs, conf, conf.seed,
                                          tm, "sift_")
+    _sync()
     info["t_sift_branch"] = time.perf_counter() - t0
 
     branches = [sift_chain]
@@ -133,6 +140,7 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
                                            conf.seed + 1, tm, "lcs_")
         branches.append(lcs_chain)
         feats = torch.cat([feats, lcs_feats.data], dim=1)
+        _sync()
         info["t_lcs_branch"] = time.perf_counter() - t1
 
     t2 = time.perf_counter()
@@ -142,6 +150,7 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
     solver = BlockWeightedLeastSquaresEstimator(
         conf.block_size, conf.solver_epochs, conf.lam, conf.mixture_weight)
     model = solver.fit(TensorDataset(feats, rt), TensorDataset(Y, rt))
+    _sync()
     info["t_solver"] = time.perf_counter() - t2
     info["feature_dim"] = int(feats.shape[1])
     info["model"] = model
