@@ -120,6 +120,14 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_pool_lastdim_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i,
+                                            c_i, c_i, ctypes.c_long, c_p]
+        lib.ks_pool_lastdim_f32.restype = c_i
+        lib.ks_pool_middim_f32.argtypes = [c_p, c_p, c_p] + [c_i] * 7 + [c_p]
+        lib.ks_pool_middim_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_sift_omap_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i, c_i,
                                          c_p]
         lib.ks_sift_omap_f32.restype = c_i
@@ -391,6 +399,39 @@ def gmm_mstep_(partials: torch.Tensor, pre,
 
 
 WOODBURY_NCMAX = 32
+
+
+def pool_lastdim(x: torch.Tensor, tap: torch.Tensor, c0: int, g: int,
+                 Q: int) -> torch.Tensor:
+    """Triangular lattice pooling of the LAST dim: [R, W] -> [R, Q]
+    (pool1d.hip — evaluates the 2*bin-1 band directly instead of the
+    dense [W, Q] GEMM whose useful flops are ~6%)."""
+    lib = _load()
+    x = _prep(x, "pool_lastdim.x")
+    tap = _prep(tap, "pool_lastdim.tap")
+    R, W = x.shape
+    out = torch.empty(R, Q, dtype=torch.float32, device=x.device)
+    _check(lib.ks_pool_lastdim_f32(
+        ctypes.c_void_p(x.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_void_p(tap.data_ptr()), tap.numel(), c0, g, W, Q,
+        ctypes.c_long(R), _stream()), "ks_pool_lastdim_f32")
+    return out
+
+
+def pool_middim(x: torch.Tensor, tap: torch.Tensor, c0: int, g: int,
+                Qy: int) -> torch.Tensor:
+    """Triangular lattice pooling of the MIDDLE dim with transpose:
+    [B, H, C] -> [B, C, Qy] (the descriptor-gather layout)."""
+    lib = _load()
+    x = _prep(x, "pool_middim.x")
+    tap = _prep(tap, "pool_middim.tap")
+    B, H, C = x.shape
+    out = torch.empty(B, C, Qy, dtype=torch.float32, device=x.device)
+    _check(lib.ks_pool_middim_f32(
+        ctypes.c_void_p(x.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_void_p(tap.data_ptr()), tap.numel(), c0, g, H, C, Qy, B,
+        _stream()), "ks_pool_middim_f32")
+    return out
 
 
 def woodbury_yz(T_all: torch.Tensor, LRt: torch.Tensor,

@@ -244,18 +244,33 @@ class SIFTExtractor(Transformer):
                     A1 = hip_ops.sift_omap_pooled(
                         imgs.contiguous(), sigma, bin_size, g, Qx) \
                         .reshape(B * N_ORI, H, Qx)
+            # banded lattice pooling kernel for the W direction when the
+            # lattice is dense (small gcd -> Q large): the dense [W, Qx]
+            # GEMM spends ~17x the useful flops there (measured: kernel
+            # 7.2 vs GEMM 10.4 ms at Q=124; a wash at Q=42, and the
+            # middim variant never beat the strided bmm -> GEMM kept)
+            use_pool_kernel = False
+            if imgs.is_cuda and dt == torch.float32 and Qx >= 64:
+                from ... import hip as hip_ops
+                use_pool_kernel = hip_ops.gpu_ready()
             if A1 is None:
                 omap = self._orientation_maps(imgs, sigma)
                 t0 = tick("omap", t0)
-                A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
-                    .reshape(B * N_ORI, H, Qx)
+                if use_pool_kernel:
+                    tap = _triangle_kernel1d(bin_size, imgs.device, dt)
+                    A1 = hip_ops.pool_lastdim(
+                        omap.reshape(B * N_ORI * H, W), tap,
+                        bin_size // 2, g, Qx).reshape(B * N_ORI, H, Qx)
+                else:
+                    A1 = (omap.reshape(B * N_ORI * H, W) @ Mw) \
+                        .reshape(B * N_ORI, H, Qx)
                 t0 = tick("a1_gemm", t0)
             # H-pooling as a batched-A @ broadcast-B bmm ([G, m, K] @
             # [K, n] — hipBLASLt strided-batched, the fast form; both the
             # broadcast-A form Mh.T @ [B*8, H, Qx] and a transpose+flat
             # rewrite measured 13 ms on rocBLAS fallbacks / transpose
-            # copies).  transpose(1, 2) is stride flags only, no copy.
-            P = torch.matmul(A1.transpose(1, 2), Mh)   # [B*8, Qx, Qy]
+            # copies).
+            P = torch.matmul(A1.transpose(1, 2), Mh)  # [B*8, Qx, Qy]
             t0 = tick("p_gemm", t0)
             # descriptor element order is (o, jx, jy) — OUR convention
             # (PCA/GMM/FV are order-agnostic; fit and serve share it):
