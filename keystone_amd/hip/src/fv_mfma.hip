@@ -52,10 +52,12 @@ typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 // accumulates stats for only HALF the components.  Halved sacc (80
 // AGPRs) + trimmed LDS (78.5 KB at K=256) let 2 workgroups co-reside
 // per CU, aiming to fill the 69% non-MFMA time seen at occupancy 1.
-// MEASURED: 48.6 ms vs 33.9 for the joint-span schedule at K=256/B=8192
-// — the duplicated logits MFMA (+93% issue work) and the 5-barrier
-// schedule cost more than the co-residency buys.  Kept opt-in as the
-// documented A/B; the joint-span f16 kernel is the production path.
+// MEASURED (two generations): Gs-based split 48.6 ms; ONLINE-softmax
+// split (no Gs panel, row reductions straight from the accumulator
+// fragments, 4 barriers) 45.2 ms — vs 33.9 for the joint-span schedule
+// at K=256/B=8192.  The duplicated logits MFMA (+93% issue work) always
+// outweighs the co-residency win.  Kept opt-in as the documented A/B;
+// the joint-span f16 kernel is the production path.
 template <int KT, bool STATS, bool F16S, bool SPLIT>
 __global__ __launch_bounds__(256, SPLIT ? 2 : 1)
 void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
@@ -80,18 +82,24 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     // 32 banks distinct.  The previous KT+1 stride gave bank row+sub:
     // ~3-way conflicts, 2.7e8 counted (profiles/pmc_summary.csv).
     constexpr int GST = KT + 8;
-    // f32 words before the f16 region (16-B aligned for b128 loads)
+    // f32 words before the f16 region (16-B aligned for b128 loads).
+    // SPLIT uses ONLINE softmax straight from the accumulator fragments:
+    // no Gs buffer at all, just two tiny [32 rows][5] reduction panels.
     constexpr int NBUF = F16S ? 1 : 2;          // Xs f32 buffers
+    constexpr int GSW = SPLIT ? 0 : CHUNK * GST;
+    constexpr int RSW = SPLIT ? 2 * 32 * 5 : 0;
     constexpr int F32W =
-        ((NBUF * CHUNK * XSTRIDE + CHUNK * GST + KT + 1) + 3) & ~3;
+        ((NBUF * CHUNK * XSTRIDE + GSW + KT + 1 + RSW) + 3) & ~3;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     // gamma needs NO double buffer: stats(c-1) reads it during the joint
     // MFMA span, and logits(c) only overwrite it in the write phase after
     // the intervening barrier.
     float* Xs0 = (float*)smem;                  // [NBUF][CHUNK][XSTRIDE]
-    float* Gs0 = Xs0 + NBUF * CHUNK * XSTRIDE;  // [CHUNK][GST]
-    float* S0s = Gs0 + CHUNK * GST;             // [KT] broadcast (+ll)
+    float* Gs0 = Xs0 + NBUF * CHUNK * XSTRIDE;  // [CHUNK][GST] (!SPLIT)
+    float* S0s = Gs0 + GSW;                     // [KT] broadcast (+ll)
+    float* smM = S0s + KT + 1;                  // [32][5] row-max panel
+    float* smS = smM + 32 * 5;                  // [32][5] row-sum panel
     // f16 region (F16S only): transposed panels for the f16 stats MFMA
     constexpr int NXH = SPLIT ? 1 : 2;          // XhT buffers
     _Float16* XhT0 = (_Float16*)(smem + (size_t)F32W * 4);  // [NXH][XCP][SH]
@@ -309,20 +317,69 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     };
 
     if (SPLIT) {
-        // 5-barrier schedule: [logits] [Gs + Xs(c+1)] [softmax -> GhT]
-        // [stats(c)] [XhT(c+1)].  Spans are no longer joint — the
-        // co-resident second workgroup fills the gaps instead.
+        // ONLINE-softmax schedule (4 barriers, no Gs round-trip):
+        //   [logits span] -> row-max shuffle/LDS reduce -> B1 ->
+        //   [exp + row-sum reduce; Xs(c+1) f32 write] -> B2 ->
+        //   [gamma -> own GhT tile] -> B3 -> [stats(c)] -> B4 ->
+        //   [XhT(c+1) f16 write]  (next chunk's B3 covers visibility).
+        // Per (lane, reg) the C-fragment row is unique:
+        // row = (r & 3) + 8 * (r >> 2) + 4 * lk; lanes sharing lk are a
+        // contiguous 32-lane group, so __shfl_xor(..., 32) reduces rows
+        // across the in-wave columns.  At K=256 / LT=2 tile t == half
+        // owns the WG's gamma half exactly.
+        float ll_acc = 0.0f;
         for (int c = 0; c < nc; ++c) {
             const bool more = c + 1 < nc;
+            const int rows_c = min(CHUNK, n - (int)chunk_start(c));
             const int rows1 = more
                 ? min(CHUNK, n - (int)chunk_start(c + 1)) : 0;
             if (more) stage_load((int)chunk_start(c + 1), rows1);
             f32x16 lacc[LT];
             logits_span(lacc);
-            __syncthreads();
-            gs_write(lacc);
+            float red[16];
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                red[r] = fmaxf(lacc[0][r], lacc[1][r]);
+#pragma unroll
+                for (int off = 16; off; off >>= 1)
+                    red[r] = fmaxf(red[r], __shfl_xor(red[r], off, 32));
+            }
+            if (l31 == 0) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                    smM[row * 5 + wave] = red[r];
+                }
+            }
+            __syncthreads();   // B1
+            float mrow[16];
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                mrow[r] = fmaxf(fmaxf(smM[row * 5 + 0], smM[row * 5 + 1]),
+                                fmaxf(smM[row * 5 + 2], smM[row * 5 + 3]));
+            }
+#pragma unroll
+            for (int t = 0; t < LT; ++t)
+#pragma unroll
+                for (int r = 0; r < 16; ++r)
+                    lacc[t][r] = __expf(lacc[t][r] - mrow[r]);
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                red[r] = lacc[0][r] + lacc[1][r];
+#pragma unroll
+                for (int off = 16; off; off >>= 1)
+                    red[r] += __shfl_xor(red[r], off, 32);
+            }
+            if (l31 == 0) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                    smS[row * 5 + wave] = red[r];
+                }
+            }
             if (more) {
-                // f32 staging only (XhT(c) still feeds stats(c))
+                // Xs(c+1): logits(c) consumed Xs before B1
 #pragma unroll
                 for (int q = 0; q < 8; ++q) {
                     Xs0[(sr0 + q) * XSTRIDE + sd] = xreg[q] * xreg[q];
@@ -331,11 +388,26 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                 if (tid < CHUNK)
                     Xs0[tid * XSTRIDE + 2 * DT] = (tid < rows1) ? 1.0f : 0.0f;
             }
-            __syncthreads();
-            softmax_chunk(c);
-            __syncthreads();
+            __syncthreads();   // B2
+            // gamma for the OWN half tile (t == blockIdx.y) -> GhT
+            {
+                const int t = (int)blockIdx.y;
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
+                    const float srow = smS[row * 5 + 0] + smS[row * 5 + 1]
+                                     + smS[row * 5 + 2] + smS[row * 5 + 3];
+                    const int col_local = wave * 32 + l31;
+                    GhT[(size_t)col_local * SH + row] =
+                        (_Float16)(lacc[t][r] / srow);
+                    if (STATS && wave == 0 && l31 == 0 && t == 0
+                            && row < rows_c)
+                        ll_acc += mrow[r] + __logf(srow);
+                }
+            }
+            __syncthreads();   // B3: GhT visible
             stats_f16(c);
-            __syncthreads();
+            __syncthreads();   // B4: XhT(c) consumed
             if (more) {
                 f16x8 hx, hx2;
 #pragma unroll
@@ -352,9 +424,10 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
                         ones[q] = (_Float16)((sr0 + q < rows1) ? 1.0f : 0.0f);
                     *(f16x8*)&XhT0[(size_t)(2 * DT) * SH + sr0] = ones;
                 }
-                __syncthreads();
             }
         }
+        if (STATS && wave == 0 && l31 == 0 && blockIdx.y == 0 && kh0 == 0)
+            atomicAdd(&S0s[KT], ll_acc);
     } else
     for (int c = 0; c < nc; ++c) {
         const int cur = c & 1;
@@ -540,10 +613,13 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
     }
 }
 
-// mode: 0 = f32 stats, 1 = f16 stats, 2 = f16 stats + K-split (2 WG/img)
+// mode: 0 = f32 stats, 1 = f16 stats, 2 = f16 stats + K-split (2 WG/img,
+// online softmax: no Gs panel, two [32][5] reduction panels instead)
 static size_t fv_lds_bytes(int K, int mode) {
     const int nbuf = mode ? 1 : 2;
-    size_t f32w = ((size_t)(nbuf * CHUNK * XSTRIDE + CHUNK * (K + 8) + K + 1)
+    const int gsw = (mode == 2) ? 0 : CHUNK * (K + 8);
+    const int rsw = (mode == 2) ? 2 * 32 * 5 : 0;
+    size_t f32w = ((size_t)(nbuf * CHUNK * XSTRIDE + gsw + K + 1 + rsw)
                    + 3) & ~(size_t)3;
     size_t b = f32w * 4;
     const int nxh = (mode == 2) ? 1 : 2;
