@@ -46,8 +46,18 @@ def _solve_spd(gram: torch.Tensor, rhs: torch.Tensor,
             jitter = 1e-6 * float(G.diagonal().abs().mean())
             L = torch.linalg.cholesky(G + jitter * eye)
         except torch.linalg.LinAlgError:
-            G64 = gram.to(torch.float64) + lam * eye.to(torch.float64)
-            L64 = torch.linalg.cholesky(G64)
+            # the fp32 GRAM itself can be non-PSD when feature magnitudes
+            # make its accumulation error exceed lam (observed: CIFAR
+            # sum-pooled conv features, diag ~1e9 -> error ~1e2 >> lam);
+            # casting to fp64 cannot repair that, so jitter scaled to the
+            # diagonal is part of the fp64 fallback as well
+            eye64 = eye.to(torch.float64)
+            G64 = gram.to(torch.float64) + lam * eye64
+            try:
+                L64 = torch.linalg.cholesky(G64)
+            except torch.linalg.LinAlgError:
+                j64 = 1e-6 * float(G64.diagonal().abs().mean())
+                L64 = torch.linalg.cholesky(G64 + j64 * eye64)
             return torch.cholesky_solve(rhs.to(torch.float64), L64).to(dt)
     W = torch.cholesky_solve(rhs.to(sdt), L)
     return W.to(dt)
