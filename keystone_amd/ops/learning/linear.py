@@ -53,11 +53,22 @@ def _solve_spd(gram: torch.Tensor, rhs: torch.Tensor,
             # diagonal is part of the fp64 fallback as well
             eye64 = eye.to(torch.float64)
             G64 = gram.to(torch.float64) + lam * eye64
-            try:
-                L64 = torch.linalg.cholesky(G64)
-            except torch.linalg.LinAlgError:
-                j64 = 1e-6 * float(G64.diagonal().abs().mean())
-                L64 = torch.linalg.cholesky(G64 + j64 * eye64)
+            # jitter LADDER: the fp32 Gram's accumulation error scales
+            # with its diagonal (observed -1e2..-1e4 eigenvalues at diag
+            # ~3e8 on CIFAR sum-pooled conv features) — one fixed step is
+            # not enough when the user's lam is tiny relative to the
+            # feature scale
+            j64 = 1e-8 * float(G64.diagonal().abs().mean())
+            last = None
+            for _ in range(6):
+                try:
+                    L64 = torch.linalg.cholesky(G64 + j64 * eye64)
+                    break
+                except torch.linalg.LinAlgError as e:
+                    last = e
+                    j64 *= 32.0
+            else:
+                raise last
             return torch.cholesky_solve(rhs.to(torch.float64), L64).to(dt)
     W = torch.cholesky_solve(rhs.to(sdt), L)
     return W.to(dt)
