@@ -120,6 +120,11 @@ def _register_extra(lib: ctypes.CDLL) -> None:
     except AttributeError:
         pass
     try:
+        lib.ks_segsum_f32.argtypes = [c_p] * 6 + [c_i] * 3 + [c_p]
+        lib.ks_segsum_f32.restype = c_i
+    except AttributeError:
+        pass
+    try:
         lib.ks_pool_lastdim_f32.argtypes = [c_p, c_p, c_p, c_i, c_i, c_i,
                                             c_i, c_i, ctypes.c_long, c_p]
         lib.ks_pool_lastdim_f32.restype = c_i
@@ -399,6 +404,23 @@ def gmm_mstep_(partials: torch.Tensor, pre,
 
 
 WOODBURY_NCMAX = 32
+
+
+def segsum(X: torch.Tensor, T: torch.Tensor, starts: torch.Tensor,
+           ncs: torch.Tensor, cols: torch.Tensor) -> torch.Tensor:
+    """Contiguous segmented reduction for the weighted solver's per-class
+    rhs: out[g] = sum_{i in class g} X[i, :] * T[i, cols[g]]."""
+    lib = _load()
+    G = starts.numel()
+    n, db = X.shape
+    K = T.shape[1]
+    out = torch.empty(G, db, dtype=torch.float32, device=X.device)
+    _check(lib.ks_segsum_f32(
+        ctypes.c_void_p(X.data_ptr()), ctypes.c_void_p(T.data_ptr()),
+        ctypes.c_void_p(starts.data_ptr()), ctypes.c_void_p(ncs.data_ptr()),
+        ctypes.c_void_p(cols.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        G, db, K, _stream()), "ks_segsum_f32")
+    return out
 
 
 def pool_lastdim(x: torch.Tensor, tap: torch.Tensor, c0: int, g: int,

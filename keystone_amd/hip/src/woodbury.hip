@@ -151,3 +151,51 @@ KS_EXPORT int ks_woodbury_f32(const float* T_all, const float* LRt,
                        T_all, LRt, starts, ncs, Yz, skipped, db);
     return (int)hipGetLastError();
 }
+
+// Per-class rhs correction for the weighted solver: with rows sorted by
+// class, seg[g] = sum_{i in class g} X[i, :] * T[i, class_g] is a
+// contiguous segmented reduction.  The torch chain (T.gather -> X*v temp
+// -> zeros -> index_add_ -> index_select) paid three extra full passes
+// over X; this kernel reads the class rows once and the T column scalar
+// per row directly.
+__global__ __launch_bounds__(256)
+void segsum_kernel(const float* __restrict__ X,   // [n, db]
+                   const float* __restrict__ T,   // [n, K]
+                   const int* __restrict__ starts,  // [G]
+                   const int* __restrict__ ncs,     // [G]
+                   const int* __restrict__ cols,    // [G] class index
+                   float* __restrict__ out,         // [G, db]
+                   int db, int K) {
+    const int g = blockIdx.x;
+    const int tid = threadIdx.x;
+    const int nc = ncs[g];
+    const int r0 = starts[g];
+    const int cg = cols[g];
+    float* og = out + (size_t)g * db;
+    __shared__ float vs[4096 / 256 * 256];  // up to 4096 v values staged
+    // stage v (one scalar per row) in chunks if nc is large
+    for (int base = 0; base < nc; base += 4096) {
+        const int m = min(4096, nc - base);
+        for (int i = tid; i < m; i += 256)
+            vs[i] = T[(size_t)(r0 + base + i) * K + cg];
+        __syncthreads();
+        for (int d = tid; d < db; d += 256) {
+            float acc = (base == 0) ? 0.0f : og[d];
+            for (int i = 0; i < m; ++i)
+                acc += vs[i] * X[(size_t)(r0 + base + i) * db + d];
+            og[d] = acc;
+        }
+        __syncthreads();
+    }
+    if (nc == 0)
+        for (int d = tid; d < db; d += 256) og[d] = 0.0f;
+}
+
+KS_EXPORT int ks_segsum_f32(const float* X, const float* T,
+                            const int* starts, const int* ncs,
+                            const int* cols, float* out, int G, int db,
+                            int K, hipStream_t stream) {
+    hipLaunchKernelGGL(segsum_kernel, dim3(G), dim3(256), 0, stream,
+                       X, T, starts, ncs, cols, out, db, K);
+    return (int)hipGetLastError();
+}

@@ -577,6 +577,7 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                 wbk_ncs = torch.tensor(
                     [cls_slice[c][1] - cls_slice[c][0] for c in own],
                     dtype=torch.int32, device=dev_)
+                wbk_cols = torch.tensor(own, dtype=torch.int32, device=dev_)
             pad = None
             pad_cls: set = set()
             if uniform and not use_wbk:
@@ -694,15 +695,26 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                             L, torch.eye(db, dtype=sdt, device=Xb.device),
                             upper=False)
                     Xb64 = Xb.to(sdt)
-                    # rhs for ALL classes in two fused passes: the class-c
-                    # correction X_c^T T[rows_c, c] is a segmented
-                    # reduction over class-sorted rows (one index_add_).
-                    v = T.gather(1, cls.unsqueeze(1)).to(sdt)  # [n,1]
-                    seg = torch.zeros(K, db, dtype=sdt,
-                                      device=Xb.device)
-                    seg.index_add_(0, cls, Xb64 * v)
-                    rhs = base_w * xtr_all.index_select(1, own_t).to(sdt) \
-                        + seg.index_select(0, own_t).T * wc_t
+                    # rhs for ALL classes: the class-c correction
+                    # X_c^T T[rows_c, c] is a segmented reduction over
+                    # class-sorted rows — ONE segsum kernel on GPU
+                    # (reads the rows once; the torch chain paid a
+                    # gather + a full X*v temp + a scatter), index_add_
+                    # otherwise.
+                    if use_wbk:
+                        from ... import hip as hip_ops
+                        seg_own = hip_ops.segsum(Xb64, T, wbk_starts,
+                                                 wbk_ncs, wbk_cols)
+                        rhs = base_w * xtr_all.index_select(
+                            1, own_t).to(sdt) + seg_own.T * wc_t
+                    else:
+                        v = T.gather(1, cls.unsqueeze(1)).to(sdt)  # [n,1]
+                        seg = torch.zeros(K, db, dtype=sdt,
+                                          device=Xb.device)
+                        seg.index_add_(0, cls, Xb64 * v)
+                        rhs = base_w * xtr_all.index_select(
+                            1, own_t).to(sdt) \
+                            + seg.index_select(0, own_t).T * wc_t
                     LR = Linv @ rhs                          # L^{-1} rhs
                     RB = Linv.T @ LR                         # B^{-1} rhs
                     t0 = tick("rhs", t0)
