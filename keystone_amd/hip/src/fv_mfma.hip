@@ -269,53 +269,6 @@ void fv_mfma_kernel(const float* __restrict__ X,    // [B, n, D]
             }
         }
     };
-    auto gs_write = [&](const f32x16* lacc) {
-#pragma unroll
-        for (int t = 0; t < LT; ++t) {
-            const int ct = wave + 4 * t;
-            if (FULL || ct < KT32) {
-#pragma unroll
-                for (int r = 0; r < 16; ++r) {
-                    const int row = (r & 3) + 8 * (r >> 2) + 4 * lk;
-                    Gs0[row * GST + ct * 32 + l31] = lacc[t][r];
-                }
-            }
-        }
-    };
-    // softmax over full K; F16S writes the own-half gamma panel
-    auto softmax_chunk = [&](int c) {
-        const int row = tid >> 3;
-        const int sub = tid & 7;
-        float m = -1e30f;
-        for (int k = sub; k < KT; k += 8)
-            m = fmaxf(m, Gs0[row * GST + k]);
-#pragma unroll
-        for (int off = 4; off; off >>= 1)
-            m = fmaxf(m, __shfl_xor(m, off, 8));
-        float s = 0.0f;
-        for (int k = sub; k < KT; k += 8) {
-            float e = __expf(Gs0[row * GST + k] - m);
-            Gs0[row * GST + k] = e;
-            s += e;
-        }
-#pragma unroll
-        for (int off = 4; off; off >>= 1)
-            s += __shfl_xor(s, off, 8);
-        const float inv = 1.0f / s;
-        for (int k = sub; k < KT; k += 8) {
-            if (F16S) {
-                if (!SPLIT || ((unsigned)(k - kh0) < (unsigned)KH))
-                    GhT[(size_t)(k - kh0) * SH + row] =
-                        (_Float16)(Gs0[row * GST + k] * inv);
-            } else {
-                Gs0[row * GST + k] *= inv;
-            }
-        }
-        if (STATS && sub == 0 && (!SPLIT || kh0 == 0)
-                && row < n - (int)chunk_start(c))
-            atomicAdd(&S0s[KT], m + __logf(s));   // logsumexp sum
-    };
-
     if (SPLIT) {
         // ONLINE-softmax schedule (4 barriers, no Gs round-trip):
         //   [logits span] -> row-max shuffle/LDS reduce -> B1 ->
