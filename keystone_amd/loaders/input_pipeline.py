@@ -20,7 +20,8 @@ from typing import Callable, Iterable, Iterator, List, Optional, Sequence, Tuple
 
 import torch
 
-__all__ = ["decode_image_bytes", "encode_jpeg_batch", "ImageBatchPipeline"]
+__all__ = ["decode_image_bytes", "encode_jpeg_batch", "ImageBatchPipeline",
+           "ProcessImagePipeline"]
 
 
 def decode_image_bytes(data: bytes, color: bool = True) -> torch.Tensor:
@@ -135,3 +136,139 @@ class ImageBatchPipeline:
         finally:
             pool.shutdown(wait=False)
             stager.shutdown(wait=False)
+
+
+def _mp_decode_worker(worker_id, nworkers, blobs, batch, color, shape,
+                      shm_names, done_q, free_q):
+    """Decode worker process: owns batches worker_id, worker_id+N, ...;
+    writes uint8 frames into its two shared-memory slots."""
+    import numpy as np
+    from multiprocessing import shared_memory
+    shms = [shared_memory.SharedMemory(name=n) for n in shm_names]
+    try:
+        n = len(blobs)
+        nb = (n + batch - 1) // batch
+        for bi in range(worker_id, nb, nworkers):
+            slot = free_q.get()          # wait until one of OUR slots frees
+            lo, hi = bi * batch, min((bi + 1) * batch, n)
+            buf = np.ndarray((batch,) + shape, dtype=np.uint8,
+                             buffer=shms[slot].buf)
+            for j in range(lo, hi):
+                from PIL import Image
+                img = Image.open(io.BytesIO(blobs[j]))
+                img = img.convert("RGB" if color else "L")
+                a = np.asarray(img)
+                if color:
+                    a = a.transpose(2, 0, 1)
+                buf[j - lo] = a
+            done_q.put((bi, slot, hi - lo))
+    finally:
+        for s in shms:
+            s.close()
+
+
+class ProcessImagePipeline:
+    """Multi-PROCESS decode -> shared-memory uint8 ring -> pinned f32
+    staging -> async H2D.  The threaded pipeline tops out ~1.3k img/s on
+    256x256 JPEG (the PIL->numpy->stack tail is GIL-bound; more threads
+    measured SLOWER); worker processes sidestep the GIL and move frames
+    as zero-copy shared-memory uint8.
+
+    Requires uniform image shapes (JPEG corpora of one resolution — the
+    tar-shard serving case); ImageBatchPipeline handles the general case.
+    """
+
+    def __init__(self, blobs: Sequence[bytes], labels: Sequence[int],
+                 batch_size: int = 256, color: bool = True,
+                 device: Optional[torch.device] = None, workers: int = 8):
+        self.blobs = list(blobs)
+        self.labels = torch.as_tensor(labels, dtype=torch.long)
+        self.batch = batch_size
+        self.color = color
+        self.device = device
+        self.workers = max(1, workers)
+        t0 = decode_image_bytes(self.blobs[0], color)
+        self.shape = tuple(t0.shape)       # (3, H, W) or (H, W)
+
+    def __len__(self) -> int:
+        return (len(self.blobs) + self.batch - 1) // self.batch
+
+    def __iter__(self):
+        import multiprocessing as mp
+        import numpy as np
+        from multiprocessing import shared_memory
+        ctx = mp.get_context("spawn")
+        n = len(self.blobs)
+        nb = len(self)
+        nworkers = min(self.workers, nb) or 1
+        frame = int(np.prod(self.shape))
+        shms, procs, free_qs = [], [], []
+        done_q = ctx.Queue()
+        use_gpu = self.device is not None and self.device.type == "cuda"
+        copy_stream = torch.cuda.Stream(self.device) if use_gpu else None
+        pinned = torch.empty((self.batch,) + self.shape,
+                             dtype=torch.float32).pin_memory() \
+            if use_gpu else None
+        try:
+            for w in range(nworkers):
+                names = []
+                for _ in range(2):
+                    shm = shared_memory.SharedMemory(
+                        create=True, size=self.batch * frame)
+                    shms.append(shm)
+                    names.append(shm.name)
+                fq = ctx.Queue()
+                fq.put(0)
+                fq.put(1)
+                free_qs.append(fq)
+                # each worker only receives ITS batches' blobs
+                my = [self.blobs[j] if (j // self.batch) % nworkers == w
+                      else b"" for j in range(n)]
+                p = ctx.Process(target=_mp_decode_worker,
+                                args=(w, nworkers, my, self.batch,
+                                      self.color, self.shape, names,
+                                      done_q, fq))
+                p.daemon = True
+                p.start()
+                procs.append(p)
+            ready = {}
+            for _ in range(nb):
+                # batches must be yielded in order
+                want = _
+                while want not in ready:
+                    bi, slot, cnt = done_q.get()
+                    ready[bi] = (slot, cnt)
+                slot, cnt = ready.pop(want)
+                w = want % nworkers
+                shm = shms[w * 2 + slot]
+                arr = np.ndarray((self.batch,) + self.shape, dtype=np.uint8,
+                                 buffer=shm.buf)
+                lo = want * self.batch
+                labs = self.labels[lo:lo + cnt]
+                if use_gpu:
+                    src = torch.from_numpy(arr[:cnt])
+                    pinned[:cnt].copy_(src.float().div_(255.0))
+                    free_qs[w].put(slot)   # shm consumed into pinned
+                    with torch.cuda.stream(copy_stream):
+                        dev = pinned[:cnt].to(self.device,
+                                              non_blocking=True)
+                    ev = torch.cuda.Event()
+                    ev.record(copy_stream)
+                    torch.cuda.current_stream(self.device).wait_event(ev)
+                    # the next pinned fill happens on the host AFTER this
+                    # H2D: wait for the copy before reusing the buffer
+                    ev.synchronize()
+                    yield dev, labs.to(self.device)
+                else:
+                    out = torch.from_numpy(arr[:cnt].copy()).float() / 255.0
+                    free_qs[w].put(slot)
+                    yield out, labs
+        finally:
+            for p in procs:
+                p.terminate()
+            for s in shms:
+                s.close()
+                try:
+                    s.unlink()
+                except FileNotFoundError:
+                    pass

@@ -100,3 +100,17 @@ def test_image_batch_pipeline_h2d():
     got = torch.cat([b for b, _ in pipe])
     assert got.is_cuda
     assert (got.cpu() - imgs).abs().mean() < 0.06
+
+
+def test_process_image_pipeline_cpu():
+    from keystone_amd.loaders.input_pipeline import ProcessImagePipeline
+    imgs = _smooth_imgs(10, 24, 24)
+    blobs = encode_jpeg_batch(imgs, workers=2)
+    pipe = ProcessImagePipeline(blobs, list(range(10)), batch_size=4,
+                                device=None, workers=2)
+    batches = list(pipe)
+    assert [b.shape[0] for b, _ in batches] == [4, 4, 2]
+    labs = torch.cat([l for _, l in batches])
+    assert labs.tolist() == list(range(10))
+    got = torch.cat([b for b, _ in batches])
+    assert (got - imgs).abs().mean() < 0.06
