@@ -231,12 +231,21 @@ class ProcessImagePipeline:
                 p.daemon = True
                 p.start()
                 procs.append(p)
+            import queue as _queue
             ready = {}
-            for _ in range(nb):
-                # batches must be yielded in order
-                want = _
+            for want in range(nb):
+                # batches must be yielded in order; never block forever
+                # on a dead worker (a crashed decode would otherwise hang
+                # the caller — and the bench — indefinitely)
                 while want not in ready:
-                    bi, slot, cnt = done_q.get()
+                    try:
+                        bi, slot, cnt = done_q.get(timeout=30)
+                    except _queue.Empty:
+                        if not all(p.is_alive() for p in procs):
+                            raise RuntimeError(
+                                "ProcessImagePipeline: a decode worker "
+                                "died; aborting")
+                        continue
                     ready[bi] = (slot, cnt)
                 slot, cnt = ready.pop(want)
                 w = want % nworkers
