@@ -142,17 +142,20 @@ def main():
     # Single-GPU only (a per-rank failure inside a multi-rank barrier
     # region could hang the job), and failure never kills the bench.
     serve_input_ips = None
+    serve_input_mode = None
     if not args.cpu_smoke and rt.world_size == 1:
-        try:
-            from keystone_amd.loaders.input_pipeline import (
-                ProcessImagePipeline, encode_jpeg_batch)
-            from keystone_amd.parallel.dataset import TensorDataset
-            n_inp = min(2048, args.images_per_gpu)
-            blobs = encode_jpeg_batch(imgs.data[:n_inp], workers=16)
-            loader = ProcessImagePipeline(blobs, list(range(n_inp)),
-                                          batch_size=256,
-                                          color=not args.gray, device=dev,
-                                          workers=8)
+        import signal
+        from keystone_amd.loaders.input_pipeline import (
+            ImageBatchPipeline, ProcessImagePipeline, encode_jpeg_batch)
+        from keystone_amd.parallel.dataset import TensorDataset
+
+        def _wd(sig, frm):
+            raise TimeoutError("serve-with-input watchdog")
+
+        n_inp = min(2048, args.images_per_gpu)
+        blobs = encode_jpeg_batch(imgs.data[:n_inp], workers=16)
+
+        def run_loader(loader):
             for batch, _l in loader:    # warm (pin alloc, worker spinup)
                 pipe.apply(TensorDataset(batch, rt)).get()
                 break
@@ -161,10 +164,30 @@ def main():
             for batch, _l in loader:
                 pipe.apply(TensorDataset(batch, rt)).get()
             rt.synchronize()
-            serve_input_ips = n_inp / (time.perf_counter() - t2)
+            return n_inp / (time.perf_counter() - t2)
+
+        old = signal.signal(signal.SIGALRM, _wd)
+        # hard watchdog: a hung decode worker must never stall the bench
+        signal.alarm(420)
+        try:
+            try:
+                serve_input_ips = run_loader(ProcessImagePipeline(
+                    blobs, list(range(n_inp)), batch_size=256,
+                    color=not args.gray, device=dev, workers=8))
+                serve_input_mode = "process"
+            except Exception as e:
+                print(f"[bench] process input pipeline failed ({e!r}); "
+                      "falling back to threads", file=sys.stderr)
+                serve_input_ips = run_loader(ImageBatchPipeline(
+                    blobs, list(range(n_inp)), batch_size=256,
+                    color=not args.gray, device=dev, workers=16))
+                serve_input_mode = "threads"
         except Exception as e:
             print(f"[bench] serve-with-input skipped: {e!r}",
                   file=sys.stderr)
+        finally:
+            signal.alarm(0)
+            signal.signal(signal.SIGALRM, old)
 
     if os.environ.get("KEYSTONE_BENCH_STAGES") and rt.rank == 0:
         stages = {k: round(v, 4) for k, v in info.items()
@@ -208,6 +231,7 @@ def main():
                 "fit_time_s_per_step": ms_per_step / 1000.0,
                 "serve_images_per_sec": serve_images_per_sec,
                 "serve_with_input_images_per_sec": serve_input_ips,
+                "serve_input_mode": serve_input_mode,
                 "serve_mode": serve_mode,
                 "parallelism": f"dp{rt.world_size}",
             },
