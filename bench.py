@@ -152,19 +152,24 @@ def main():
         def _wd(sig, frm):
             raise TimeoutError("serve-with-input watchdog")
 
-        n_inp = min(2048, args.images_per_gpu)
+        n_inp = min(4096, args.images_per_gpu)
         blobs = encode_jpeg_batch(imgs.data[:n_inp], workers=16)
 
         def run_loader(loader):
-            for batch, _l in loader:    # warm (pin alloc, worker spinup)
-                pipe.apply(TensorDataset(batch, rt)).get()
-                break
+            # steady-state rate: time from after the FIRST batch (worker
+            # spinup + torch import in spawned processes) to the end of
+            # ONE pass
+            it = iter(loader)
+            batch, _l = next(it)
+            pipe.apply(TensorDataset(batch, rt)).get()
             rt.synchronize()
             t2 = time.perf_counter()
-            for batch, _l in loader:
+            cnt = 0
+            for batch, _l in it:
                 pipe.apply(TensorDataset(batch, rt)).get()
+                cnt += int(batch.shape[0])
             rt.synchronize()
-            return n_inp / (time.perf_counter() - t2)
+            return cnt / (time.perf_counter() - t2)
 
         old = signal.signal(signal.SIGALRM, _wd)
         # hard watchdog: a hung decode worker must never stall the bench
