@@ -14,9 +14,8 @@ is therefore decode-bound and scales with host cores — the honest
 from __future__ import annotations
 
 import io
-import os
 from concurrent.futures import ThreadPoolExecutor
-from typing import Callable, Iterable, Iterator, List, Optional, Sequence, Tuple
+from typing import Iterator, List, Optional, Sequence, Tuple
 
 import torch
 

@@ -29,13 +29,13 @@ columns cross xGMI.
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import List, Optional, Sequence
 
 import torch
 
 from ...parallel.dataset import Dataset, TensorDataset
-from ...workflow.pipeline import (BatchTransformer, LabelEstimator,
-                                  Transformer, TupleDataset)
+from ...workflow.pipeline import (LabelEstimator, Transformer,
+                                  TupleDataset)
 from .linear import _solve_dtype, _solve_spd
 
 __all__ = ["BlockLinearMapper", "BlockLeastSquaresEstimator",

@@ -15,8 +15,6 @@ fp32 shards (keystone_amd/hip); torch fallback elsewhere.
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import torch
 
 from ...parallel.dataset import TensorDataset

@@ -34,7 +34,7 @@ from typing import Callable, Dict, List, Optional
 import torch
 
 from .cache import Cacher
-from .graph import (DatasetOperator, Graph, GraphId, NodeId,
+from .graph import (DatasetOperator, Graph, NodeId,
                     TransformerOperator)
 from .pipeline import Transformer
 
