@@ -142,6 +142,9 @@ def _register_extra(lib: ctypes.CDLL) -> None:
         lib.ks_sift_norm_f32.argtypes = [c_p, ctypes.c_long, c_i, c_f, c_f,
                                          c_p]
         lib.ks_sift_norm_f32.restype = c_i
+        lib.ks_sift_norm2_f32.argtypes = [c_p, ctypes.c_long, c_i,
+                                          ctypes.c_long, c_i, c_f, c_f, c_p]
+        lib.ks_sift_norm2_f32.restype = c_i
     except AttributeError:
         pass
 
@@ -600,6 +603,25 @@ def sift_norm_(D: torch.Tensor, clamp: float = 0.2,
     _check(lib.ks_sift_norm_f32(
         ctypes.c_void_p(D.data_ptr()), ctypes.c_long(rows), dim, clamp, eps,
         _stream()), "ks_sift_norm_f32")
+    return D
+
+
+def sift_norm_slice_(D: torch.Tensor, clamp: float = 0.2,
+                     eps: float = 1e-20) -> torch.Tensor:
+    """In-place SIFT normalization of a [B, n, dim] SLICE of a larger
+    [B, n_total, dim] buffer (rows contiguous, batch pitch arbitrary) —
+    lets multi-scale descriptor assembly write each scale straight into
+    the final buffer with no torch.cat pass."""
+    lib = _load()
+    if D.dtype != torch.float32 or not D.is_cuda:
+        raise RuntimeError("sift_norm_slice_: needs CUDA float32")
+    B, nrows, dim = D.shape
+    if D.stride(2) != 1 or D.stride(1) != dim:
+        raise RuntimeError("sift_norm_slice_: rows must be contiguous")
+    _check(lib.ks_sift_norm2_f32(
+        ctypes.c_void_p(D.data_ptr()), ctypes.c_long(D.stride(0)), nrows,
+        ctypes.c_long(B * nrows), dim, clamp, eps, _stream()),
+        "ks_sift_norm2_f32")
     return D
 
 

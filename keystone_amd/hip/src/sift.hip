@@ -333,6 +333,59 @@ void sift_norm_reg_kernel(float* __restrict__ D, long rows, float clampv,
     for (int q = 0; q < NV; ++q) d[lane + 64 * q] = v[q] * inv2;
 }
 
+// 2-level-addressed variant: descriptor (b, i) lives at
+// D + b * bpitch + i * dim — normalizes IN PLACE inside a strided slice
+// of a preallocated [B, n_total, dim] buffer, so multi-scale assembly
+// needs no torch.cat pass.
+template <int NV>
+__global__ __launch_bounds__(256)
+void sift_norm_reg2_kernel(float* __restrict__ D, long bpitch, int nrows,
+                           long rows, float clampv, float eps) {
+    const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= rows) return;
+    const int lane = threadIdx.x & 63;
+    float* d = D + (row / nrows) * bpitch + (row % nrows) * (64 * NV);
+    float v[NV];
+    float ss = 0.0f;
+#pragma unroll
+    for (int q = 0; q < NV; ++q) {
+        v[q] = d[lane + 64 * q];
+        ss += v[q] * v[q];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss += __shfl_down(ss, off, 64);
+    ss = __shfl(ss, 0, 64);
+    const float inv = rsqrtf(fmaxf(ss, eps));
+    float ss2 = 0.0f;
+#pragma unroll
+    for (int q = 0; q < NV; ++q) {
+        v[q] = fminf(v[q] * inv, clampv);
+        ss2 += v[q] * v[q];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ss2 += __shfl_down(ss2, off, 64);
+    ss2 = __shfl(ss2, 0, 64);
+    const float inv2 = rsqrtf(fmaxf(ss2, eps));
+#pragma unroll
+    for (int q = 0; q < NV; ++q) d[lane + 64 * q] = v[q] * inv2;
+}
+
+KS_EXPORT int ks_sift_norm2_f32(float* D, long bpitch, int nrows, long rows,
+                                int dim, float clampv, float eps,
+                                hipStream_t stream) {
+    if (rows == 0) return 0;
+    long blocks = (rows + 3) / 4;
+    dim3 g((unsigned)blocks), b(256);
+    switch (dim) {
+        case 64:  hipLaunchKernelGGL(sift_norm_reg2_kernel<1>, g, b, 0, stream, D, bpitch, nrows, rows, clampv, eps); break;
+        case 128: hipLaunchKernelGGL(sift_norm_reg2_kernel<2>, g, b, 0, stream, D, bpitch, nrows, rows, clampv, eps); break;
+        case 192: hipLaunchKernelGGL(sift_norm_reg2_kernel<3>, g, b, 0, stream, D, bpitch, nrows, rows, clampv, eps); break;
+        case 256: hipLaunchKernelGGL(sift_norm_reg2_kernel<4>, g, b, 0, stream, D, bpitch, nrows, rows, clampv, eps); break;
+        default: return (int)hipErrorInvalidValue;
+    }
+    return (int)hipGetLastError();
+}
+
 KS_EXPORT int ks_sift_norm_f32(float* D, long rows, int dim, float clampv,
                                float eps, hipStream_t stream) {
     if (rows == 0) return 0;

@@ -188,8 +188,8 @@ class SIFTExtractor(Transformer):
         omap.scatter_add_(1, b1.unsqueeze(1), (mag * frac).unsqueeze(1))
         return omap
 
-    def _one_scale(self, imgs: torch.Tensor, bin_size: int, sigma: float
-                   ) -> torch.Tensor:
+    def _one_scale(self, imgs: torch.Tensor, bin_size: int, sigma: float,
+                   out: torch.Tensor = None) -> torch.Tensor:
         """[B, H, W] -> [B, n, 128] descriptors at one scale."""
         import os
         import time
@@ -279,11 +279,28 @@ class SIFTExtractor(Transformer):
             # 21k-element orientation plane (o-innermost measured 8.3 ms
             # of gather per step; this order ~3)
             s0, s1, s2 = P.stride()
-            D = P.as_strided(
+            src6 = P.as_strided(
                 (B, ny, nx, N_ORI, N_SPATIAL, N_SPATIAL),
                 (N_ORI * s0, sy_y * s2, sy_x * s1,
-                 s0, sj_x * s1, sj_y * s2)) \
-                .reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
+                 s0, sj_x * s1, sj_y * s2))
+            if out is not None:
+                # assemble straight into the caller's [B, n_total, 128]
+                # slice (no torch.cat pass across scales): ONE strided
+                # copy + the pitch-aware norm kernel
+                dst6 = out.as_strided(
+                    (B, ny, nx, N_ORI, N_SPATIAL, N_SPATIAL),
+                    (out.stride(0), nx * 128, 128, 16, 4, 1))
+                dst6.copy_(src6)
+                t0 = tick("gather", t0)
+                from ... import hip as hip_ops
+                hip_ops.sift_norm_slice_(out, self.clamp)
+                t0 = tick("norm", t0)
+                if dbg:
+                    print(f"[sift bin={bin_size}] "
+                          + " ".join(f"{k}={v * 1000:.1f}ms"
+                                     for k, v in _ph.items()), flush=True)
+                return out
+            D = src6.reshape(B, ny * nx, N_SPATIAL * N_SPATIAL * N_ORI)
             t0 = tick("gather", t0)
         else:
             Mw, nx = _sampling_matrix(W, bin_size, self.step,
@@ -334,11 +351,42 @@ class SIFTExtractor(Transformer):
         """[B, H, W] -> [B, n_total, 128(+2)] over all scales."""
         if imgs.dim() == 2:
             return self.apply_batch(imgs.unsqueeze(0)).squeeze(0)
-        outs = []
+        scales = []
         for s in range(self.num_scales):
             bs = max(1, int(round(self.bin_size * (self.scale_step ** s))))
-            sigma = bs / 3.0  # VLFeat magnif=3 convention
-            outs.append(self._one_scale(imgs, bs, sigma))
+            scales.append((bs, bs / 3.0))  # VLFeat magnif=3 convention
+
+        # GPU fast assembly: when every scale takes the merged lattice
+        # path, write each scale's descriptors straight into ONE
+        # preallocated [B, n_total, 128] buffer (skips the torch.cat)
+        if (imgs.is_cuda and imgs.dtype == torch.float32
+                and not self.append_xy):
+            from ... import hip as hip_ops
+            if hip_ops.gpu_ready():
+                B, H, W = imgs.shape
+                counts = []
+                ok = True
+                for bs, _sg in scales:
+                    Mwm, nx, Qx, _, _ = _merged_pool_matrix(
+                        W, bs, self.step, imgs.device, imgs.dtype)
+                    _, ny, _, _, _ = _merged_pool_matrix(
+                        H, bs, self.step, imgs.device, imgs.dtype)
+                    if not (nx and ny and Qx <= N_SPATIAL * nx):
+                        ok = False
+                        break
+                    counts.append(ny * nx)
+                if ok:
+                    Dtot = torch.empty(B, sum(counts),
+                                       N_SPATIAL * N_SPATIAL * N_ORI,
+                                       device=imgs.device, dtype=imgs.dtype)
+                    off = 0
+                    for (bs, sg), n_s in zip(scales, counts):
+                        self._one_scale(imgs, bs, sg,
+                                        out=Dtot[:, off:off + n_s])
+                        off += n_s
+                    return Dtot
+
+        outs = [self._one_scale(imgs, bs, sg) for bs, sg in scales]
         return torch.cat(outs, dim=1)
 
     def apply(self, img: torch.Tensor) -> torch.Tensor:
