@@ -467,3 +467,53 @@ def test_woodbury_kernel_matches_torch_path(hip, monkeypatch):
     Wt = torch.cat(mt.blocks, dim=0)
     err = (Wk - Wt).abs().max().item()
     assert err < 1e-4, err
+
+
+def test_pool1d_kernels_match_gemm(hip):
+    from keystone_amd.ops.images.sift import (_merged_pool_matrix,
+                                              _triangle_kernel1d)
+    import math as _m
+    torch.manual_seed(33)
+    R, W = 4096, 256
+    x = torch.randn(R, W, device="cuda")
+    for bins, step in [(6, 6), (8, 6), (4, 4)]:
+        g = _m.gcd(step, bins)
+        Mw, npos, Q, _, _ = _merged_pool_matrix(W, bins, step, x.device,
+                                                x.dtype)
+        tap = _triangle_kernel1d(bins, x.device, x.dtype)
+        ref = x @ Mw
+        out = hip.pool_lastdim(x, tap, bins // 2, g, Q)
+        assert (out - ref).abs().max().item() < 1e-4, (bins, step)
+        A1 = ref.reshape(16, R // 16, Q)
+        Mh, _, Qy, _, _ = _merged_pool_matrix(R // 16, bins, step, x.device,
+                                              x.dtype)
+        ref2 = torch.matmul(A1.transpose(1, 2), Mh)
+        out2 = hip.pool_middim(A1, tap, bins // 2, g, Qy)
+        assert (out2 - ref2).abs().max().item() < 1e-4, (bins, step)
+
+
+def test_sift_norm_slice_matches_plain(hip):
+    torch.manual_seed(34)
+    B, ntot, dim = 8, 100, 128
+    D = torch.randn(B, ntot, dim, device="cuda")
+    sl = D[:, 10:40].contiguous()          # reference: plain kernel
+    ref = hip.sift_norm_(sl.reshape(-1, dim).clone(), 0.2).reshape(
+        B, 30, dim)
+    hip.sift_norm_slice_(D[:, 10:40], 0.2)  # in place, pitched
+    assert torch.equal(D[:, 10:40], ref)
+
+
+def test_segsum_kernel_matches_index_add(hip):
+    torch.manual_seed(35)
+    n, db, K, G = 500, 96, 11, 11
+    X = torch.randn(n, db, device="cuda")
+    T = torch.randn(n, K, device="cuda")
+    cls = torch.sort(torch.randint(0, K, (n,), device="cuda")).values
+    counts = torch.bincount(cls, minlength=K)
+    ends = counts.cumsum(0)
+    starts = (ends - counts).to(torch.int32)
+    out = hip.segsum(X, T, starts, counts.to(torch.int32),
+                     torch.arange(K, dtype=torch.int32, device="cuda"))
+    v = T.gather(1, cls.unsqueeze(1))
+    ref = torch.zeros(K, db, device="cuda").index_add_(0, cls, X * v)
+    assert (out - ref).abs().max().item() < 1e-3
