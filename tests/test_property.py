@@ -88,3 +88,32 @@ def test_linear_mapper_roundtrip_property(d, k, seed, tmp_path_factory):
     a = pipe.apply(ka.from_rows(X)).get().data
     bb = loaded.apply(ka.from_rows(X)).get().data
     assert torch.allclose(a, bb, atol=1e-6)
+
+
+@settings(max_examples=10, deadline=None)
+@given(n=st.integers(24, 80), d=st.integers(4, 24), K=st.integers(2, 6),
+       bs=st.integers(2, 10), alpha=st.floats(0.05, 0.9),
+       seed=st.integers(0, 10_000))
+def test_weighted_bcd_matches_weighted_oracle(n, d, K, bs, alpha, seed):
+    """Per-column weighted normal equations over random shapes, class
+    distributions (possibly-empty classes included) and mixture weights."""
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    g = torch.Generator().manual_seed(seed)
+    X = torch.randn(n, d, generator=g, dtype=torch.float64)
+    y = torch.randint(0, K, (n,), generator=g)
+    Y = torch.full((n, K), -1.0, dtype=torch.float64)
+    Y[torch.arange(n), y] = 1.0
+    lam = 0.1
+    m = BlockWeightedLeastSquaresEstimator(
+        block_size=bs, num_iter=60, lam=lam, mixture_weight=alpha,
+        fit_intercept=False).fit(ka.from_rows(X), ka.from_rows(Y))
+    W = torch.cat(m.blocks, dim=0)
+    counts = torch.bincount(y, minlength=K).to(torch.float64)
+    for c in range(K):
+        w = torch.full((n,), (1 - alpha) / n, dtype=torch.float64)
+        if counts[c] > 0:
+            w[y == c] += alpha / counts[c]
+        G = X.T @ (w.unsqueeze(1) * X) + lam * torch.eye(
+            d, dtype=torch.float64)
+        ref = torch.linalg.solve(G, X.T @ (w * Y[:, c]))
+        assert torch.allclose(W[:, c], ref, atol=2e-4), (c, n, d, K, bs)
