@@ -125,6 +125,28 @@ def test_state_cache_reuses_fit(tmp_path):
         state_cache.disable_state_cache()
 
 
+def test_state_cache_does_not_alias_different_data(tmp_path):
+    """identical estimators fit on DIFFERENT datasets
+    must not return each other's cached models — the cache key folds in a
+    fit-data fingerprint."""
+    from keystone_amd.workflow import state_cache
+    from keystone_amd.ops.learning import LinearMapEstimator
+    torch.manual_seed(3)
+    X1, Y1 = torch.randn(30, 5), torch.randn(30, 2)
+    X2, Y2 = torch.randn(30, 5), torch.randn(30, 2)
+    state_cache.enable_state_cache(str(tmp_path / "sc2"))
+    try:
+        p1 = ka.Identity().and_then(LinearMapEstimator(0.1),
+                                    ka.from_rows(X1), ka.from_rows(Y1)).fit()
+        p2 = ka.Identity().and_then(LinearMapEstimator(0.1),
+                                    ka.from_rows(X2), ka.from_rows(Y2)).fit()
+        a = p1.apply(ka.from_rows(X1)).get().data
+        b = p2.apply(ka.from_rows(X1)).get().data
+        assert not torch.allclose(a, b, atol=1e-6)  # distinct fits
+    finally:
+        state_cache.disable_state_cache()
+
+
 def test_shuffler_preserves_rows():
     from keystone_amd.ops.util import Shuffler
     X = torch.arange(20, dtype=torch.float32).reshape(10, 2)
