@@ -231,3 +231,23 @@ def test_krr_distributed_matches_local():
     results = run_distributed(_krr_fit_predict, 2)
     for pred in results:
         assert torch.allclose(pred, ref, atol=1e-3)
+
+
+def _async_reduce(rank, world):
+    import torch as t
+    from keystone_amd.parallel.runtime import get_runtime
+    rt = get_runtime()
+    xs = [t.full((4, 3), float(rank + i)) for i in range(3)]
+    handles = [rt.all_reduce_async_(x) for x in xs]
+    for h in handles:
+        if h is not None:
+            h.wait()
+    return xs
+
+
+def test_all_reduce_async_matches_sum():
+    results = run_distributed(_async_reduce, 2)
+    for i in range(3):
+        expect = torch.full((4, 3), float(0 + i) + float(1 + i))
+        assert torch.equal(results[0][i], expect)
+        assert torch.equal(results[1][i], expect)

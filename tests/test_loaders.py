@@ -114,3 +114,26 @@ def test_process_image_pipeline_cpu():
     assert labs.tolist() == list(range(10))
     got = torch.cat([b for b, _ in batches])
     assert (got - imgs).abs().mean() < 0.06
+
+
+def _tar_loader_world2(rank, world, tmpdir):
+    from keystone_amd.loaders.files import ImageNetLoader
+    imgs, labels = ImageNetLoader(tmpdir)
+    return labels.data, torch.stack(list(imgs.items())).shape
+
+
+def test_imagenet_tar_shards_split_across_ranks(tmp_path):
+    """4 tar shards, world 2: each rank streams 2 whole shards."""
+    from tests.dist_util import run_distributed
+    for shard in range(4):
+        with tarfile.open(tmp_path / f"s{shard}.tar", "w") as tf:
+            for ci, cls in enumerate(["n01", "n02"]):
+                data = _jpeg_bytes(seed=shard * 100 + ci)
+                info = tarfile.TarInfo(f"{cls}/im{shard}.JPEG")
+                info.size = len(data)
+                tf.addfile(info, io.BytesIO(data))
+    results = run_distributed(_tar_loader_world2, 2, args=(str(tmp_path),))
+    all_labels = torch.cat([r[0] for r in results])
+    assert sorted(all_labels.tolist()) == [0, 0, 0, 0, 1, 1, 1, 1]
+    for _labels, shape in results:
+        assert shape == (4, 3, 32, 32)   # 2 shards x 2 images per rank
