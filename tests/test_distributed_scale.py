@@ -242,3 +242,32 @@ def test_weighted_lazy_blocks_world4():
     for W, b in results:
         assert torch.allclose(W, ref_W, atol=1e-6)
         assert torch.allclose(b, ref_b, atol=1e-6)
+
+
+def _gather_rows(rank, world):
+    import torch as t
+    from keystone_amd.parallel.runtime import get_runtime
+    rt = get_runtime()
+    # unequal local row counts (rank+1 rows)
+    local = t.arange(rank + 1, dtype=t.float64).unsqueeze(1) + 10 * rank
+    return rt.all_gather_rows(local)
+
+
+def test_all_gather_rows_unequal_world3():
+    results = run_distributed(_gather_rows, 3)
+    expect = torch.cat([torch.arange(r + 1, dtype=torch.float64)
+                        .unsqueeze(1) + 10 * r for r in range(3)])
+    for got in results:
+        assert torch.equal(got, expect)
+
+
+def test_weighted_world3_remainder_shards():
+    """Odd world size: shard_bounds remainders + class re-shard with
+    uneven per-rank row counts."""
+    K = 4
+    X, Y = _make_xy(K)
+    kw = dict(block_size=7, num_iter=15, lam=1e-2, mixture_weight=0.3)
+    ref_W, _ = _weighted_ref(X, Y, kw)
+    results = run_distributed(_weighted_fit, 3, args=(X, Y, kw))
+    for W, _b in results:
+        assert torch.allclose(W, ref_W, atol=1e-6)
