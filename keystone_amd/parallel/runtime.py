@@ -103,6 +103,10 @@ class Runtime:
         call; stripes better across the 7 xGMI links than many small calls)."""
         if not self.is_distributed or not tensors:
             return
+        if any(t.dtype != tensors[0].dtype for t in tensors):
+            # a mixed-dtype cat would silently upcast some summands
+            raise TypeError("all_reduce_many_ requires uniform dtype; got "
+                            + str([str(t.dtype) for t in tensors]))
         dev = tensors[0].device
         flat = torch.cat([t.reshape(-1).to(dev) for t in tensors])
         self.all_reduce_(flat, op)
