@@ -185,3 +185,34 @@ def test_profiler_produces_stats():
     assert prof, "profiler returned no records"
     rec = next(iter(prof.values()))
     assert rec["seconds"] >= 0 and rec["bytes"] > 0
+
+
+def test_util_nodes_semantics():
+    """TopK ordering, Densify/Sparsify round trip, AugmentedExamples vote
+    (ref: nodes.util + AugmentedExamplesEvaluator semantics)."""
+    from keystone_amd.ops.util import (Densify, Sparsify, TopKClassifier)
+    s = torch.tensor([[0.1, 0.9, 0.5, 0.7], [1.0, -1.0, 0.0, 0.5]])
+    top = TopKClassifier(3).apply_batch(s)
+    assert top.tolist() == [[1, 3, 2], [0, 3, 2]]   # best first
+
+    X = torch.tensor([[0.0, 2.0, 0.0], [1.0, 0.0, 0.0]])
+    sp = Sparsify().apply_batch(X)
+    assert sp.is_sparse_csr or sp.layout != torch.strided
+    back = Densify().apply_batch(sp)
+    assert torch.equal(back, X)
+
+
+def test_augmented_examples_evaluator_vote():
+    from keystone_amd.evaluation import AugmentedExamplesEvaluator
+    import keystone_amd as ka
+    # 2 originals x 3 augmentations; scores vote per original
+    scores = torch.tensor([[1.0, 0.0], [0.8, 0.1], [0.0, 2.0],
+                           [0.0, 1.0], [0.2, 0.9], [0.1, 0.8]])
+    ids = torch.tensor([0, 0, 0, 1, 1, 1])
+    labels = torch.tensor([0, 0, 0, 1, 1, 1])   # per-augmentation rows
+    ev = AugmentedExamplesEvaluator(2)
+    m = ev.evaluate(ka.from_rows(scores), ka.from_rows(ids),
+                    ka.from_rows(labels))
+    # image 0: summed scores (1.8, 2.1) -> argmax 1 != label 0 (the big
+    # augmentation outlier wins); image 1: (0.3, 2.7) -> 1 == label
+    assert abs(m.accuracy - 0.5) < 1e-9
