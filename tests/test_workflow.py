@@ -215,4 +215,4 @@ def test_augmented_examples_evaluator_vote():
                     ka.from_rows(labels))
     # image 0: summed scores (1.8, 2.1) -> argmax 1 != label 0 (the big
     # augmentation outlier wins); image 1: (0.3, 2.7) -> 1 == label
-    assert abs(m.accuracy - 0.5) < 1e-9
+    assert abs(m.total_accuracy - 0.5) < 1e-9
