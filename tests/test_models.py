@@ -169,3 +169,20 @@ def test_cli_cifar_and_voc(tmp_path):
         timeout=300)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "train mAP" in out.stdout
+
+
+def test_mean_ap_hand_computed():
+    """AP against a worked example, plain and 11-point interpolated."""
+    # one class, ranked relevance pattern [1, 0, 1, 0]:
+    # plain AP = (1/1 + 2/3) / 2 = 5/6
+    scores = torch.tensor([[0.9], [0.8], [0.7], [0.1]])
+    acts = torch.tensor([[1.0], [0.0], [1.0], [0.0]])
+    ev = MeanAveragePrecisionEvaluator(1)
+    res = ev.evaluate(ka.from_rows(scores), ka.from_rows(acts))
+    assert abs(res["mAP"] - 5.0 / 6.0) < 1e-9
+    # 11-pt: recall thresholds 0..1; max precision at recall>=t is 1.0
+    # for t <= 0.5 (6 thresholds) and 2/3 for t > 0.5 (5 thresholds)
+    ev11 = MeanAveragePrecisionEvaluator(1, interpolated_11pt=True)
+    res11 = ev11.evaluate(ka.from_rows(scores), ka.from_rows(acts))
+    expect = (6 * 1.0 + 5 * (2.0 / 3.0)) / 11
+    assert abs(res11["mAP"] - expect) < 1e-9
