@@ -435,8 +435,10 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
                 return torch.cat(rt.all_to_all(parts), dim=0)
 
             Y = _exchange(Y)
-            cls = _exchange(cls.unsqueeze(1).to(torch.float32)) \
-                .squeeze(1).long()
+            # int64 end-to-end: both all_to_all backends move integer
+            # tensors natively (a float32 bounce would lose exactness
+            # past 2^24 classes)
+            cls = _exchange(cls.unsqueeze(1)).squeeze(1)
             if isinstance(data, LazyBlocks):
                 raw = data.raw
                 if not isinstance(raw, TensorDataset):
