@@ -114,3 +114,20 @@ def test_loaders_files(tmp_path):
                   json.dumps({"reviewText": "bad", "overall": 1}) + "\n")
     docs, labels = AmazonReviewsDataLoader(str(aj))
     assert labels.data.tolist() == [1, 0]
+
+
+def test_stupid_backoff_unseen_and_backoff_weighting():
+    """Backoff semantics: an unseen trigram backs off to the bigram with
+    an alpha penalty; fully-unseen words get the floor, not a crash."""
+    from keystone_amd.ops.nlp import StupidBackoffEstimator
+    corpus = [["a", "b", "c"], ["a", "b", "d"], ["x", "b", "c"]] * 2
+    lm = StupidBackoffEstimator(3, alpha=0.4).fit(ka.from_items(corpus))
+    seen = lm.apply(["a", "b", "c"])          # seen trigram path
+    backed = lm.apply(["x", "b", "d"])        # (x b d) unseen -> backoff
+    unseen = lm.apply(["zz", "qq"])           # OOV words -> floor
+    assert seen > backed > unseen
+    # direct alpha check: S(d | x b) should equal alpha * S(d | b)
+    xv = lm.vocab
+    s_backed = lm._score_ids([xv["x"], xv["b"]], xv["d"])
+    s_bigram = lm._score_ids([xv["b"]], xv["d"])
+    assert abs(s_backed - 0.4 * s_bigram) < 1e-12
