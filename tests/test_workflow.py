@@ -216,3 +216,26 @@ def test_augmented_examples_evaluator_vote():
     # image 0: summed scores (1.8, 2.1) -> argmax 1 != label 0 (the big
     # augmentation outlier wins); image 1: (0.3, 2.7) -> 1 == label
     assert abs(m.total_accuracy - 0.5) < 1e-9
+
+
+def test_vector_splitter_feeds_block_solver():
+    """VectorSplitter -> TupleDataset -> block solver: the reference's
+    feature-block 'model parallel' axis end to end."""
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    from keystone_amd.ops.util import Cast, VectorCombiner, VectorSplitter
+    torch.manual_seed(6)
+    X = torch.randn(80, 12, dtype=torch.float64)
+    W_true = torch.randn(12, 2, dtype=torch.float64)
+    Y = X @ W_true
+    parts = VectorSplitter(5).apply_dataset(ka.from_rows(X))
+    m = BlockLeastSquaresEstimator(block_size=5, num_iter=40, lam=1e-8,
+                                   fit_intercept=False).fit(
+        parts, ka.from_rows(Y))
+    assert [b.shape[0] for b in m.blocks] == [5, 5, 2]  # ragged last block
+    pred = m.apply(X)
+    assert torch.allclose(pred, Y, atol=1e-5)
+    # datum-level splitter + combiner round trip, and Cast semantics
+    x = torch.randn(12)
+    back = VectorCombiner().apply(VectorSplitter(5).apply(x))
+    assert torch.equal(back, x)
+    assert Cast("float64").apply_batch(x).dtype == torch.float64
