@@ -281,3 +281,30 @@ def test_fast_windower_matches_windower():
             assert torch.equal(slow, fast), (shape, stride, size)
     x = torch.randn(3, 12, 14)
     assert torch.equal(Windower(2, 4).apply(x), FastWindower(2, 4).apply(x))
+
+
+def test_random_image_transformer_flip_and_crop():
+    from keystone_amd.ops.images import RandomImageTransformer
+    x = torch.arange(24.0).reshape(4, 6)
+    always = RandomImageTransformer(flip_prob=1.0, seed=1)
+    assert torch.equal(always.apply(x), x.flip(-1))
+    never = RandomImageTransformer(flip_prob=0.0, crop_size=3, seed=2)
+    out = never.apply(x)
+    assert out.shape[-2:] == (3, 3)
+
+
+def test_padded_fft_matches_manual():
+    from keystone_amd.ops.stats import PaddedFFT
+    torch.manual_seed(13)
+    X = torch.randn(5, 6)                  # pads to 8
+    out = PaddedFFT().apply_batch(X)
+    ref = torch.fft.rfft(X, n=8, dim=-1).abs()[..., 1:]
+    assert torch.allclose(out, ref, atol=1e-6)
+    assert out.shape == (5, 4)             # rfft(8) -> 5 bins, DC dropped
+
+
+def test_linear_rectifier_semantics():
+    from keystone_amd.ops.stats import LinearRectifier
+    x = torch.tensor([[-1.0, 0.2, 2.0]])
+    out = LinearRectifier(max_val=0.0, alpha=0.5).apply_batch(x)
+    assert torch.equal(out, torch.tensor([[0.0, 0.0, 1.5]]))

@@ -417,3 +417,18 @@ def test_block_weighted_rejects_multilabel():
     with pytest.raises(ValueError, match="single-label"):
         BlockWeightedLeastSquaresEstimator(block_size=3, num_iter=2).fit(
             ka.from_rows(X), ka.from_rows(Y))
+
+
+def test_minibatch_sgd_converges():
+    from keystone_amd.ops.learning import LeastSquaresGradient
+    from keystone_amd.ops.learning.lbfgs import MiniBatchSGDwithL2
+    torch.manual_seed(12)
+    X = torch.randn(400, 8)
+    W_true = torch.randn(8, 2)
+    Y = X @ W_true
+    m = MiniBatchSGDwithL2(LeastSquaresGradient(), num_iterations=1500,
+                           step_size=0.1, batch_fraction=0.5,
+                           reg_param=0.0).fit(ka.from_rows(X),
+                                              ka.from_rows(Y))
+    rel = (m.apply_batch(X) - Y).norm() / Y.norm()
+    assert rel.item() < 0.05, rel.item()
