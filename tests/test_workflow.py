@@ -239,3 +239,42 @@ def test_vector_splitter_feeds_block_solver():
     back = VectorCombiner().apply(VectorSplitter(5).apply(x))
     assert torch.equal(back, x)
     assert Cast("float64").apply_batch(x).dtype == torch.float64
+
+
+def test_checkpointer_spills_and_reloads(tmp_path):
+    from keystone_amd.workflow.cache import Checkpointer
+    X = torch.randn(12, 5)
+    ds = ka.from_rows(X)
+    out = (ka.Identity().and_then(Checkpointer(str(tmp_path / "ck")))
+           .apply(ds).get())
+    assert torch.equal(out.data, X)          # identity semantics
+    re = Checkpointer.load(str(tmp_path / "ck"))
+    assert torch.equal(re.data, X)           # per-rank shard round-trips
+
+
+def test_unused_branch_removal_rule():
+    from keystone_amd.workflow.optimizer import UnusedBranchRemovalRule
+    X = torch.randn(6, 3)
+    pipe = ka.Identity().and_then(ka.FunctionTransformer(lambda t: t * 2))
+    lazy = pipe.apply(ka.from_rows(X))
+    g = lazy.graph.copy()
+    # graft a dead branch off an existing node: never feeds any sink
+    anchor = next(iter(g.operators))
+    dead = g.add_node(TransformerOperator(ka.FunctionTransformer(
+        lambda t: t + 99, name="dead")), [anchor])
+    n_before = len(g.operators)
+    g2 = UnusedBranchRemovalRule().apply(g)
+    assert dead not in g2.operators
+    assert len(g2.operators) == n_before - 1
+    from keystone_amd.workflow.executor import GraphExecutor
+    out = GraphExecutor(g2, optimize=False).execute(lazy.sink)
+    assert torch.allclose(torch.stack(list(out.items())), X * 2)
+
+
+def test_function_transformer_batch_and_name():
+    f = ka.FunctionTransformer(lambda t: t + 1, name="plus1")
+    assert f.name == "plus1"
+    X = torch.randn(4, 3)
+    # datum-only transformer: the dataset lift maps per row -> ItemDataset
+    out = ka.Identity().and_then(f).apply(ka.from_rows(X)).get()
+    assert torch.allclose(torch.stack(list(out.items())), X + 1)
