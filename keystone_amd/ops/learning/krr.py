@@ -143,7 +143,11 @@ class KernelRidgeRegression(LabelEstimator):
                 ab = rt.all_gather_rows(TensorDataset(alpha[sl], rt).data) \
                     if rt.is_distributed else alpha[sl]
                 K_bb = self.kernel(Xb, Xb)
-                resid = Yb - pred + K_bb @ ab + lam_n * ab
+                # exact block coordinate update of (K + lam n I) alpha = Y:
+                # (K_bb + lam n I) a_b_new = Y_b - K_{b,:} alpha + K_bb a_b
+                # (pred contains K_bb a_b_old, so add it back; the lam_n
+                # term belongs ONLY on the left side)
+                resid = Yb - pred + K_bb @ ab
                 # solve (K_bb + lam n I) a_b = resid  (replicated, fp64)
                 A = K_bb.to(torch.float64) + lam_n * torch.eye(
                     Xb.shape[0], dtype=torch.float64, device=Xb.device)

@@ -117,3 +117,29 @@ def test_weighted_bcd_matches_weighted_oracle(n, d, K, bs, alpha, seed):
             d, dtype=torch.float64)
         ref = torch.linalg.solve(G, X.T @ (w * Y[:, c]))
         assert torch.allclose(W[:, c], ref, atol=2e-4), (c, n, d, K, bs)
+
+
+@settings(max_examples=10, deadline=None)
+@given(n=st.integers(12, 60), d=st.integers(2, 8), k=st.integers(1, 3),
+       bs=st.integers(4, 16), gamma=st.floats(0.05, 0.5),
+       seed=st.integers(0, 10_000))
+def test_krr_dual_bcd_matches_exact_oracle(n, d, k, bs, gamma, seed):
+    """Dual BCD converges to alpha = (K + lam*n I)^-1 Y over random
+    shapes/kernel widths; prediction on fresh points must match the
+    closed form."""
+    from keystone_amd.ops.learning import (GaussianKernelGenerator,
+                                           KernelRidgeRegression)
+    g = torch.Generator().manual_seed(seed)
+    X = torch.randn(n, d, generator=g, dtype=torch.float64)
+    Y = torch.randn(n, k, generator=g, dtype=torch.float64)
+    lam = 0.1
+    kern = GaussianKernelGenerator(gamma)
+    m = KernelRidgeRegression(kern, lam, block_size=bs, num_epochs=40).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    K = kern(X, X)
+    alpha_ref = torch.linalg.solve(
+        K + lam * n * torch.eye(n, dtype=torch.float64), Y)
+    Xt = torch.randn(7, d, generator=g, dtype=torch.float64)
+    pred = m.apply_batch(Xt)
+    ref = kern(Xt, X) @ alpha_ref
+    assert torch.allclose(pred, ref, atol=1e-5), (pred - ref).abs().max()
