@@ -432,3 +432,57 @@ def test_minibatch_sgd_converges():
                                               ka.from_rows(Y))
     rel = (m.apply_batch(X) - Y).norm() / Y.norm()
     assert rel.item() < 0.05, rel.item()
+
+
+def test_lbfgs_matches_ridge_oracle_with_l2():
+    """Regularized fixed point: the objective is (1/n)·0.5||XW−Y||² +
+    0.5·reg·||W||², so the minimizer solves (XᵀX/n + reg·I) W = XᵀY/n.
+    Guards against the lam-scaling bug family (cf. the KRR block-update
+    fix) that a reg=0 test cannot see."""
+    g = torch.Generator().manual_seed(3)
+    n, d, k, reg = 120, 8, 3, 0.3
+    X = torch.randn(n, d, generator=g)
+    Y = torch.randn(n, k, generator=g)
+    m = DenseLBFGSwithL2(LeastSquaresGradient(), num_iterations=200,
+                         reg_param=reg).fit(ka.from_rows(X), ka.from_rows(Y))
+    Xd, Yd = X.to(torch.float64), Y.to(torch.float64)
+    ref = torch.linalg.solve(
+        Xd.T @ Xd / n + reg * torch.eye(d, dtype=torch.float64),
+        Xd.T @ Yd / n)
+    assert torch.allclose(m.W.to(torch.float64), ref, atol=1e-3), \
+        (m.W.to(torch.float64) - ref).abs().max()
+
+
+def test_logistic_matches_autograd_oracle():
+    """Same objective ((1/n)·softmax-CE + 0.5·reg·||W||², bias column
+    augmented and regularized like the solver does) optimized with torch
+    autograd LBFGS in fp64 must land on the same W."""
+    g = torch.Generator().manual_seed(11)
+    n, d, K, reg = 150, 6, 3, 0.05
+    X = torch.randn(n, d, generator=g)
+    y = torch.randint(0, K, (n,), generator=g)
+    model = LogisticRegressionEstimator(num_classes=K, num_iters=300,
+                                        reg_param=reg).fit(
+        ka.from_rows(X), ka.from_rows(y))
+    W_ours = torch.cat([model.W, model.b.unsqueeze(0)]).to(torch.float64)
+
+    Xa = torch.cat([X, torch.ones(n, 1)], dim=1).to(torch.float64)
+    Yh = torch.nn.functional.one_hot(y, K).to(torch.float64)
+    W = torch.zeros(d + 1, K, dtype=torch.float64, requires_grad=True)
+    opt = torch.optim.LBFGS([W], max_iter=400, tolerance_grad=1e-12,
+                            line_search_fn="strong_wolfe")
+
+    def closure():
+        opt.zero_grad()
+        S = Xa @ W
+        loss = (torch.logsumexp(S, dim=1) - (S * Yh).sum(1)).mean() \
+            + 0.5 * reg * (W ** 2).sum()
+        loss.backward()
+        return loss
+
+    opt.step(closure)
+    ref = W.detach()
+    # softmax W has a shift gauge only without regularization; with reg>0
+    # the optimum is unique, compare directly
+    assert torch.allclose(W_ours, ref, atol=2e-3), \
+        (W_ours - ref).abs().max()
