@@ -486,3 +486,37 @@ def test_logistic_matches_autograd_oracle():
     # the optimum is unique, compare directly
     assert torch.allclose(W_ours, ref, atol=2e-3), \
         (W_ours - ref).abs().max()
+
+
+def test_gmm_em_fixed_point_oracle():
+    """At convergence the fitted (mu, var, w) must satisfy the EM
+    fixed-point equations with posteriors recomputed from scratch via the
+    textbook diagonal-Gaussian Bayes formula in fp64 — validates both the
+    E-step (log-prob/softmax) and M-step (weighted MLE) algebra, which
+    the recovery test alone cannot."""
+    torch.manual_seed(7)
+    mu0 = torch.tensor([[3.0, -1.0, 0.0], [-2.0, 2.0, 1.0], [0.0, 0.0, -3.0]])
+    X = torch.cat([mu0[i] + torch.randn(250, 3) * (0.5 + 0.3 * i)
+                   for i in range(3)])
+    gmm = GaussianMixtureModelEstimator(3, max_iterations=200, tol=1e-12,
+                                        seed=1).fit(ka.from_rows(X))
+    Xd = X.to(torch.float64)
+    m = gmm.means.to(torch.float64)
+    v = gmm.variances.to(torch.float64)
+    w = gmm.weights.to(torch.float64)
+    # hand E-step
+    lp = (-0.5 * ((Xd.unsqueeze(1) - m) ** 2 / v).sum(2)
+          - 0.5 * v.log().sum(1) - 0.5 * 3 * torch.log(
+              torch.tensor(2 * torch.pi, dtype=torch.float64))
+          + w.log())
+    gam = torch.softmax(lp, dim=1)
+    # model's own posteriors must match the textbook formula
+    assert torch.allclose(gmm.apply_batch(X).to(torch.float64), gam,
+                          atol=1e-5)
+    # hand M-step from those posteriors reproduces the model (fixed point)
+    Nk = gam.sum(0)
+    mu_next = gam.T @ Xd / Nk.unsqueeze(1)
+    var_next = gam.T @ (Xd ** 2) / Nk.unsqueeze(1) - mu_next ** 2
+    assert torch.allclose(m, mu_next, atol=1e-3), (m - mu_next).abs().max()
+    assert torch.allclose(v, var_next, atol=1e-3)
+    assert torch.allclose(w, Nk / X.shape[0], atol=1e-4)
