@@ -174,9 +174,15 @@ def main():
         pipe, info = fit_imagenet_fv(imgs, labels, conf)
         if args.save:
             pipe.save(args.save)
+        import keystone_amd as ka
+        # the pipeline ends in TopKClassifier: column 0 is the top-1 class
+        pred = pipe.apply(imgs).get().data[:, 0]
+        m = MulticlassClassifierEvaluator(args.num_classes).evaluate(
+            ka.from_rows(pred), labels)
         if rt.rank == 0:
             print(f"fit {info['t_total']:.2f}s, feature_dim "
-                  f"{info['feature_dim']}  ({time.time()-t0:.1f}s total)")
+                  f"{info['feature_dim']}, train accuracy "
+                  f"{m.total_accuracy:.3f}  ({time.time()-t0:.1f}s total)")
 
 
 if __name__ == "__main__":

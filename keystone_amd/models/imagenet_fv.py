@@ -156,6 +156,7 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
     info["model"] = model
 
     pipe = (gather(branches).and_then(VectorCombiner())
-            .and_then(model).and_then(TopKClassifier(conf.top_k)))
+            .and_then(model)
+            .and_then(TopKClassifier(min(conf.top_k, conf.num_classes))))
     info["t_total"] = time.perf_counter() - t0
     return pipe, info

@@ -99,6 +99,25 @@ def test_cli_mnist_smoke():
     assert "accuracy" in out.stdout
 
 
+@pytest.mark.parametrize("cmd", [
+    ["newsgroups", "--n-docs", "200", "--num-classes", "4",
+     "--common-features", "500"],
+    ["amazon", "--n-docs", "200"],
+    ["timit", "--n", "1500", "--num-features", "512", "--epochs", "1"],
+    ["imagenet-fv", "--images", "12", "--image-size", "64",
+     "--num-classes", "4", "--gmm-k", "4"],
+])
+def test_cli_apps_smoke(cmd):
+    """Every reference example app's CLI twin runs end-to-end on tiny
+    synthetic data (mnist/cifar/voc covered elsewhere)."""
+    out = subprocess.run(
+        [sys.executable, "bin/run_pipeline.py"] + cmd,
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "accuracy" in out.stdout or "mAP" in out.stdout
+
+
 def test_state_cache_reuses_fit(tmp_path):
     from keystone_amd.workflow import state_cache
     from keystone_amd.ops.learning import LinearMapEstimator
