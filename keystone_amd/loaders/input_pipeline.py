@@ -236,14 +236,23 @@ class ProcessImagePipeline:
                 # batches must be yielded in order; never block forever
                 # on a dead worker (a crashed decode would otherwise hang
                 # the caller — and the bench — indefinitely)
+                waited = 0.0
                 while want not in ready:
                     try:
-                        bi, slot, cnt = done_q.get(timeout=30)
+                        bi, slot, cnt = done_q.get(timeout=2)
+                        waited = 0.0
                     except _queue.Empty:
+                        # short poll => a crashed decode worker (bad blob,
+                        # OOM-killed) is detected in ~2s, not 30
                         if not all(p.is_alive() for p in procs):
                             raise RuntimeError(
                                 "ProcessImagePipeline: a decode worker "
                                 "died; aborting")
+                        waited += 2.0
+                        if waited >= 60.0:
+                            raise RuntimeError(
+                                "ProcessImagePipeline: no batch for 60s "
+                                "with live workers; aborting")
                         continue
                     ready[bi] = (slot, cnt)
                 slot, cnt = ready.pop(want)

@@ -137,3 +137,20 @@ def test_imagenet_tar_shards_split_across_ranks(tmp_path):
     assert sorted(all_labels.tolist()) == [0, 0, 0, 0, 1, 1, 1, 1]
     for _labels, shape in results:
         assert shape == (4, 3, 32, 32)   # 2 shards x 2 images per rank
+
+
+def test_process_pipeline_dead_worker_aborts_fast():
+    """A corrupt blob kills its decode worker; the iterator must raise
+    within a few seconds instead of hanging the caller (the short-poll
+    liveness check)."""
+    import time
+    from keystone_amd.loaders.input_pipeline import ProcessImagePipeline
+    imgs = _smooth_imgs(8, 24, 24)
+    blobs = encode_jpeg_batch(imgs, workers=2)
+    blobs[5] = b"not a jpeg at all"
+    pipe = ProcessImagePipeline(blobs, list(range(8)), batch_size=4,
+                                device=None, workers=2)
+    t0 = time.monotonic()
+    with pytest.raises(RuntimeError, match="decode worker"):
+        list(pipe)
+    assert time.monotonic() - t0 < 30
