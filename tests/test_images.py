@@ -308,3 +308,25 @@ def test_linear_rectifier_semantics():
     x = torch.tensor([[-1.0, 0.2, 2.0]])
     out = LinearRectifier(max_val=0.0, alpha=0.5).apply_batch(x)
     assert torch.equal(out, torch.tensor([[0.0, 0.0, 1.5]]))
+
+
+def test_convolver_matches_naive_loop_oracle():
+    """All three modes vs a quadruple-loop correlation oracle — cross-mode
+    agreement alone could hide a shared convention error (flip vs no
+    flip, offset)."""
+    torch.manual_seed(4)
+    B, C, H, W, Fn, hh, ww = 2, 3, 9, 8, 4, 3, 3
+    X = torch.randn(B, C, H, W)
+    filt = torch.randn(Fn, C, hh, ww)
+    Ho, Wo = H - hh + 1, W - ww + 1
+    ref = torch.zeros(B, Fn, Ho, Wo, dtype=torch.float64)
+    Xd, fd = X.to(torch.float64), filt.to(torch.float64)
+    for b in range(B):
+        for f in range(Fn):
+            for i in range(Ho):
+                for j in range(Wo):
+                    ref[b, f, i, j] = (Xd[b, :, i:i + hh, j:j + ww]
+                                       * fd[f]).sum()
+    for mode in ("im2col", "conv", "fft"):
+        out = Convolver(filt, mode=mode).apply_batch(X)
+        assert torch.allclose(out.to(torch.float64), ref, atol=1e-4), mode
