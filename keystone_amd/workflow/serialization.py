@@ -106,6 +106,12 @@ def load_pipeline(path: str) -> Pipeline:
     id_map: Dict[int, GraphId] = {manifest["source"]: src}
     for entry in manifest["nodes"]:
         cls = _load_class(entry["class"])
+        # the manifest names arbitrary import paths: refuse anything that
+        # is not a Transformer (clearer error on corrupt/hostile files
+        # than an AttributeError deep in from_saved)
+        if not (isinstance(cls, type) and issubclass(cls, Transformer)):
+            raise ValueError(f"manifest names non-Transformer class "
+                             f"{entry['class']!r}")
         state = {}
         if "state" in entry:
             state = load_file(os.path.join(path, entry["state"]))

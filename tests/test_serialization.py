@@ -133,3 +133,25 @@ def test_weighted_block_mapper_with_intercept_roundtrips(tmp_path):
     a = pipe.apply(X).get()
     b = loaded.apply(X).get()
     assert torch.allclose(a, b, atol=1e-12)
+
+
+def test_load_rejects_newer_format_and_foreign_class(tmp_path):
+    import json
+    import pytest
+    import keystone_amd as ka
+    from keystone_amd.ops.stats import NormalizeRows
+    path = str(tmp_path / "m")
+    (ka.Identity().and_then(NormalizeRows())).save(path)
+
+    mpath = tmp_path / "m" / "manifest.json"
+    man = json.loads(mpath.read_text())
+    man2 = dict(man, format_version=man["format_version"] + 1)
+    mpath.write_text(json.dumps(man2))
+    with pytest.raises(ValueError, match="newer"):
+        ka.Pipeline.load(path)
+
+    man3 = json.loads(json.dumps(man))
+    man3["nodes"][0]["class"] = "os.path.join"   # not a Transformer
+    mpath.write_text(json.dumps(man3))
+    with pytest.raises(ValueError, match="non-Transformer"):
+        ka.Pipeline.load(path)
