@@ -251,3 +251,37 @@ def test_all_reduce_async_matches_sum():
         expect = torch.full((4, 3), float(0 + i) + float(1 + i))
         assert torch.equal(results[0][i], expect)
         assert torch.equal(results[1][i], expect)
+
+
+def _nlp_fit(rank, world, docs_all):
+    import keystone_amd as ka
+    from keystone_amd.ops.nlp import (CommonSparseFeatures, NGramsFeaturizer,
+                                      StupidBackoffEstimator, TermFrequency,
+                                      Tokenizer)
+    from keystone_amd.parallel.dataset import ItemDataset
+    from keystone_amd.parallel.runtime import get_runtime
+    rt = get_runtime()
+    base, rem = divmod(len(docs_all), world)
+    lo = rank * base + min(rank, rem)
+    hi = lo + base + (1 if rank < rem else 0)
+    docs = ItemDataset(docs_all[lo:hi], rt)
+    toks = Tokenizer().apply_dataset(docs)
+    tf = TermFrequency().apply_dataset(
+        NGramsFeaturizer(1, 2).apply_dataset(toks))
+    vec = CommonSparseFeatures(30)._fit(tf)
+    lm = StupidBackoffEstimator(max_order=2)._fit(toks)
+    score = lm.apply(["the", "cat"])
+    return sorted(vec.index.items(), key=lambda kv: kv[1]), float(score)
+
+
+def test_nlp_featurization_distributed_matches_local():
+    """CommonSparseFeatures top-K (reduceByKey analogue) and the backoff
+    LM's merged counts at world 2 must equal the single-process fit."""
+    docs = ["the cat sat on the mat", "the dog ate the cat food",
+            "a cat and a dog", "the mat was red", "dogs eat food",
+            "red cats sit"] * 3
+    ref_idx, ref_score = _nlp_fit(0, 1, docs)
+    results = run_distributed(_nlp_fit, 2, args=(docs,))
+    for idx, score in results:
+        assert idx == ref_idx
+        assert abs(score - ref_score) < 1e-9
