@@ -285,3 +285,23 @@ def test_nlp_featurization_distributed_matches_local():
     for idx, score in results:
         assert idx == ref_idx
         assert abs(score - ref_score) < 1e-9
+
+
+def _nb_fit(rank, world):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import NaiveBayesEstimator
+    g = torch.Generator().manual_seed(21)
+    X = (torch.rand(60, 12, generator=g) < 0.3).float() * \
+        torch.randint(1, 5, (60, 12), generator=g)
+    y = torch.randint(0, 3, (60,), generator=g)
+    m = NaiveBayesEstimator(3, lam=1.0)._fit(
+        ka.from_rows(_shard(rank, world, X)),
+        ka.from_rows(_shard(rank, world, y)))
+    return m.log_prior, m.log_theta
+
+
+def test_naive_bayes_distributed_matches_local():
+    lp_ref, lt_ref = _nb_fit(0, 1)
+    for lp, lt in run_distributed(_nb_fit, 2):
+        assert torch.allclose(lp, lp_ref, atol=1e-6)
+        assert torch.allclose(lt, lt_ref, atol=1e-6)
