@@ -249,9 +249,12 @@ class ProcessImagePipeline:
                                 "ProcessImagePipeline: a decode worker "
                                 "died; aborting")
                         waited += 2.0
-                        if waited >= 60.0:
+                        # generous: the FIRST batch includes spawn +
+                        # torch import in every worker (minutes on a
+                        # cold box); crashes are caught by is_alive
+                        if waited >= 300.0:
                             raise RuntimeError(
-                                "ProcessImagePipeline: no batch for 60s "
+                                "ProcessImagePipeline: no batch for 300s "
                                 "with live workers; aborting")
                         continue
                     ready[bi] = (slot, cnt)
