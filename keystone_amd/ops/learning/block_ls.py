@@ -211,6 +211,19 @@ def _blocked_trtri(L: torch.Tensor, base: int = 1024) -> torch.Tensor:
     return out
 
 
+def _agreed_epoch(rt, my_epoch: int) -> int:
+    """Checkpoint epoch all ranks agree on: the saved epoch if identical
+    everywhere, else 0 (fresh start).  Resuming mixed epochs would blend
+    residuals from different optimization states through the all-reduced
+    block solves — silently wrong, so it is refused."""
+    if not rt.is_distributed:
+        return my_epoch
+    mm = torch.tensor([float(my_epoch), -float(my_epoch)],
+                      dtype=torch.float64)
+    rt.all_reduce_(mm, op="max")       # [max_e, -min_e]
+    return my_epoch if mm[0].item() == -mm[1].item() else 0
+
+
 def _as_block_getters(data: Dataset, block_size: int):
     """Return (list of zero-arg block getters, runtime, n_local)."""
     if isinstance(data, LazyBlocks):
@@ -273,12 +286,20 @@ class BlockLeastSquaresEstimator(LabelEstimator):
         Ws = [None] * nb
         R = Y.clone()  # residual Y - X W, local shard rows
         start_epoch = 0
-        if self.checkpoint_dir and os.path.exists(self._ckpt_path(rt.rank)):
-            st = torch.load(self._ckpt_path(rt.rank), map_location=Y.device)
-            start_epoch = st["epoch"]
-            Ws = [w.to(Y.device) if w is not None else None
-                  for w in st["Ws"]]
-            R = st["R"].to(Y.device)
+        if self.checkpoint_dir:
+            st = None
+            if os.path.exists(self._ckpt_path(rt.rank)):
+                st = torch.load(self._ckpt_path(rt.rank),
+                                map_location=Y.device)
+            # resume ONLY from an epoch every rank completed: a crash
+            # between per-rank saves leaves mixed epochs, and mixed-epoch
+            # residuals would be silently blended by the all-reduced
+            # block solves
+            start_epoch = _agreed_epoch(rt, 0 if st is None else st["epoch"])
+            if st is not None and start_epoch > 0:
+                Ws = [w.to(Y.device) if w is not None else None
+                      for w in st["Ws"]]
+                R = st["R"].to(Y.device)
 
         # ---- batched factorization pre-pass (same rationale as the
         # weighted solver): block Grams + lam I don't depend on residuals,
@@ -651,13 +672,19 @@ class BlockWeightedLeastSquaresEstimator(LabelEstimator):
         Ws: List[Optional[torch.Tensor]] = [None] * nb
         R = Y.clone()
         start_epoch = 0
-        if self.checkpoint_dir and os.path.exists(self._ckpt_path(rt.rank)):
-            st = torch.load(self._ckpt_path(rt.rank), map_location=Y.device)
-            start_epoch = st["epoch"]
-            Ws = [w.to(Y.device) if w is not None else None for w in st["Ws"]]
-            R = st["R"].to(Y.device)
-            if b_int is not None and st.get("b_int") is not None:
-                b_int = st["b_int"].to(Y.device)
+        if self.checkpoint_dir:
+            st = None
+            if os.path.exists(self._ckpt_path(rt.rank)):
+                st = torch.load(self._ckpt_path(rt.rank),
+                                map_location=Y.device)
+            # see the plain solver: only resume an epoch ALL ranks saved
+            start_epoch = _agreed_epoch(rt, 0 if st is None else st["epoch"])
+            if st is not None and start_epoch > 0:
+                Ws = [w.to(Y.device) if w is not None else None
+                      for w in st["Ws"]]
+                R = st["R"].to(Y.device)
+                if b_int is not None and st.get("b_int") is not None:
+                    b_int = st["b_int"].to(Y.device)
         for _epoch in range(start_epoch, self.num_iter):
             for b in range(nb):
                 Xb = fetch(b)

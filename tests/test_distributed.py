@@ -305,3 +305,41 @@ def test_naive_bayes_distributed_matches_local():
     for lp, lt in run_distributed(_nb_fit, 2):
         assert torch.allclose(lp, lp_ref, atol=1e-6)
         assert torch.allclose(lt, lt_ref, atol=1e-6)
+
+
+def _ckpt_mixed_resume(rank, world, tmpdir):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    g = torch.Generator().manual_seed(8)
+    X = torch.randn(40, 12, generator=g, dtype=torch.float64)
+    Y = torch.randn(40, 2, generator=g, dtype=torch.float64)
+    ds = ka.from_rows(_shard(rank, world, X))
+    ys = ka.from_rows(_shard(rank, world, Y))
+    est = BlockLeastSquaresEstimator(4, 6, 0.1, checkpoint_dir=tmpdir)
+    m = est.fit(ds, ys)
+    return torch.cat(m.blocks, dim=0)
+
+
+def test_block_ls_refuses_mixed_epoch_checkpoints(tmp_path):
+    """A crash between per-rank checkpoint saves leaves ranks at
+    different epochs; resuming must fall back to a fresh fit (identical
+    result), never silently blend mixed-epoch residuals."""
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    g = torch.Generator().manual_seed(8)
+    X = torch.randn(40, 12, generator=g, dtype=torch.float64)
+    Y = torch.randn(40, 2, generator=g, dtype=torch.float64)
+    ref = BlockLeastSquaresEstimator(4, 6, 0.1).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    ref_W = torch.cat(ref.blocks, dim=0)
+
+    ck = tmp_path / "ck"
+    ck.mkdir()
+    # plant a stale epoch-3 checkpoint for rank 0 ONLY (rank 1 has none)
+    stale = {"epoch": 3,
+             "Ws": [torch.zeros(4, 2, dtype=torch.float64)] * 3,
+             "R": torch.zeros(20, 2, dtype=torch.float64)}
+    torch.save(stale, ck / "bls_rank0.pt")
+    results = run_distributed(_ckpt_mixed_resume, 2, args=(str(ck),))
+    for W in results:
+        assert torch.allclose(W, ref_W, atol=1e-8)
