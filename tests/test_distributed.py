@@ -343,3 +343,35 @@ def test_block_ls_refuses_mixed_epoch_checkpoints(tmp_path):
     results = run_distributed(_ckpt_mixed_resume, 2, args=(str(ck),))
     for W in results:
         assert torch.allclose(W, ref_W, atol=1e-8)
+
+
+def _ckpt_two_phase(rank, world, tmpdir):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    g = torch.Generator().manual_seed(8)
+    X = torch.randn(40, 12, generator=g, dtype=torch.float64)
+    Y = torch.randn(40, 2, generator=g, dtype=torch.float64)
+    ds = ka.from_rows(_shard(rank, world, X))
+    ys = ka.from_rows(_shard(rank, world, Y))
+    # phase 1: run 3 epochs with checkpointing (all ranks save epoch 3)
+    BlockLeastSquaresEstimator(4, 3, 0.1, checkpoint_dir=tmpdir).fit(ds, ys)
+    # phase 2: resume to 6 epochs
+    m = BlockLeastSquaresEstimator(4, 6, 0.1, checkpoint_dir=tmpdir).fit(ds, ys)
+    return torch.cat(m.blocks, dim=0)
+
+
+def test_block_ls_distributed_checkpoint_resume(tmp_path):
+    """Clean resume at world 2 (every rank saved the same epoch) must
+    equal the uninterrupted 6-epoch fit exactly."""
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockLeastSquaresEstimator
+    g = torch.Generator().manual_seed(8)
+    X = torch.randn(40, 12, generator=g, dtype=torch.float64)
+    Y = torch.randn(40, 2, generator=g, dtype=torch.float64)
+    ref = BlockLeastSquaresEstimator(4, 6, 0.1).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    ref_W = torch.cat(ref.blocks, dim=0)
+    ck = tmp_path / "ck2"
+    ck.mkdir()
+    for W in run_distributed(_ckpt_two_phase, 2, args=(str(ck),)):
+        assert torch.allclose(W, ref_W, atol=1e-10)
