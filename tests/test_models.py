@@ -186,3 +186,34 @@ def test_mean_ap_hand_computed():
     res11 = ev11.evaluate(ka.from_rows(scores), ka.from_rows(acts))
     expect = (6 * 1.0 + 5 * (2.0 / 3.0)) / 11
     assert abs(res11["mAP"] - expect) < 1e-9
+
+
+def test_classifier_metrics_hand_oracle():
+    """Confusion / per-class / macro-F1 / binary PRF against hand-counted
+    values on a tiny fixed prediction set."""
+    import keystone_amd as ka
+    from keystone_amd.evaluation import (BinaryClassifierEvaluator,
+                                         MulticlassClassifierEvaluator)
+    act = torch.tensor([0, 0, 0, 1, 1, 2])
+    pred = torch.tensor([0, 1, 0, 1, 2, 2])
+    m = MulticlassClassifierEvaluator(3).evaluate(
+        ka.from_rows(pred), ka.from_rows(act))
+    # confusion rows=actual: [[2,1,0],[0,1,1],[0,0,1]]
+    assert m.confusion.tolist() == [[2, 1, 0], [0, 1, 1], [0, 0, 1]]
+    assert abs(m.total_accuracy - 4 / 6) < 1e-12
+    assert torch.allclose(m.per_class_accuracy,
+                          torch.tensor([2 / 3, 1 / 2, 1.0],
+                                       dtype=torch.float64))
+    # per-class precision: [1, 1/2, 1/2]; recall: [2/3, 1/2, 1]
+    f1 = [2 * 1 * (2 / 3) / (1 + 2 / 3), 2 * 0.5 * 0.5 / 1.0,
+          2 * 0.5 * 1 / 1.5]
+    assert abs(m.macro_f1 - sum(f1) / 3) < 1e-12
+
+    b_act = torch.tensor([1, 1, 1, 0, 0, 0])
+    b_pred = torch.tensor([1, 0, 1, 1, 0, 0])
+    bm = BinaryClassifierEvaluator().evaluate(
+        ka.from_rows(b_pred), ka.from_rows(b_act))
+    assert (bm.tp, bm.fp, bm.tn, bm.fn) == (2, 1, 2, 1)
+    assert abs(bm.precision - 2 / 3) < 1e-12
+    assert abs(bm.recall - 2 / 3) < 1e-12
+    assert abs(bm.f1 - 2 / 3) < 1e-12
