@@ -375,3 +375,23 @@ def test_block_ls_distributed_checkpoint_resume(tmp_path):
     ck.mkdir()
     for W in run_distributed(_ckpt_two_phase, 2, args=(str(ck),)):
         assert torch.allclose(W, ref_W, atol=1e-10)
+
+
+def test_gmm_distributed_replicated_world4():
+    """World 4 with uneven shards: the vocab fit must stay bitwise
+    replicated (the 8-GPU GMM path at a non-power world size)."""
+    results = run_distributed(_gmm_fit, 4)
+    for r in results[1:]:
+        for a, b in zip(results[0], r):
+            assert torch.equal(a, b)
+
+
+def test_krr_distributed_world4_matches_local():
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import (GaussianKernelGenerator,
+                                           KernelRidgeRegression)
+    m = KernelRidgeRegression(GaussianKernelGenerator(0.2), 1e-4, 16, 3).fit(
+        ka.from_rows(X_ALL.float()), ka.from_rows(Y_ALL.float()))
+    ref = m.apply_batch(X_ALL.float())
+    for pred in run_distributed(_krr_fit_predict, 4):
+        assert torch.allclose(pred, ref, atol=1e-3)
