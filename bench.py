@@ -153,6 +153,9 @@ def main():
             raise TimeoutError("serve-with-input watchdog")
 
         n_inp = min(4096, args.images_per_gpu)
+        # >= 5 batches so the steady-state window (which excludes the
+        # first batch's worker spinup) still times real work at small n
+        inp_bs = 256 if n_inp >= 1280 else max(1, n_inp // 5)
         blobs = encode_jpeg_batch(imgs.data[:n_inp], workers=16)
 
         def run_loader(loader):
@@ -177,14 +180,14 @@ def main():
         try:
             try:
                 serve_input_ips = run_loader(ProcessImagePipeline(
-                    blobs, list(range(n_inp)), batch_size=256,
+                    blobs, list(range(n_inp)), batch_size=inp_bs,
                     color=not args.gray, device=dev, workers=8))
                 serve_input_mode = "process"
             except Exception as e:
                 print(f"[bench] process input pipeline failed ({e!r}); "
                       "falling back to threads", file=sys.stderr)
                 serve_input_ips = run_loader(ImageBatchPipeline(
-                    blobs, list(range(n_inp)), batch_size=256,
+                    blobs, list(range(n_inp)), batch_size=inp_bs,
                     color=not args.gray, device=dev, workers=16))
                 serve_input_mode = "threads"
         except Exception as e:
