@@ -330,3 +330,24 @@ def test_convolver_matches_naive_loop_oracle():
     for mode in ("im2col", "conv", "fft"):
         out = Convolver(filt, mode=mode).apply_batch(X)
         assert torch.allclose(out.to(torch.float64), ref, atol=1e-4), mode
+
+
+def test_sift_three_scales_shapes_and_norm():
+    """Multi-scale beyond the bench default: descriptor counts follow the
+    per-scale grids (bin x sqrt(2) per scale, VLFeat convention) and
+    normalization holds."""
+    import math as _m
+    imgs = torch.rand(2, 64, 64)
+    s3 = SIFTExtractor(step=4, bin_size=4, num_scales=3)
+    D = s3.apply_batch(imgs)
+    # per-scale anchor count along one axis for L=64:
+    # anchors = range(bin//2, L - 3*bin - bin//2, step)
+    def npos(L, b, st):
+        return len(range(b // 2, L - 3 * b - b // 2, st))
+    bins = [max(1, int(round(4 * (_m.sqrt(2.0) ** s)))) for s in range(3)]
+    assert bins == [4, 6, 8]
+    expect = sum(npos(64, b, 4) ** 2 for b in bins)
+    assert D.shape == (2, expect, 128)
+    nz = D.norm(dim=2)
+    nz = nz[nz > 1e-6]
+    assert torch.allclose(nz, torch.ones_like(nz), atol=1e-4)
