@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import torch
 
-from ..parallel.dataset import Dataset, TensorDataset
 from ..parallel.runtime import get_runtime
 from .classification import _as_label_tensor
 

@@ -17,8 +17,7 @@ streams (weak-scaling benchmarks fix per-rank work, not the global set).
 
 from __future__ import annotations
 
-import math
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -17,7 +17,7 @@ from ..ops.learning import LinearMapEstimator
 from ..ops.learning.whitening import ZCAWhitenerEstimator
 from ..ops.util import ClassLabelIndicatorsFromIntLabels, MaxClassifier
 from ..parallel.dataset import TensorDataset
-from ..workflow.pipeline import FunctionTransformer, Pipeline
+from ..workflow.pipeline import Pipeline
 
 
 @dataclass

@@ -13,7 +13,7 @@ from __future__ import annotations
 
 from dataclasses import dataclass
 
-from ..ops.learning import BlockLeastSquaresEstimator, LinearMapEstimator
+from ..ops.learning import BlockLeastSquaresEstimator
 from ..ops.stats import (CosineRandomFeatures, PaddedFFT, RandomSignNode,
                          StandardScaler)
 from ..ops.util import (ClassLabelIndicatorsFromIntLabels, MaxClassifier,

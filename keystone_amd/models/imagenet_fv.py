@@ -17,7 +17,7 @@ sample fits and for the solver features — "cache" on MI355X means the
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional, Tuple
 
 import torch

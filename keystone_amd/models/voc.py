@@ -8,9 +8,7 @@ GMM -> FisherVector -> normalization -> BlockLeastSquares (multi-label
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Tuple
 
-import torch
 
 from ..ops.images import FisherVector, GrayScaler, SIFTExtractor
 from ..ops.learning import BlockLeastSquaresEstimator, DistributedPCAEstimator

@@ -23,7 +23,6 @@ all-reduces.
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 

@@ -18,7 +18,6 @@ grows with d^2 (Gram) vs per-iteration d*k (LBFGS) — so:
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 

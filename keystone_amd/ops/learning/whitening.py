@@ -8,7 +8,6 @@ LinearDiscriminantAnalysis (Fisher LDA projection).
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 

@@ -21,7 +21,7 @@ decision, not a failure-recovery mechanism.
 
 from __future__ import annotations
 
-from typing import Any, Callable, Iterable, List, Optional, Sequence, Union
+from typing import Any, Callable, Iterable, List, Optional, Sequence
 
 import torch
 

@@ -38,7 +38,7 @@ Per-op debug / dispatch switches (read at the call site, not here):
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass

@@ -18,7 +18,6 @@ Constraints (checked/owned by the caller):
 
 from __future__ import annotations
 
-from typing import Callable
 
 import torch
 

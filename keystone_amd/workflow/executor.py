@@ -15,7 +15,7 @@ and the optional JSONL trace (utils.tracing).
 from __future__ import annotations
 
 import time
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 import torch
 

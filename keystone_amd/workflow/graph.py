@@ -19,7 +19,7 @@ from __future__ import annotations
 
 import itertools
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 __all__ = [
     "GraphId", "NodeId", "SourceId", "SinkId", "Graph", "Operator",

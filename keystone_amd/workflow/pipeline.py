@@ -25,12 +25,11 @@ inside estimators (Gram all-reduce etc.), never in the executor.
 
 from __future__ import annotations
 
-from typing import Any, Callable, List, Optional, Sequence, Union
+from typing import Any, Callable, List, Optional, Sequence
 
 import torch
 
 from ..parallel.dataset import Dataset, ItemDataset, TensorDataset, from_items, from_rows
-from ..parallel.runtime import get_runtime
 from .graph import (DatasetOperator, DatumOperator, DelegatingOperator,
                     EstimatorOperator, Graph, GraphId, NodeId, SourceId,
                     TransformerOperator)

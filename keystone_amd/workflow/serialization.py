@@ -26,7 +26,7 @@ from typing import Dict
 import torch
 
 from ..parallel.runtime import get_runtime
-from .graph import Graph, GraphId, NodeId, SinkId, SourceId, TransformerOperator
+from .graph import Graph, GraphId, TransformerOperator
 from .pipeline import Pipeline, Transformer
 
 FORMAT_VERSION = 1
