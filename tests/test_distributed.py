@@ -395,3 +395,41 @@ def test_krr_distributed_world4_matches_local():
     ref = m.apply_batch(X_ALL.float())
     for pred in run_distributed(_krr_fit_predict, 4):
         assert torch.allclose(pred, ref, atol=1e-3)
+
+
+def _wbls_ckpt(rank, world, tmpdir):
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    g = torch.Generator().manual_seed(15)
+    X = torch.randn(48, 10, generator=g, dtype=torch.float64)
+    y = torch.randint(0, 4, (48,), generator=g)
+    Y = torch.full((48, 4), -1.0, dtype=torch.float64)
+    Y[torch.arange(48), y] = 1.0
+    m = BlockWeightedLeastSquaresEstimator(
+        5, 8, 1e-2, 0.5, checkpoint_dir=tmpdir).fit(
+        ka.from_rows(_shard(rank, world, X)),
+        ka.from_rows(_shard(rank, world, Y)))
+    return torch.cat(m.blocks, dim=0)
+
+
+def test_weighted_refuses_mixed_epoch_checkpoints(tmp_path):
+    """Weighted-solver flavor of the mixed-epoch guard: a stale rank-0
+    checkpoint must be ignored (fresh fit), matching the no-checkpoint
+    reference bitwise across ranks."""
+    import keystone_amd as ka
+    from keystone_amd.ops.learning import BlockWeightedLeastSquaresEstimator
+    g = torch.Generator().manual_seed(15)
+    X = torch.randn(48, 10, generator=g, dtype=torch.float64)
+    y = torch.randint(0, 4, (48,), generator=g)
+    Y = torch.full((48, 4), -1.0, dtype=torch.float64)
+    Y[torch.arange(48), y] = 1.0
+    ref = BlockWeightedLeastSquaresEstimator(5, 8, 1e-2, 0.5).fit(
+        ka.from_rows(X), ka.from_rows(Y))
+    ref_W = torch.cat(ref.blocks, dim=0)
+    ck = tmp_path / "wck"
+    ck.mkdir()
+    stale = {"epoch": 5, "Ws": [torch.zeros(5, 4, dtype=torch.float64)] * 2,
+             "R": torch.zeros(24, 4, dtype=torch.float64), "b_int": None}
+    torch.save(stale, ck / "bwls_rank0.pt")
+    for W in run_distributed(_wbls_ckpt, 2, args=(str(ck),)):
+        assert torch.allclose(W, ref_W, atol=1e-10)
