@@ -163,6 +163,8 @@ def gather(branches: Sequence[PipelineBase]) -> Pipeline:
     """``gather([p1..pk]) : A -> (B1..Bk)`` — apply every branch to the same
     input and tuple the results (ref: Pipeline.gather).  Follow with
     ``VectorCombiner`` to concatenate feature branches."""
+    if not branches:
+        raise ValueError("gather() needs at least one branch")
     g = Graph()
     s = g.add_source()
     outs = []
