@@ -246,6 +246,15 @@ def main():
         }
         print(json.dumps(out))
 
+    # tear the process group down EXPLICITLY: exiting with a live group
+    # races its comm threads against interpreter shutdown — observed as a
+    # rare SIGABRT ("terminate called without an active exception") on
+    # gloo rank teardown, and it is the documented-clean path for RCCL
+    sys.stdout.flush()
+    rt.barrier()
+    from keystone_amd.parallel.runtime import shutdown_runtime
+    shutdown_runtime()
+
 
 if __name__ == "__main__":
     main()
