@@ -56,6 +56,10 @@ class ImageNetFVConfig:
     lam: float = 1e-5
     mixture_weight: float = 0.5
     top_k: int = 5
+    # epoch-boundary solver checkpointing (failure recovery): a re-run
+    # with the same dir resumes the block solver from the last epoch
+    # every rank completed
+    solver_checkpoint_dir: Optional[str] = None
     seed: int = 0
     gmm_iters: int = 20
     kmeans_iters: int = 5
@@ -148,7 +152,8 @@ def fit_imagenet_fv(train_imgs: TensorDataset, labels_int: TensorDataset,
     ind = ClassLabelIndicatorsFromIntLabels(conf.num_classes)
     Y = ind.apply_batch(labels_int.data).float()
     solver = BlockWeightedLeastSquaresEstimator(
-        conf.block_size, conf.solver_epochs, conf.lam, conf.mixture_weight)
+        conf.block_size, conf.solver_epochs, conf.lam, conf.mixture_weight,
+        checkpoint_dir=conf.solver_checkpoint_dir)
     model = solver.fit(TensorDataset(feats, rt), TensorDataset(Y, rt))
     _sync()
     info["t_solver"] = time.perf_counter() - t2

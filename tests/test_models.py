@@ -217,3 +217,28 @@ def test_classifier_metrics_hand_oracle():
     assert abs(bm.precision - 2 / 3) < 1e-12
     assert abs(bm.recall - 2 / 3) < 1e-12
     assert abs(bm.f1 - 2 / 3) < 1e-12
+
+
+def test_imagenet_fv_solver_checkpoint_resume(tmp_path):
+    """Flagship-API failure recovery: interrupt after 2 solver epochs,
+    resume to 4 via solver_checkpoint_dir — identical model to the
+    uninterrupted 4-epoch fit (featurization is deterministic per seed)."""
+    import torch
+    from keystone_amd.loaders.synthetic import synthetic_images
+    from keystone_amd.models.imagenet_fv import (ImageNetFVConfig,
+                                                 fit_imagenet_fv)
+    imgs, labels = synthetic_images(16, h=48, w=48, num_classes=3, seed=3)
+    kw = dict(num_classes=3, sift_step=6, sift_bin=4, sift_scales=1,
+              lcs_step=6, lcs_subpatch=4, pca_dims=8, gmm_k=4,
+              n_sample=2000, block_size=16, lam=0.1, top_k=1, gmm_iters=3)
+    _, full = fit_imagenet_fv(imgs, labels,
+                              ImageNetFVConfig(solver_epochs=4, **kw))
+    ck = str(tmp_path / "solver_ck")
+    fit_imagenet_fv(imgs, labels, ImageNetFVConfig(
+        solver_epochs=2, solver_checkpoint_dir=ck, **kw))
+    _, resumed = fit_imagenet_fv(imgs, labels, ImageNetFVConfig(
+        solver_epochs=4, solver_checkpoint_dir=ck, **kw))
+    W_full = torch.cat(list(full["model"].blocks), dim=0)
+    W_res = torch.cat(list(resumed["model"].blocks), dim=0)
+    assert torch.allclose(W_full, W_res, atol=1e-5), \
+        (W_full - W_res).abs().max()
