@@ -50,6 +50,8 @@ def main():
     im.add_argument("--num-classes", type=int, default=1000)
     im.add_argument("--gmm-k", type=int, default=64)
     im.add_argument("--save", help="save the fitted pipeline here")
+    im.add_argument("--solver-checkpoint", help="epoch checkpoint dir "
+                    "(re-run with the same dir to resume the solver)")
 
     args = p.parse_args()
 
@@ -170,7 +172,8 @@ def main():
                                         w=args.image_size,
                                         num_classes=args.num_classes)
         conf = ImageNetFVConfig(num_classes=args.num_classes,
-                                gmm_k=args.gmm_k)
+                                gmm_k=args.gmm_k,
+                                solver_checkpoint_dir=args.solver_checkpoint)
         pipe, info = fit_imagenet_fv(imgs, labels, conf)
         if args.save:
             pipe.save(args.save)
