@@ -25,7 +25,8 @@ class AugmentedExamplesEvaluator:
         agg = torch.zeros(uniq.shape[0], S.shape[1], dtype=S.dtype,
                           device=S.device)
         agg.index_add_(0, inv, S)
-        # first occurrence's actual label per image
+        # one occurrence's label per image (augmentations of the same
+        # image share a label, so which occurrence wins is immaterial)
         lab = torch.zeros(uniq.shape[0], dtype=torch.long, device=S.device)
         lab.scatter_(0, inv, act)
         pred = agg.argmax(1)
